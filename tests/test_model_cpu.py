@@ -1,0 +1,103 @@
+"""CPU tests for the Llama model + training engine (fp32, tiny config)."""
+import os
+
+import pytest
+import torch
+
+from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
+from torch_on_k8s_amd.models.llama import LlamaModel, get_config
+from torch_on_k8s_amd.parallel.env import DistContext
+
+
+def tiny_cfg(**kw):
+    return TrainerConfig(model="llama-tiny", micro_batch=2, seq_len=64, **kw)
+
+
+def test_forward_loss_finite():
+    cfg = get_config("llama-tiny")
+    model = LlamaModel(cfg)
+    ids = torch.randint(0, cfg.vocab_size, (2, 32))
+    loss = model(ids, ids)
+    assert torch.isfinite(loss)
+
+
+def test_loss_decreases_on_fixed_batch():
+    torch.manual_seed(0)
+    ctx = DistContext()
+    tr = Trainer(tiny_cfg(lr=1e-3), ctx)
+    # fix the batch: same seed + step index => same data
+    losses = []
+    for _ in range(8):
+        tr.step_count = 0  # same synthetic batch every time
+        losses.append(tr.train_step())
+        tr.step_count = 1
+    assert losses[-1] < losses[0], losses
+
+
+def test_activation_checkpointing_matches():
+    torch.manual_seed(0)
+    cfg = get_config("llama-tiny")
+    m1 = LlamaModel(cfg)
+    torch.manual_seed(0)
+    m2 = LlamaModel(cfg, activation_checkpointing=True)
+    m2.load_state_dict(m1.state_dict())
+    ids = torch.randint(0, cfg.vocab_size, (2, 32))
+    l1 = m1(ids, ids)
+    l2 = m2(ids, ids)
+    l1.backward()
+    l2.backward()
+    assert torch.allclose(l1, l2, atol=1e-6)
+    g1 = m1.layers[0].attn.q_proj.weight.grad
+    g2 = m2.layers[0].attn.q_proj.weight.grad
+    assert torch.allclose(g1, g2, atol=1e-5)
+
+
+def test_checkpoint_save_resume(tmp_path):
+    ctx = DistContext()
+    tr = Trainer(tiny_cfg(), ctx)
+    for _ in range(2):
+        tr.train_step()
+    ck = str(tmp_path / "ckpt")
+    tr.save_checkpoint(ck)
+    assert os.path.exists(os.path.join(ck, "meta.json"))
+
+    tr2 = Trainer(tiny_cfg(), ctx)
+    tr2.load_checkpoint(ck)
+    assert tr2.step_count == 2
+    # identical state => identical next loss
+    l1 = tr.train_step()
+    l2 = tr2.train_step()
+    assert abs(l1 - l2) < 1e-6
+
+
+def test_flat_buckets_preserve_params():
+    """Flattening must not change parameter values or training math."""
+    torch.manual_seed(0)
+    cfg = get_config("llama-tiny")
+    model = LlamaModel(cfg)
+    before = {k: v.clone() for k, v in model.state_dict().items()}
+    from torch_on_k8s_amd.parallel.ddp import FlatBucketModel
+    fb = FlatBucketModel(model, bucket_mb=1)
+    after = model.state_dict()
+    for k in before:
+        assert torch.equal(before[k], after[k]), k
+    assert len(fb.buckets) > 1  # tiny bucket size forces multiple buckets
+
+
+def test_grad_accumulates_into_flat_buffer():
+    torch.manual_seed(0)
+    cfg = get_config("llama-tiny")
+    model = LlamaModel(cfg)
+    from torch_on_k8s_amd.parallel.ddp import FlatBucketModel
+    fb = FlatBucketModel(model, bucket_mb=4)
+    fb.zero_grads()
+    ids = torch.randint(0, cfg.vocab_size, (2, 32))
+    model(ids, ids).backward()
+    # every param's grad view must alias its bucket's flat_grad
+    for b in fb.buckets:
+        for s in b.segs:
+            assert s.param.grad is not None
+            assert s.param.grad.data_ptr() == \
+                b.flat_grad[s.offset:s.offset + s.numel].data_ptr()
+    total = sum(float(b.flat_grad.abs().sum()) for b in fb.buckets)
+    assert total > 0

@@ -1,0 +1,160 @@
+"""Training engine: the data plane the reference outsources to user
+containers, built MI355X-native.
+
+One process per GPU; DP gradient sync over RCCL/xGMI via flat buckets
+(parallel/ddp.py); fused gfx950 kernels for the hot ops; checkpoint /
+resume compatible with the control plane's elastic protocol (SURVEY.md
+§5.4: the operator coordinates checkpoints via annotations, the data
+plane writes them).
+"""
+from __future__ import annotations
+
+import json
+import os
+import time
+from dataclasses import dataclass, field, asdict
+
+import torch
+
+from torch_on_k8s_amd.engine.data import SyntheticTokens
+from torch_on_k8s_amd.models.llama import LlamaModel, get_config
+from torch_on_k8s_amd.parallel.ddp import FlatBucketModel, FlatAdamW
+from torch_on_k8s_amd.parallel.env import DistContext
+
+
+@dataclass
+class TrainerConfig:
+    model: str = "llama3-8b"
+    model_overrides: dict = field(default_factory=dict)
+    micro_batch: int = 2
+    seq_len: int = 4096
+    lr: float = 3e-4
+    weight_decay: float = 0.1
+    betas: tuple = (0.9, 0.95)
+    grad_clip: float = 0.0          # 0 disables (extra HBM pass when on)
+    bucket_mb: int = 256
+    activation_checkpointing: bool = False
+    overlap_grad_sync: bool = True
+    dtype: str = "bf16"             # compute dtype on GPU
+    seed: int = 1234
+    metrics_path: str | None = None  # structured metrics for the elastic
+                                     # autoscaler (replaces the reference's
+                                     # stdout log-regex, observation.go:40-106)
+
+    def to_dict(self):
+        d = asdict(self)
+        return d
+
+
+class Trainer:
+    def __init__(self, cfg: TrainerConfig, ctx: DistContext):
+        self.cfg = cfg
+        self.ctx = ctx
+        self.device = ctx.device
+        torch.manual_seed(cfg.seed)  # same init on every rank
+
+        mcfg = get_config(cfg.model, **cfg.model_overrides)
+        self.model_cfg = mcfg
+        model = LlamaModel(mcfg, activation_checkpointing=cfg.activation_checkpointing)
+        if self.device.type == "cuda" and cfg.dtype == "bf16":
+            model = model.to(torch.bfloat16)
+        model = model.to(self.device)
+        self.fb = FlatBucketModel(
+            model, bucket_mb=cfg.bucket_mb, overlap=cfg.overlap_grad_sync)
+        self.opt = FlatAdamW(
+            self.fb, lr=cfg.lr, betas=tuple(cfg.betas),
+            weight_decay=cfg.weight_decay)
+        self.data = SyntheticTokens(
+            mcfg.vocab_size, cfg.micro_batch, cfg.seq_len, self.device,
+            rank=ctx.rank, seed=cfg.seed)
+        self.step_count = 0
+        self._t_last = None
+
+    @property
+    def module(self) -> LlamaModel:
+        return self.fb.module
+
+    def train_step(self, sync: bool = True):
+        """One full step: data -> fwd -> bwd (overlapped RCCL all-reduce)
+        -> optional clip -> fused AdamW.
+
+        sync=True returns the host loss (device sync); sync=False returns
+        the detached loss tensor without host synchronisation (bench path).
+        """
+        t0 = time.perf_counter()
+        self.fb.zero_grads()
+        inp, lab = self.data.batch(self.step_count)
+        loss = self.fb(inp, lab)
+        loss.backward()
+        self.fb.finish_grad_sync()
+        scale = 1.0 / self.fb.world_size
+        if self.cfg.grad_clip > 0:
+            norm = self.fb.grad_norm()
+            coef = self.cfg.grad_clip / (norm + 1e-6)
+            coef = torch.clamp(coef, max=1.0)
+            scale = scale * coef.item()
+        self.opt.step(grad_scale=scale)
+        self.step_count += 1
+        if not sync:
+            return loss.detach()
+        out = loss.detach().float().item()
+        self._t_last = time.perf_counter() - t0
+        if self.cfg.metrics_path and self.ctx.is_main:
+            self._write_metrics(out)
+        return out
+
+    def tokens_per_step(self) -> int:
+        return self.cfg.micro_batch * self.cfg.seq_len * self.fb.world_size
+
+    def _write_metrics(self, loss: float):
+        """Structured metrics endpoint for the elastic autoscaler
+        (reference parses worker-0 stdout with a regex,
+        torchelastic/observation.go:40-106 - we write JSON instead)."""
+        rec = {
+            "step": self.step_count,
+            "loss": loss,
+            "step_time_s": self._t_last,
+            "tokens_per_s": self.tokens_per_step() / self._t_last
+            if self._t_last else None,
+            "world_size": self.fb.world_size,
+            "ts": time.time(),
+        }
+        tmp = self.cfg.metrics_path + ".tmp"
+        with open(tmp, "w") as f:
+            json.dump(rec, f)
+        os.replace(tmp, self.cfg.metrics_path)
+
+    # ---- checkpoint / resume (elastic protocol) ------
+    def save_checkpoint(self, path: str):
+        """Atomic checkpoint written by rank 0 (DP replicas are identical)."""
+        if not self.ctx.is_main:
+            return
+        os.makedirs(path + ".tmp", exist_ok=True)
+        # state_dict values are views into flat buckets; clone so torch.save
+        # serialises each tensor's bytes, not the whole bucket storage.
+        msd = {k: v.detach().cpu().clone()
+               for k, v in self.module.state_dict().items()}
+        torch.save(msd, os.path.join(path + ".tmp", "model.pt"))
+        torch.save(self.opt.state_dict(), os.path.join(path + ".tmp", "optim.pt"))
+        meta = {
+            "step": self.step_count,
+            "model": self.cfg.model,
+            "model_config": self.model_cfg.to_dict(),
+            "trainer_config": self.cfg.to_dict(),
+        }
+        with open(os.path.join(path + ".tmp", "meta.json"), "w") as f:
+            json.dump(meta, f, indent=2)
+        if os.path.exists(path):
+            import shutil
+            shutil.rmtree(path)
+        os.replace(path + ".tmp", path)
+
+    def load_checkpoint(self, path: str):
+        sd = torch.load(os.path.join(path, "model.pt"), map_location=self.device,
+                        weights_only=True)
+        self.module.load_state_dict(sd)
+        osd = torch.load(os.path.join(path, "optim.pt"), map_location=self.device,
+                         weights_only=True)
+        self.opt.load_state_dict(osd)
+        with open(os.path.join(path, "meta.json")) as f:
+            self.step_count = json.load(f)["step"]

@@ -1,0 +1,219 @@
+"""Llama-family transformer, MI355X-first.
+
+This is the flagship model of the data plane (SURVEY.md §7 step 2 /
+BASELINE.json: gang-scheduled Llama-3-8B tokens/sec is the headline
+metric). Design choices for MI355X:
+
+  * bf16 weights/activations end-to-end; fp32 accumulation inside the
+    fused HIP ops (RMSNorm, RoPE, flash attention, AdamW).
+  * Plain projection GEMMs go through hipBLASLt/rocBLAS via torch.matmul;
+    everything fusable is a handwritten gfx950 kernel in ops/.
+  * RoPE cos/sin tables precomputed on host (fp32) -- no device trig.
+  * 288 GB HBM3E per GPU: default training config holds the whole model,
+    grads and fp32 Adam state resident without sharding; activation
+    checkpointing is optional and off by default.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field, asdict
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from torch_on_k8s_amd import ops
+
+
+@dataclass
+class LlamaConfig:
+    name: str = "llama3-8b"
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    max_seq_len: int = 8192
+    rope_theta: float = 500000.0
+    rms_eps: float = 1e-5
+    tie_embeddings: bool = False
+    # "hip": fused gfx950 flash-attention kernel (CPU: fp32 reference);
+    # "sdpa": torch scaled_dot_product_attention (used for A/B comparison).
+    attn_impl: str = "hip"
+    init_std: float = 0.02
+
+    def to_dict(self):
+        return asdict(self)
+
+
+PRESETS = {
+    # The headline config named by BASELINE.json.
+    "llama3-8b": LlamaConfig(),
+    # Single-layer-scale configs for tests / smoke.
+    "llama-tiny": LlamaConfig(
+        name="llama-tiny", vocab_size=512, hidden_size=256,
+        intermediate_size=512, num_layers=2, num_heads=4, num_kv_heads=2,
+        head_dim=64, max_seq_len=256, rope_theta=10000.0),
+    # Mid-size config for single-GPU kernel iteration.
+    "llama-1b": LlamaConfig(
+        name="llama-1b", vocab_size=32000, hidden_size=2048,
+        intermediate_size=5504, num_layers=16, num_heads=16, num_kv_heads=8,
+        head_dim=128, max_seq_len=4096, rope_theta=500000.0),
+}
+
+
+def get_config(name: str, **overrides) -> LlamaConfig:
+    cfg = PRESETS[name]
+    if overrides:
+        d = cfg.to_dict()
+        d.update(overrides)
+        cfg = LlamaConfig(**d)
+    return cfg
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, hidden: int, eps: float):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden))
+        self.eps = eps
+
+    def forward(self, x):
+        return ops.rmsnorm(x, self.weight, self.eps)
+
+
+def build_rope_table(cfg: LlamaConfig, seq_len: int, device) -> tuple[torch.Tensor, torch.Tensor]:
+    """Host-precomputed fp32 [S, D/2] cos/sin tables."""
+    half = cfg.head_dim // 2
+    inv_freq = 1.0 / (cfg.rope_theta ** (torch.arange(0, half, dtype=torch.float64) / half))
+    t = torch.arange(seq_len, dtype=torch.float64)
+    ang = torch.outer(t, inv_freq)
+    return (ang.cos().float().contiguous().to(device),
+            ang.sin().float().contiguous().to(device))
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        H, D = cfg.num_heads, cfg.head_dim
+        Hkv = cfg.num_kv_heads
+        self.q_proj = nn.Linear(cfg.hidden_size, H * D, bias=False)
+        self.k_proj = nn.Linear(cfg.hidden_size, Hkv * D, bias=False)
+        self.v_proj = nn.Linear(cfg.hidden_size, Hkv * D, bias=False)
+        self.o_proj = nn.Linear(H * D, cfg.hidden_size, bias=False)
+
+    def forward(self, x, cos, sin):
+        B, S, _ = x.shape
+        cfg = self.cfg
+        q = self.q_proj(x).view(B, S, cfg.num_heads, cfg.head_dim)
+        k = self.k_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        v = self.v_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        q = ops.apply_rope(q, cos, sin)
+        k = ops.apply_rope(k, cos, sin)
+        if cfg.attn_impl == "sdpa" and x.is_cuda:
+            rep = cfg.num_heads // cfg.num_kv_heads
+            qt = q.transpose(1, 2)
+            kt = k.transpose(1, 2).repeat_interleave(rep, dim=1)
+            vt = v.transpose(1, 2).repeat_interleave(rep, dim=1)
+            o = F.scaled_dot_product_attention(qt, kt, vt, is_causal=True)
+            o = o.transpose(1, 2).contiguous()
+        else:
+            o = ops.attention(q, k, v, causal=True)
+        return self.o_proj(o.reshape(B, S, cfg.num_heads * cfg.head_dim))
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
+        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
+        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
+
+    def forward(self, x):
+        return self.down_proj(F.silu(self.gate_proj(x)) * self.up_proj(x))
+
+
+class Block(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.input_norm = RMSNorm(cfg.hidden_size, cfg.rms_eps)
+        self.attn = Attention(cfg)
+        self.post_attn_norm = RMSNorm(cfg.hidden_size, cfg.rms_eps)
+        self.mlp = MLP(cfg)
+
+    def forward(self, x, cos, sin):
+        x = x + self.attn(self.input_norm(x), cos, sin)
+        x = x + self.mlp(self.post_attn_norm(x))
+        return x
+
+
+class LlamaModel(nn.Module):
+    """Decoder-only Llama. forward() returns mean next-token CE loss."""
+
+    def __init__(self, cfg: LlamaConfig, activation_checkpointing: bool = False):
+        super().__init__()
+        self.cfg = cfg
+        self.activation_checkpointing = activation_checkpointing
+        self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.layers = nn.ModuleList(Block(cfg) for _ in range(cfg.num_layers))
+        self.norm = RMSNorm(cfg.hidden_size, cfg.rms_eps)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        if cfg.tie_embeddings:
+            self.lm_head.weight = self.embed_tokens.weight
+        self._rope_cache: tuple[int, torch.Tensor, torch.Tensor] | None = None
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        std = self.cfg.init_std
+        for m in self.modules():
+            if isinstance(m, nn.Linear):
+                nn.init.normal_(m.weight, mean=0.0, std=std)
+            elif isinstance(m, nn.Embedding):
+                nn.init.normal_(m.weight, mean=0.0, std=std)
+            elif isinstance(m, RMSNorm):
+                nn.init.ones_(m.weight)
+        # scale down output projections per residual-branch count
+        scale = (2 * self.cfg.num_layers) ** -0.5
+        for blk in self.layers:
+            with torch.no_grad():
+                blk.attn.o_proj.weight.mul_(scale)
+                blk.mlp.down_proj.weight.mul_(scale)
+
+    def _rope(self, S: int, device):
+        if self._rope_cache is None or self._rope_cache[0] != S or \
+                self._rope_cache[1].device != device:
+            cos, sin = build_rope_table(self.cfg, S, device)
+            self._rope_cache = (S, cos, sin)
+        return self._rope_cache[1], self._rope_cache[2]
+
+    def forward_hidden(self, input_ids: torch.Tensor) -> torch.Tensor:
+        B, S = input_ids.shape
+        cos, sin = self._rope(S, input_ids.device)
+        x = self.embed_tokens(input_ids)
+        for blk in self.layers:
+            if self.activation_checkpointing and self.training:
+                x = torch.utils.checkpoint.checkpoint(
+                    blk, x, cos, sin, use_reentrant=False)
+            else:
+                x = blk(x, cos, sin)
+        return self.norm(x)
+
+    def forward(self, input_ids: torch.Tensor,
+                labels: torch.Tensor | None = None):
+        x = self.forward_hidden(input_ids)
+        logits = self.lm_head(x)
+        if labels is None:
+            return logits
+        loss = F.cross_entropy(
+            logits.float().view(-1, self.cfg.vocab_size), labels.view(-1))
+        return loss
+
+    def num_params(self) -> int:
+        seen, total = set(), 0
+        for p in self.parameters():
+            if id(p) not in seen:
+                seen.add(id(p))
+                total += p.numel()
+        return total

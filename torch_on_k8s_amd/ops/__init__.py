@@ -1,0 +1,199 @@
+"""MI355X-native fused ops.
+
+Dispatch policy:
+  * GPU tensors ALWAYS run the handwritten gfx950 HIP kernels from the
+    in-tree extension ``torch_on_k8s_amd.ops._C``. If the extension is
+    missing on a GPU machine, ops raise instead of silently falling back
+    to eager PyTorch.
+  * CPU tensors use plain fp32 PyTorch reference implementations (used by
+    the CPU test suite and as the numerics oracle for the HIP kernels).
+
+Capability parity: the reference (hliangzhao/torch-on-k8s) ships no
+kernels at all -- it schedules opaque containers. These ops
+are the data plane its env-var contract assumes exists inside the
+container, built MI355X-first per SURVEY.md §7 step 2.
+"""
+from __future__ import annotations
+
+import torch
+
+try:  # the in-tree gfx950 extension; built by `setup.py build_ext --inplace`
+    from . import _C  # type: ignore
+except ImportError:  # pragma: no cover - exercised only on unbuilt checkouts
+    _C = None
+
+
+def hip_ext_available() -> bool:
+    return _C is not None
+
+
+def _require_ext(op: str):
+    if _C is None:
+        raise RuntimeError(
+            f"torch_on_k8s_amd HIP extension is required for {op} on GPU but "
+            "is not built. Run: PYTORCH_ROCM_ARCH=gfx950 python setup.py "
+            "build_ext --inplace"
+        )
+    return _C
+
+
+# --------------------------------------------------------------------------
+# RMSNorm
+# --------------------------------------------------------------------------
+def rmsnorm_ref(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+    """fp32 reference: y = x * w / sqrt(mean(x^2) + eps)."""
+    xf = x.float()
+    r = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (xf * r * w.float()).to(x.dtype)
+
+
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        if x.is_cuda:
+            y, invr = _require_ext("rmsnorm").rmsnorm_fwd(x, w, eps)
+        else:
+            xf = x.float()
+            invr = torch.rsqrt(xf.pow(2).mean(-1) + eps).reshape(-1)
+            y = (xf * invr.view(*x.shape[:-1], 1) * w.float()).to(x.dtype)
+        ctx.save_for_backward(x, w, invr)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, invr = ctx.saved_tensors
+        dy = dy.contiguous()
+        if x.is_cuda:
+            dx, dw = _C.rmsnorm_bwd(x, w, dy, invr)
+        else:
+            xf, wf, dyf = x.float(), w.float(), dy.float()
+            H = x.shape[-1]
+            r = invr.view(*x.shape[:-1], 1)
+            c = (dyf * wf * xf).sum(-1, keepdim=True)
+            dx = (r * (wf * dyf - xf * (r * r / H) * c)).to(x.dtype)
+            dw = (dyf * xf * r).reshape(-1, H).sum(0)
+        return dx, dw.to(w.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    return _RMSNormFn.apply(x.contiguous(), w, eps)
+
+
+# --------------------------------------------------------------------------
+# RoPE (Llama rotate-half convention)
+# --------------------------------------------------------------------------
+def rope_ref(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+             sign: float = 1.0) -> torch.Tensor:
+    """x: [B, S, H, D]; cos/sin: [S, D/2] fp32."""
+    B, S, H, D = x.shape
+    xf = x.float()
+    x1, x2 = xf[..., : D // 2], xf[..., D // 2:]
+    cs = cos.view(1, S, 1, D // 2)
+    sn = sin.view(1, S, 1, D // 2) * sign
+    return torch.cat([x1 * cs - x2 * sn, x2 * cs + x1 * sn], dim=-1).to(x.dtype)
+
+
+class _RopeFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos, sin):
+        ctx.save_for_backward(cos, sin)
+        ctx.heads = x.shape[2]
+        if x.is_cuda:
+            return _require_ext("rope").rope(x.contiguous(), cos, sin, x.shape[2], 1.0)
+        return rope_ref(x, cos, sin, 1.0)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos, sin = ctx.saved_tensors
+        if dy.is_cuda:
+            dx = _C.rope(dy.contiguous(), cos, sin, ctx.heads, -1.0)
+        else:
+            dx = rope_ref(dy, cos, sin, -1.0)
+        return dx, None, None
+
+
+def apply_rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+    """Apply rotary embedding to [B, S, H, D] activations."""
+    return _RopeFn.apply(x, cos, sin)
+
+
+# --------------------------------------------------------------------------
+# Fused AdamW on a flat bucket
+# --------------------------------------------------------------------------
+def fused_adamw_(p: torch.Tensor, g: torch.Tensor, m: torch.Tensor,
+                 v: torch.Tensor, *, lr: float, beta1: float, beta2: float,
+                 eps: float, weight_decay: float, step: int,
+                 grad_scale: float = 1.0) -> None:
+    """In-place AdamW over one flat bucket (p/g bf16, m/v fp32)."""
+    if p.is_cuda:
+        _require_ext("fused_adamw").adamw_(p, g, m, v, lr, beta1, beta2, eps,
+                                           weight_decay, step, grad_scale)
+        return
+    gf = g.float() * grad_scale
+    pf = p.float()
+    pf -= lr * weight_decay * pf
+    m.mul_(beta1).add_(gf, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
+    bc1 = 1.0 / (1.0 - beta1 ** step)
+    bc2 = 1.0 / (1.0 - beta2 ** step)
+    pf -= lr * (m * bc1) / ((v * bc2).sqrt() + eps)
+    p.copy_(pf.to(p.dtype))
+
+
+# --------------------------------------------------------------------------
+# Flash attention (causal, GQA)
+# --------------------------------------------------------------------------
+def attention_ref(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                  causal: bool = True) -> torch.Tensor:
+    """fp32 reference attention. q: [B,S,Hq,D], k/v: [B,S,Hkv,D]."""
+    B, S, Hq, D = q.shape
+    Hkv = k.shape[2]
+    rep = Hq // Hkv
+    qf = q.float().permute(0, 2, 1, 3)  # B,Hq,S,D
+    kf = k.float().permute(0, 2, 1, 3).repeat_interleave(rep, dim=1)
+    vf = v.float().permute(0, 2, 1, 3).repeat_interleave(rep, dim=1)
+    s = torch.matmul(qf, kf.transpose(-1, -2)) / (D ** 0.5)
+    if causal:
+        mask = torch.triu(torch.ones(S, S, dtype=torch.bool, device=q.device), 1)
+        s = s.masked_fill(mask, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    o = torch.matmul(p, vf)
+    return o.permute(0, 2, 1, 3).to(q.dtype)
+
+
+class _FlashAttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal):
+        if q.is_cuda:
+            ext = _require_ext("flash_attention")
+            if not hasattr(ext, "attn_fwd"):
+                raise RuntimeError(
+                    "HIP flash attention kernel missing from extension; "
+                    "rebuild the extension")
+            o, lse = ext.attn_fwd(q, k, v, causal)
+            ctx.save_for_backward(q, k, v, o, lse)
+            ctx.causal = causal
+            return o
+        # CPU path: differentiable reference (no custom backward needed).
+        raise RuntimeError("use attention() entry point for CPU tensors")
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = _C.attn_bwd(q, k, v, o, lse, do.contiguous(), ctx.causal)
+        return dq, dk, dv, None
+
+
+def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+              causal: bool = True) -> torch.Tensor:
+    """Causal GQA attention: HIP flash kernel on GPU, reference on CPU."""
+    if q.is_cuda:
+        return _FlashAttnFn.apply(q.contiguous(), k.contiguous(),
+                                  v.contiguous(), causal)
+    return attention_ref(q, k, v, causal)
+
+
+__all__ = [
+    "rmsnorm", "rmsnorm_ref", "apply_rope", "rope_ref", "fused_adamw_",
+    "attention", "attention_ref", "hip_ext_available",
+]

@@ -1,0 +1,151 @@
+// RMSNorm forward/backward for MI355X (gfx950).
+//
+// Capability parity note: the reference operator (hliangzhao/torch-on-k8s)
+// delegates all compute to user containers; this kernel is part of the
+// MI355X-native data plane that SURVEY.md §0/§7 requires (fused hot ops:
+// RMSNorm / RoPE / Adam / flash-attention).
+//
+// Design (memory-bound op, target HBM ceiling ~6.3 TB/s):
+//  - bf16 I/O loaded 16 B/lane (bf16x8) — hipcc does not auto-vectorize
+//    scalar bf16 loads (guide G13: scalar is ~2x slower).
+//  - fp32 accumulation; one block per row, wave64 + LDS cross-wave reduce.
+//  - inv_rms saved (fp32 per row) for backward.
+//  - backward dweight uses a column-parallel kernel with per-block
+//    row-striding and one atomicAdd per column per block (G12).
+#include "common.h"
+
+namespace {
+
+constexpr int BLOCK = 256;
+
+__global__ void rmsnorm_fwd_kernel(const bf16x8* __restrict__ x,
+                                   const bf16x8* __restrict__ w,
+                                   bf16x8* __restrict__ y,
+                                   float* __restrict__ invr,
+                                   long nrows, int hc /* H/8 */, float eps) {
+  __shared__ float red[BLOCK / WAVE];
+  const int H = hc * 8;
+  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const bf16x8* xr = x + row * hc;
+    float ss = 0.f;
+    for (int c = threadIdx.x; c < hc; c += BLOCK) {
+      bf16x8 v = xr[c];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bfbits2f(v.h[j]);
+        ss = fmaf(f, f, ss);
+      }
+    }
+    ss = block_reduce_sum(ss, red);
+    const float r = rsqrtf(ss / (float)H + eps);
+    if (threadIdx.x == 0) invr[row] = r;
+    bf16x8* yr = y + row * hc;
+    for (int c = threadIdx.x; c < hc; c += BLOCK) {
+      bf16x8 xv = xr[c], wv = w[c], ov;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bfbits2f(xv.h[j]) * r * bfbits2f(wv.h[j]);
+        ov.h[j] = f2bfbits(f);
+      }
+      yr[c] = ov;
+    }
+    __syncthreads();  // red[] reused next row
+  }
+}
+
+// dx_i = r * (w_i*dy_i - x_i * r^2/H * sum_j(dy_j*w_j*x_j))
+__global__ void rmsnorm_bwd_dx_kernel(const bf16x8* __restrict__ x,
+                                      const bf16x8* __restrict__ w,
+                                      const bf16x8* __restrict__ dy,
+                                      const float* __restrict__ invr,
+                                      bf16x8* __restrict__ dx,
+                                      long nrows, int hc) {
+  __shared__ float red[BLOCK / WAVE];
+  const int H = hc * 8;
+  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const bf16x8* xr = x + row * hc;
+    const bf16x8* dyr = dy + row * hc;
+    const float r = invr[row];
+    float acc = 0.f;
+    for (int c = threadIdx.x; c < hc; c += BLOCK) {
+      bf16x8 xv = xr[c], wv = w[c], dv = dyr[c];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        acc = fmaf(bfbits2f(dv.h[j]) * bfbits2f(wv.h[j]), bfbits2f(xv.h[j]), acc);
+    }
+    acc = block_reduce_sum(acc, red);
+    const float k = acc * r * r / (float)H;
+    bf16x8* dxr = dx + row * hc;
+    for (int c = threadIdx.x; c < hc; c += BLOCK) {
+      bf16x8 xv = xr[c], wv = w[c], dv = dyr[c], ov;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = r * (bfbits2f(wv.h[j]) * bfbits2f(dv.h[j]) - bfbits2f(xv.h[j]) * k);
+        ov.h[j] = f2bfbits(g);
+      }
+      dxr[c] = ov;
+    }
+    __syncthreads();
+  }
+}
+
+// dw_j = sum_rows dy_j * x_j * r  — column-parallel: blockIdx.x covers
+// columns, blockIdx.y splits rows; fp32 accumulation in registers then one
+// atomicAdd per column per block.
+__global__ void rmsnorm_bwd_dw_kernel(const bf16x8* __restrict__ x,
+                                      const bf16x8* __restrict__ dy,
+                                      const float* __restrict__ invr,
+                                      float* __restrict__ dw,
+                                      long nrows, int hc) {
+  const int c = blockIdx.x * BLOCK + threadIdx.x;  // vec-column index
+  if (c >= hc) return;
+  const long row0 = blockIdx.y;
+  const long rstride = gridDim.y;
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (long row = row0; row < nrows; row += rstride) {
+    const float r = invr[row];
+    bf16x8 xv = x[row * hc + c];
+    bf16x8 dv = dy[row * hc + c];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      acc[j] = fmaf(bfbits2f(dv.h[j]) * r, bfbits2f(xv.h[j]), acc[j]);
+  }
+  float* out = dw + (long)c * 8;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(out + j, acc[j]);
+}
+
+}  // namespace
+
+extern "C" {
+
+hipError_t tok_rmsnorm_fwd(const void* x, const void* w, void* y, float* invr,
+                           long nrows, int H, float eps, hipStream_t stream) {
+  const int hc = H / 8;
+  int grid = (int)(nrows < 8192 ? nrows : 8192);
+  if (grid < 1) grid = 1;
+  rmsnorm_fwd_kernel<<<grid, BLOCK, 0, stream>>>(
+      (const bf16x8*)x, (const bf16x8*)w, (bf16x8*)y, invr, nrows, hc, eps);
+  return hipGetLastError();
+}
+
+hipError_t tok_rmsnorm_bwd(const void* x, const void* w, const void* dy,
+                           const float* invr, void* dx, float* dw_f32,
+                           long nrows, int H, hipStream_t stream) {
+  const int hc = H / 8;
+  int grid = (int)(nrows < 8192 ? nrows : 8192);
+  if (grid < 1) grid = 1;
+  rmsnorm_bwd_dx_kernel<<<grid, BLOCK, 0, stream>>>(
+      (const bf16x8*)x, (const bf16x8*)w, (const bf16x8*)dy, invr, (bf16x8*)dx,
+      nrows, hc);
+  // dw: column blocks x row splits. Aim for ~2048 blocks total.
+  int cblocks = (hc + BLOCK - 1) / BLOCK;
+  int rsplit = 2048 / (cblocks > 0 ? cblocks : 1);
+  if (rsplit < 1) rsplit = 1;
+  if ((long)rsplit > nrows) rsplit = (int)nrows;
+  dim3 g(cblocks, rsplit);
+  rmsnorm_bwd_dw_kernel<<<g, BLOCK, 0, stream>>>(
+      (const bf16x8*)x, (const bf16x8*)dy, invr, dw_f32, nrows, hc);
+  return hipGetLastError();
+}
+}
