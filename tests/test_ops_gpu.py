@@ -1,0 +1,126 @@
+"""GPU numerics tests: gfx950 HIP kernels vs plain PyTorch fp32 references.
+
+Every test here compares the handwritten kernel against the fp32 CPU/eager
+reference of the same op (tolerances account for bf16 I/O).
+"""
+import pytest
+import torch
+
+from torch_on_k8s_amd import ops
+
+pytestmark = pytest.mark.gpu
+
+
+def dev():
+    return torch.device("cuda", 0)
+
+
+def test_hip_ext_loaded():
+    # on a GPU box the native extension must be present - no silent fallback
+    assert ops.hip_ext_available(), "gfx950 extension not built/loaded"
+
+
+def test_mfma_16x16x32_layout():
+    """Transpose-detecting probe: asymmetric A and B (guide §5.4 r16)."""
+    torch.manual_seed(0)
+    A = (torch.randn(16, 32) * 0.5).bfloat16().to(dev())
+    B = (torch.randn(32, 16) * 0.5).bfloat16().to(dev())
+    D = ops._C.mfma_probe(A, B)
+    ref = A.float().cpu() @ B.float().cpu()
+    err = (D.cpu() - ref).abs().max().item()
+    assert err < 1e-2, f"MFMA layout mismatch, max err {err}\nD={D.cpu()}\nref={ref}"
+
+
+@pytest.mark.parametrize("shape", [(32, 256), (1024, 4096), (33, 1024)])
+def test_rmsnorm_fwd_bwd(shape):
+    torch.manual_seed(0)
+    # quantize inputs to bf16 FIRST so the fp32 reference sees the same
+    # values as the kernel (isolates kernel math error from input rounding)
+    x = torch.randn(shape).bfloat16().float()
+    w = torch.randn(shape[-1]).bfloat16().float()
+    dy = torch.randn(shape).bfloat16().float()
+
+    # fp32 CPU reference with autograd
+    xr = x.clone().requires_grad_(True)
+    wr = w.clone().requires_grad_(True)
+    yr = ops.rmsnorm(xr, wr, 1e-5)
+    yr.backward(dy)
+
+    xg = x.bfloat16().to(dev()).requires_grad_(True)
+    wg = w.bfloat16().to(dev()).requires_grad_(True)
+    yg = ops.rmsnorm(xg, wg, 1e-5)
+    yg.backward(dy.bfloat16().to(dev()))
+
+    assert (yg.float().cpu() - yr.detach()).abs().max() < 0.05
+    assert (xg.grad.float().cpu() - xr.grad).abs().max() < 0.05
+    # dw accumulates over rows - compare relative
+    dw_err = (wg.grad.float().cpu() - wr.grad).abs().max()
+    assert dw_err < 0.05 * max(1.0, wr.grad.abs().max().item())
+
+
+def test_rope_fwd_bwd():
+    from torch_on_k8s_amd.models.llama import build_rope_table, get_config
+    torch.manual_seed(0)
+    cfg = get_config("llama-tiny", head_dim=128)
+    B, S, H, D = 2, 64, 4, 128
+    cos, sin = build_rope_table(cfg, S, dev())
+    x = torch.randn(B, S, H, D)
+    dy = torch.randn(B, S, H, D)
+
+    y_ref = ops.rope_ref(x, cos.cpu(), sin.cpu(), 1.0)
+    dx_ref = ops.rope_ref(dy, cos.cpu(), sin.cpu(), -1.0)
+
+    xg = x.bfloat16().to(dev()).requires_grad_(True)
+    yg = ops.apply_rope(xg, cos, sin)
+    yg.backward(dy.bfloat16().to(dev()))
+
+    assert (yg.float().cpu() - y_ref.float()).abs().max() < 0.03
+    assert (xg.grad.float().cpu() - dx_ref.float()).abs().max() < 0.03
+
+
+def test_fused_adamw_vs_reference():
+    torch.manual_seed(0)
+    n = 4096 + 8
+    p0 = torch.randn(n)
+    g = torch.randn(n)
+
+    # CPU fp32 reference (itself verified vs torch.optim.AdamW on CPU)
+    p_ref = p0.clone()
+    m_ref = torch.zeros(n)
+    v_ref = torch.zeros(n)
+    for step in range(1, 5):
+        ops.fused_adamw_(p_ref, g, m_ref, v_ref, lr=1e-2, beta1=0.9,
+                         beta2=0.95, eps=1e-8, weight_decay=0.1, step=step,
+                         grad_scale=0.5)
+
+    pg = p0.bfloat16().to(dev())
+    gg = g.bfloat16().to(dev())
+    mg = torch.zeros(n, device=dev())
+    vg = torch.zeros(n, device=dev())
+    for step in range(1, 5):
+        ops.fused_adamw_(pg, gg, mg, vg, lr=1e-2, beta1=0.9, beta2=0.95,
+                         eps=1e-8, weight_decay=0.1, step=step, grad_scale=0.5)
+
+    assert (pg.float().cpu() - p_ref).abs().max() < 0.05
+    assert (mg.cpu() - m_ref).abs().max() < 0.02
+    assert (vg.cpu() - v_ref).abs().max() < 0.02
+
+
+def test_trainer_single_gpu_step():
+    """End-to-end: tiny Llama fwd+bwd+fused AdamW on GPU, loss finite and
+    decreasing on a fixed batch."""
+    from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
+    from torch_on_k8s_amd.parallel.env import DistContext
+    torch.manual_seed(0)
+    ctx = DistContext(device=dev())
+    cfg = TrainerConfig(model="llama-tiny", micro_batch=2, seq_len=128,
+                        lr=1e-3)  # default attn_impl="hip": the fused
+    # flash-attention kernel is the path under test
+    tr = Trainer(cfg, ctx)
+    losses = []
+    for _ in range(6):
+        tr.step_count = 0
+        losses.append(tr.train_step())
+        tr.step_count = 1
+    assert all(l == l for l in losses)
+    assert losses[-1] < losses[0], losses

@@ -19,6 +19,16 @@ hipError_t tok_rope(const void* x, void* y, const float* cos_tab,
 hipError_t tok_adamw(void* p, const void* g, float* m, float* v, long n,
                      float lr, float beta1, float beta2, float eps, float wd,
                      int step, float gscale, hipStream_t stream);
+hipError_t tok_mfma_probe_16x16x32(const void* A, const void* B, float* D,
+                                   hipStream_t stream);
+hipError_t tok_attn_fwd(const void* q, const void* k, const void* v, void* o,
+                        float* lse, int B, int S, int Hq, int Hkv, int D,
+                        int causal, hipStream_t stream);
+hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
+                        const void* o, const void* dout, const float* lse,
+                        float* dsum_ws, void* dq, void* dk, void* dv, int B,
+                        int S, int Hq, int Hkv, int D, int causal,
+                        hipStream_t stream);
 }
 
 namespace {
@@ -103,9 +113,61 @@ void adamw_(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double lr,
                        (float)gscale, current_stream()));
 }
 
+// q: [B,S,Hq,D], k/v: [B,S,Hkv,D] bf16 contiguous -> (o, lse[B,Hq,S] f32)
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 bool causal) {
+  CHECK_BF16_CUDA(q);
+  CHECK_BF16_CUDA(k);
+  CHECK_BF16_CUDA(v);
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(D == 128 || D == 64, "head_dim must be 64 or 128");
+  TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
+  TORCH_CHECK(k.sizes() == v.sizes() && k.size(0) == B && k.size(1) == S);
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  TOK_HIP_OK(tok_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                          o.data_ptr(), lse.data_ptr<float>(), B, S, Hq, Hkv,
+                          D, causal ? 1 : 0, current_stream()));
+  return {o, lse};
+}
+
+std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor o, at::Tensor lse, at::Tensor dout,
+                                 bool causal) {
+  CHECK_BF16_CUDA(q);
+  CHECK_BF16_CUDA(dout);
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto dsum = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  TOK_HIP_OK(tok_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                          o.data_ptr(), dout.data_ptr(),
+                          lse.data_ptr<float>(), dsum.data_ptr<float>(),
+                          dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), B, S,
+                          Hq, Hkv, D, causal ? 1 : 0, current_stream()));
+  return {dq, dk, dv};
+}
+
+at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
+  CHECK_BF16_CUDA(A);
+  CHECK_BF16_CUDA(B);
+  TORCH_CHECK(A.sizes() == at::IntArrayRef({16, 32}) &&
+              B.sizes() == at::IntArrayRef({32, 16}));
+  auto D = at::empty({16, 16}, A.options().dtype(at::kFloat));
+  TOK_HIP_OK(tok_mfma_probe_16x16x32(A.data_ptr(), B.data_ptr(),
+                                     D.data_ptr<float>(), current_stream()));
+  return D;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 bf16 layout probe");
+  mod.def("attn_fwd", &attn_fwd, "Flash attention forward (bf16, gfx950)");
+  mod.def("attn_bwd", &attn_bwd, "Flash attention backward (bf16, gfx950)");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16, gfx950)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (bf16, gfx950)");
   mod.def("rope", &rope, "Rotary embedding rotate-half (bf16, gfx950)");
