@@ -1,0 +1,377 @@
+"""Control-plane tests with FakeRuntime (the envtest analog: tests flip
+task phases by hand — SURVEY.md §4)."""
+import time
+
+import pytest
+
+from torch_on_k8s_amd.controlplane.api import (DAGCondition, ElasticPolicy,
+                                               JobConditionType, RestartPolicy,
+                                               RunPolicy, SchedulingPolicy,
+                                               TaskPhase, TaskSpec, TaskType,
+                                               TorchJob, set_defaults,
+                                               ANN_CKPT_REQUESTED,
+                                               ANN_CKPT_COMPLETED)
+from torch_on_k8s_amd.controlplane.controller import (ControllerConfig,
+                                                      JobController)
+from torch_on_k8s_amd.controlplane.coordinator import Coordinator
+from torch_on_k8s_amd.controlplane.elastic import (ElasticScaler,
+                                                   TorchElasticAutoscaler,
+                                                   Observation)
+from torch_on_k8s_amd.controlplane import failover as fo
+from torch_on_k8s_amd.controlplane.modelregistry import (ModelRegistry,
+                                                         StorageProvider)
+from torch_on_k8s_amd.controlplane.node import NodeState
+from torch_on_k8s_amd.controlplane.runtime import FakeRuntime
+
+
+def mk_job(name="job1", workers=2, master=True, aimaster=False, **kw):
+    tasks = {}
+    if aimaster:
+        tasks[TaskType.AIMASTER] = TaskSpec(replicas=1, gpus_per_task=0)
+    if master:
+        tasks[TaskType.MASTER] = TaskSpec(replicas=1)
+    if workers:
+        tasks[TaskType.WORKER] = TaskSpec(replicas=workers)
+    return TorchJob(name=name, tasks=tasks, **kw)
+
+
+def mk_ctl(num_gpus=8, coordinator=None, **cfg_kw):
+    node = NodeState(num_gpus=num_gpus)
+    rt = FakeRuntime()
+    ctl = JobController(node, rt, ControllerConfig(**cfg_kw),
+                        coordinator=coordinator)
+    return ctl, node, rt
+
+
+# ---------------------------------------------------------------------------
+# defaults
+# ---------------------------------------------------------------------------
+def test_defaults_dag_edges_and_policies():
+    job = set_defaults(mk_job("My_Job", aimaster=True))
+    assert job.name == "my-job"
+    assert job.tasks[TaskType.MASTER].restart_policy == RestartPolicy.ON_EXIT_CODE
+    assert job.tasks[TaskType.WORKER].restart_policy == RestartPolicy.ON_FAILURE
+    assert job.tasks[TaskType.MASTER].dag_conditions[0].upstream == TaskType.AIMASTER
+    assert job.tasks[TaskType.WORKER].dag_conditions[0].upstream == TaskType.MASTER
+    # min_members populated from task specs (reference bug fixed)
+    assert job.min_members[TaskType.WORKER] == 2
+
+
+# ---------------------------------------------------------------------------
+# DAG gating
+# ---------------------------------------------------------------------------
+def test_dag_workers_wait_for_master():
+    ctl, node, rt = mk_ctl()
+    job = ctl.create_job(mk_job())
+    # FakeRuntime marks tasks Running on poll, but at creation time the
+    # first reconcile sees master Pending -> workers must NOT start yet
+    started_types = [k[1] for k in rt.started]
+    assert TaskType.MASTER in started_types
+    assert TaskType.WORKER not in started_types
+    # next reconcile: master polled Running -> workers start
+    ctl.reconcile(job)
+    started_types = [k[1] for k in rt.started]
+    assert TaskType.WORKER in started_types
+
+
+def test_dag_off_starts_all():
+    ctl, node, rt = mk_ctl(enable_dag_scheduling=False)
+    ctl.create_job(mk_job())
+    types = [k[1] for k in rt.started]
+    assert TaskType.MASTER in types and TaskType.WORKER in types
+
+
+# ---------------------------------------------------------------------------
+# gang scheduling
+# ---------------------------------------------------------------------------
+def test_gang_blocks_until_enough_gpus():
+    ctl, node, rt = mk_ctl(num_gpus=2)
+    job = ctl.create_job(mk_job(workers=4))  # needs 5 GPUs total
+    assert not rt.started  # gang refused
+    assert any(e.reason == "GangNotAdmitted" for e in ctl.events)
+
+
+def test_gang_min_member_running_rule():
+    """Gang job becomes Running at MinMember running tasks (README fix)."""
+    ctl, node, rt = mk_ctl(num_gpus=8)
+    job = mk_job(workers=4, master=False)
+    job.scheduling = SchedulingPolicy(min_available=2)
+    job.min_members = {TaskType.WORKER: 2}
+    ctl.create_job(job)
+    ctl.reconcile(job)  # workers running via FakeRuntime auto_run
+    assert job.status.phase == JobConditionType.RUNNING
+
+
+def test_gpu_slots_allocated_and_released():
+    ctl, node, rt = mk_ctl()
+    job = ctl.create_job(mk_job(workers=2))
+    ctl.reconcile(job)  # master running -> workers start
+    ctl.reconcile(job)
+    assert len(node.free_slots) == 8 - 3
+    for key in list(rt.tasks):
+        rt.set_phase(key, TaskPhase.SUCCEEDED, exit_code=0)
+    ctl.reconcile(job)
+    assert job.status.phase == JobConditionType.SUCCEEDED
+    assert len(node.free_slots) == 8
+
+
+# ---------------------------------------------------------------------------
+# failover / exit codes
+# ---------------------------------------------------------------------------
+def test_exit_code_policy():
+    assert fo.exit_code_retryable(137)
+    assert fo.exit_code_retryable(143)
+    assert fo.exit_code_retryable(138)   # SIGUSR1 user-defined
+    assert fo.exit_code_retryable(None, "OOMKilled")
+    assert not fo.exit_code_retryable(1)
+    assert not fo.exit_code_retryable(139)
+    assert not fo.exit_code_retryable(126)
+
+
+def test_master_retryable_failure_restarts():
+    ctl, node, rt = mk_ctl()
+    job = ctl.create_job(mk_job(workers=0))
+    master_key = ("job1", TaskType.MASTER, 0)
+    rt.set_phase(master_key, TaskPhase.FAILED, exit_code=137, reason="Killed")
+    ctl.reconcile(job)
+    assert job.status.restart_count == 1
+    # restarted: a fresh handle exists and is running after next poll
+    ctl.reconcile(job)
+    assert ctl.handles["job1"][master_key].phase == TaskPhase.RUNNING
+    assert job.status.has_condition(JobConditionType.RESTARTING)
+
+
+def test_master_permanent_failure_fails_job():
+    ctl, node, rt = mk_ctl()
+    job = ctl.create_job(mk_job(workers=0))
+    rt.set_phase(("job1", TaskType.MASTER, 0), TaskPhase.FAILED, exit_code=1)
+    ctl.reconcile(job)
+    assert job.status.phase == JobConditionType.FAILED
+
+
+def test_backoff_limit():
+    ctl, node, rt = mk_ctl()
+    job = mk_job(workers=0, run_policy=RunPolicy(backoff_limit=2))
+    ctl.create_job(job)
+    for i in range(4):
+        key = ("job1", TaskType.MASTER, 0)
+        if key in rt.tasks and not rt.tasks[key].finished:
+            rt.set_phase(key, TaskPhase.FAILED, exit_code=137)
+        ctl.reconcile(job)
+        if job.status.phase == JobConditionType.FAILED:
+            break
+    assert job.status.phase == JobConditionType.FAILED
+    assert job.status.restart_count == 2
+
+
+def test_active_deadline():
+    ctl, node, rt = mk_ctl()
+    job = mk_job(workers=0,
+                 run_policy=RunPolicy(active_deadline_seconds=0.01))
+    ctl.create_job(job)
+    ctl.reconcile(job)  # running; start_time set
+    time.sleep(0.05)
+    ctl.reconcile(job)
+    assert job.status.phase == JobConditionType.FAILED
+
+
+# ---------------------------------------------------------------------------
+# coordinator
+# ---------------------------------------------------------------------------
+def test_coordinator_quota_and_wrr():
+    ctl, node, rt = mk_ctl()
+    coord = Coordinator(dequeue_fn=lambda j: ctl.reconcile(j),
+                        tenant_usage_fn=ctl.tenant_gpu_usage,
+                        quotas={"teamA": 8, "teamB": 4}, default_quota=8)
+    ctl.coordinator = coord
+
+    jobA = mk_job("job-a", workers=5,
+                  scheduling=SchedulingPolicy(queue="teamA"))
+    jobB = mk_job("job-b", workers=5,
+                  scheduling=SchedulingPolicy(queue="teamB"))
+    ctl.create_job(jobA)
+    ctl.create_job(jobB)
+    assert coord.is_queuing(jobA.uid) and coord.is_queuing(jobB.uid)
+
+    # schedule passes: jobA (6 GPUs <= 8) admits; jobB (6 > 4) never does
+    for _ in range(8):
+        coord.schedule_once()
+    assert not coord.is_queuing(jobA.uid)
+    assert coord.is_queuing(jobB.uid)
+    # jobA proceeded into reconcile (master started)
+    assert ("job-a", TaskType.MASTER, 0) in rt.tasks
+
+
+def test_coordinator_priority_selects_higher():
+    admitted = []
+    coord = Coordinator(dequeue_fn=lambda j: admitted.append(j.name),
+                        default_quota=100)
+    lo = mk_job("lo", scheduling=SchedulingPolicy(queue="q", priority=1))
+    hi = mk_job("hi", scheduling=SchedulingPolicy(queue="q", priority=9))
+    set_defaults(lo)
+    set_defaults(hi)
+    coord.enqueue_or_update(lo)
+    coord.enqueue_or_update(hi)
+    coord.schedule_once()
+    assert admitted == ["hi"]
+
+
+def test_wrr_weights_favor_bigger_queue():
+    from torch_on_k8s_amd.controlplane.coordinator import (
+        Queue, QueueUnit, WeightedRoundRobinSelector)
+    qa, qb = Queue("a"), Queue("b")
+    for i in range(3):
+        j = set_defaults(mk_job(f"a{i}", workers=3,
+                                scheduling=SchedulingPolicy(queue="a")))
+        qa.add(QueueUnit(j))
+    j = set_defaults(mk_job("b0", workers=0,
+                            scheduling=SchedulingPolicy(queue="b")))
+    qb.add(QueueUnit(j))
+    sel = WeightedRoundRobinSelector()
+    picks = [sel.next([qa, qb]).tenant for _ in range(13)]
+    # qa weight 12, qb weight 1 -> qa picked much more often
+    assert picks.count("a") > picks.count("b")
+    assert "b" in picks  # but b is not starved
+
+
+# ---------------------------------------------------------------------------
+# elastic
+# ---------------------------------------------------------------------------
+def test_elastic_checkpoint_protocol_and_scale_out():
+    ctl, node, rt = mk_ctl()
+    ctl.elastic = ElasticScaler()
+    job = mk_job(workers=2, elastic=ElasticPolicy(min_replicas=2,
+                                                  max_replicas=4))
+    ctl.create_job(job)
+    ctl.reconcile(job)
+    ctl.reconcile(job)
+    assert sum(1 for k in rt.tasks if k[1] == TaskType.WORKER) == 2
+
+    # scale out 2 -> 4 (generation bump)
+    ElasticScaler.scale(job, 4)
+    ctl.reconcile(job)
+    # stage 1: checkpoint requested, no new workers yet
+    assert ANN_CKPT_REQUESTED in job.annotations
+    live_workers = [h for h in ctl.handles["job1"].values()
+                    if h.task_type == TaskType.WORKER]
+    assert len(live_workers) == 2
+
+    # data plane completes the checkpoint
+    ElasticScaler.complete_checkpoint(job)
+    assert ANN_CKPT_COMPLETED in job.annotations
+    ctl.reconcile(job)   # executes the scale: kills stale, recreates
+    ctl.reconcile(job)   # recreate remaining
+    workers = [h for h in ctl.handles["job1"].values()
+               if h.task_type == TaskType.WORKER]
+    assert len(workers) == 4
+    assert all(h.generation == job.generation for h in workers)
+    # WORLD_SIZE refreshed for the new generation (master + 4 workers)
+    env = rt.tasks[("job1", TaskType.WORKER, 3)].env
+    assert env["WORLD_SIZE"] == "5"
+
+
+def test_autoscaler_doubles_then_reverts():
+    job = mk_job(workers=2, elastic=ElasticPolicy(min_replicas=2,
+                                                  max_replicas=8,
+                                                  metric_window=2))
+    set_defaults(job)
+    feed = {"obs": []}
+    auto = TorchElasticAutoscaler(lambda j: feed["obs"].pop(0)
+                                  if feed["obs"] else None)
+    # 2 replicas: latency 1.0
+    feed["obs"] = [Observation(1, 1.0, 100, 5.0),
+                   Observation(2, 1.0, 100, 5.0)]
+    auto.observe(job)
+    auto.observe(job)
+    act = auto.decide(job)
+    assert act is not None and job.tasks[TaskType.WORKER].replicas == 4
+    gen_after_scale = job.generation
+
+    # 4 replicas: latency 0.4 (per-replica improves: 0.4*4 < 1.0*2) -> 8
+    feed["obs"] = [Observation(3, 0.4, 250, 5.0),
+                   Observation(4, 0.4, 250, 5.0)]
+    auto.observe(job)
+    auto.observe(job)
+    act = auto.decide(job)
+    assert job.tasks[TaskType.WORKER].replicas == 8
+
+    # 8 replicas: latency 0.39 (0.39*8 > 0.4*4) -> revert to 4, stop
+    feed["obs"] = [Observation(5, 0.39, 260, 5.0),
+                   Observation(6, 0.39, 260, 5.0)]
+    auto.observe(job)
+    auto.observe(job)
+    act = auto.decide(job)
+    from torch_on_k8s_amd.controlplane.api import ElasticCondition
+    assert act == ElasticCondition.STOP
+    assert job.tasks[TaskType.WORKER].replicas == 4
+    assert job.generation > gen_after_scale
+
+
+# ---------------------------------------------------------------------------
+# model registry
+# ---------------------------------------------------------------------------
+def test_model_version_build_and_extract(tmp_path):
+    store = StorageProvider(str(tmp_path / "store"))
+    reg = ModelRegistry(store)
+    out = store.job_output_dir("jobx")
+    with open(f"{out}/weights.bin", "wb") as f:
+        f.write(b"\x01\x02\x03" * 100)
+    mv = reg.build_version("llama", "v1", out, source_job="jobx")
+    assert mv.build_phase == "Succeeded"
+    assert mv.digest.startswith("sha256:")
+    assert reg.models["llama"].latest_version == "v1"
+    dest = reg.extract("llama", "v1", str(tmp_path / "run"))
+    import os
+    assert os.path.exists(os.path.join(dest, "outputs", "jobx",
+                                       "weights.bin")) or \
+        any("weights.bin" in f for _, _, fs in os.walk(dest) for f in fs)
+
+
+def test_model_version_created_on_job_success(tmp_path):
+    store = StorageProvider(str(tmp_path / "store"))
+    reg = ModelRegistry(store)
+    ctl, node, rt = mk_ctl()
+    ctl.model_registry = reg
+    job = mk_job(workers=0, model_name="llama")
+    ctl.create_job(job)
+    out = store.job_output_dir(job.name)
+    with open(f"{out}/ck.pt", "wb") as f:
+        f.write(b"x" * 10)
+    rt.set_phase(("job1", TaskType.MASTER, 0), TaskPhase.SUCCEEDED, 0)
+    ctl.reconcile(job)
+    assert job.status.model_version is not None
+    assert reg.models["llama"].latest_version == job.status.model_version
+
+
+# ---------------------------------------------------------------------------
+# two contending jobs end-to-end (BASELINE config 5 shape, CPU-only)
+# ---------------------------------------------------------------------------
+def test_two_jobs_contending_for_node():
+    ctl, node, rt = mk_ctl(num_gpus=8)
+    coord = Coordinator(dequeue_fn=lambda j: ctl.reconcile(j),
+                        tenant_usage_fn=ctl.tenant_gpu_usage,
+                        quotas={"qa": 8, "qb": 8})
+    ctl.coordinator = coord
+    j1 = mk_job("j1", workers=7, scheduling=SchedulingPolicy(queue="qa"))
+    j2 = mk_job("j2", workers=7, scheduling=SchedulingPolicy(queue="qb"))
+    ctl.create_job(j1)
+    ctl.create_job(j2)
+    for _ in range(6):
+        coord.schedule_once()
+        ctl.reconcile_all()
+    # exactly one job holds the node (8 GPUs each; only one fits)
+    phases = {j1.status.phase, j2.status.phase}
+    assert JobConditionType.RUNNING in phases
+    running = j1 if j1.status.phase == JobConditionType.RUNNING else j2
+    waiting = j2 if running is j1 else j1
+    # the other is gang-blocked (created but no tasks)
+    assert not ctl.handles[waiting.name]
+    # finish the running job -> the waiting one admits on later passes
+    for key in list(rt.tasks):
+        if key[0] == running.name:
+            rt.set_phase(key, TaskPhase.SUCCEEDED, 0)
+    ctl.reconcile(running)
+    assert running.status.phase == JobConditionType.SUCCEEDED
+    for _ in range(4):
+        ctl.reconcile_all()
+    assert ctl.handles[waiting.name]  # tasks now created

@@ -1,0 +1,382 @@
+"""TorchJob controller: the reconcile engine.
+
+Node-native rebuild of the reference's JobController + TorchJobReconciler
+(controllers/common/job.go:55-342, controllers/train/torchjob_controller.go,
+controllers/train/job.go:99-207):
+
+  reconcile(job):
+    deletion cleanup -> finished/TTL -> coordinator queue gate ->
+    poll tasks -> termination checks (backoff limit / active deadline) ->
+    elastic protocol hooks -> gang admission -> per-task-type loop in
+    AIMaster->Master->Worker order with DAG gating -> failover of failed
+    tasks by exit-code policy -> job status state machine -> model
+    packaging on success -> metrics.
+
+Differences by design (MI355X node, not a cluster): tasks are local GPU
+processes (runtime.py); "services" (the reference's master headless svc,
+service.go:251-308) collapse to a per-job master port reservation —
+reservation/rewrite semantics live in _master_port.
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+
+from torch_on_k8s_amd.controlplane import failover as fo
+from torch_on_k8s_amd.controlplane.api import (CleanPodPolicy,
+                                               JobConditionType, RestartPolicy,
+                                               TASK_ORDER, TaskPhase, TaskType,
+                                               TorchJob, set_defaults)
+from torch_on_k8s_amd.controlplane.dag import dag_condition_ready
+from torch_on_k8s_amd.controlplane.gang import GangScheduler
+from torch_on_k8s_amd.controlplane.node import NodeState
+from torch_on_k8s_amd.controlplane.runtime import Runtime, TaskHandle
+
+
+@dataclass
+class Event:
+    job: str
+    type: str         # Normal | Warning
+    reason: str
+    message: str
+    ts: float = field(default_factory=time.time)
+
+
+@dataclass
+class ControllerConfig:
+    """JobControllerConfiguration analog (controllers/common/config.go)."""
+    enable_gang_scheduling: bool = True
+    enable_dag_scheduling: bool = True
+    master_port_range: tuple = (20000, 30000)  # hostnetwork range analog
+
+
+class JobController:
+    def __init__(self, node: NodeState, runtime: Runtime,
+                 cfg: ControllerConfig | None = None, coordinator=None,
+                 metrics=None, model_registry=None, elastic=None):
+        self.node = node
+        self.runtime = runtime
+        self.cfg = cfg or ControllerConfig()
+        self.coordinator = coordinator
+        self.metrics = metrics
+        self.model_registry = model_registry
+        self.elastic = elastic
+        self.gang = GangScheduler(node, self.cfg.enable_dag_scheduling) \
+            if self.cfg.enable_gang_scheduling else None
+        self.jobs: dict[str, TorchJob] = {}
+        self.handles: dict[str, dict] = {}   # job -> {key: TaskHandle}
+        self.events: list[Event] = []
+        self._ports: dict[str, int] = {}
+        self._next_port = self.cfg.master_port_range[0]
+        self._first_task_ts: dict[str, float] = {}
+
+    # ------------------------------------------------------------------
+    def event(self, job: str, etype: str, reason: str, msg: str = ""):
+        self.events.append(Event(job, etype, reason, msg))
+
+    def create_job(self, job: TorchJob) -> TorchJob:
+        """OnOwnerCreate analog (eventhandler.go:38-64): default, mark
+        Created, enqueue to the coordinator (or reconcile directly)."""
+        set_defaults(job)
+        self.jobs[job.name] = job
+        self.handles.setdefault(job.name, {})
+        job.status.set_condition(JobConditionType.CREATED, "JobCreated")
+        if self.metrics:
+            self.metrics.created()
+        if self.coordinator is not None:
+            self.coordinator.enqueue_or_update(job)
+        else:
+            self.reconcile(job)
+        return job
+
+    def delete_job(self, name: str):
+        job = self.jobs.get(name)
+        if job is None:
+            return
+        job.deleted = True
+        self.reconcile(job)
+        if self.metrics:
+            self.metrics.deleted()
+
+    def _master_port(self, job: TorchJob) -> int:
+        """Per-job master port from the configured range (the reference
+        picks random host ports from --hostnetwork-port-range,
+        pod.go:531-544; collisions here are avoided by monotonic sweep)."""
+        if job.name not in self._ports:
+            lo, hi = self.cfg.master_port_range
+            self._ports[job.name] = lo + (self._next_port - lo) % (hi - lo)
+            self._next_port += 1
+        return self._ports[job.name]
+
+    # ------------------------------------------------------------------
+    def reconcile(self, job: TorchJob) -> None:
+        hs = self.handles.setdefault(job.name, {})
+
+        if job.deleted:
+            self._cleanup(job, kill_all=True)
+            self.jobs.pop(job.name, None)
+            self.handles.pop(job.name, None)
+            if self.coordinator is not None:
+                self.coordinator.dequeue(job.uid)
+            return
+
+        phase = job.status.phase
+        if phase in (JobConditionType.SUCCEEDED, JobConditionType.FAILED):
+            self._maybe_ttl_cleanup(job)
+            return
+
+        # coordinator queue gate (torchjob_controller.go:190-197 +
+        # eventhandler: while queuing the controller does not act)
+        if self.coordinator is not None and self.coordinator.is_queuing(job.uid):
+            return
+
+        # refresh task phases
+        for h in hs.values():
+            self.runtime.poll(h)
+
+        # release GPU slots of finished tasks
+        for h in hs.values():
+            if h.finished and h.gpu_slots:
+                self.node.release(h.gpu_slots)
+                h.gpu_slots = ()
+
+        # termination checks (job.go:105-200)
+        if self._check_terminated(job, hs):
+            return
+
+        # elastic protocol (checkpoint gating + generation scale,
+        # job.go:221-248)
+        if self.elastic is not None and self.elastic.enabled(job):
+            if self.elastic.reconcile(self, job, hs):
+                return  # elastic transaction in flight; hold other actions
+
+        # gang admission before any task creation (job.go:214-219)
+        if self.gang is not None:
+            self.gang.create_pod_group(job)
+            if not hs and not self.gang.can_admit(job):
+                self.event(job.name, "Normal", "GangNotAdmitted",
+                           "waiting for gang quota")
+                return
+
+        # per-task-type reconcile in AIMaster -> Master -> Worker order
+        for t in TASK_ORDER:
+            if t not in job.tasks:
+                continue
+            if self.cfg.enable_dag_scheduling and \
+                    not dag_condition_ready(job, t, hs):
+                break  # downstream tasks wait too
+            self._reconcile_task(job, t, hs)
+
+        self._update_status(job, hs)
+
+    # ------------------------------------------------------------------
+    def _reconcile_task(self, job: TorchJob, t: TaskType, hs: dict):
+        spec = job.tasks[t]
+        desired = spec.replicas
+        # create missing indices (pod.go:361-464 index-sliced semantics)
+        for idx in range(desired):
+            key = (job.name, t, idx)
+            h = hs.get(key)
+            if h is None:
+                self._start_task(job, t, idx, hs)
+                continue
+            if h.finished:
+                self._handle_finished(job, t, h, hs)
+        # remove excess indices (scale-in)
+        for key, h in list(hs.items()):
+            if h.task_type == t and h.index >= desired:
+                self.runtime.kill(h)
+                if h.gpu_slots:
+                    self.node.release(h.gpu_slots)
+                hs.pop(key, None)
+
+    def _start_task(self, job: TorchJob, t: TaskType, idx: int, hs: dict):
+        spec = job.tasks[t]
+        need = spec.gpus_per_task if t != TaskType.AIMASTER else 0
+        if need and len(self.node.free_slots) < need:
+            self.event(job.name, "Warning", "NoGPUSlots",
+                       f"{t.value}-{idx} waiting for {need} GPUs")
+            return
+        slots = self.node.allocate(need, (job.name, t.value, idx)) \
+            if need else ()
+        extra_env = {"MASTER_PORT": str(self._master_port(job))}
+        h = self.runtime.start_task(job, t, idx, slots, extra_env)
+        hs[h.key] = h
+        if job.name not in self._first_task_ts:
+            self._first_task_ts[job.name] = time.time()
+            if self.metrics:
+                self.metrics.first_task_delay(job.name)
+        self.event(job.name, "Normal", "TaskCreated",
+                   f"{t.value}-{idx} on GPUs {slots}")
+
+    def _handle_finished(self, job: TorchJob, t: TaskType, h: TaskHandle,
+                         hs: dict):
+        spec = job.tasks[t]
+        if h.phase == TaskPhase.SUCCEEDED:
+            if spec.restart_policy == RestartPolicy.ALWAYS:
+                self._restart_task(job, h, hs)
+            return
+        # failed:
+        policy = spec.restart_policy
+        retry = (policy == RestartPolicy.ALWAYS or
+                 policy == RestartPolicy.ON_FAILURE or
+                 (policy == RestartPolicy.ON_EXIT_CODE and
+                  fo.exit_code_retryable(h.exit_code, h.reason)))
+        if retry and job.status.restart_count < job.run_policy.backoff_limit:
+            job.status.restart_count += 1
+            job.status.set_condition(JobConditionType.RESTARTING,
+                                     "TaskFailed",
+                                     f"{t.value}-{h.index} exit {h.exit_code}")
+            if self.metrics:
+                self.metrics.restarted()
+            self._restart_task(job, h, hs)
+
+    def _restart_task(self, job: TorchJob, h: TaskHandle, hs: dict):
+        """Recreate failover (failover.go:117-127): drop the handle; the
+        next reconcile pass recreates the missing index."""
+        self.runtime.kill(h)
+        if h.gpu_slots:
+            self.node.release(h.gpu_slots)
+        rc = h.restart_count + 1
+        hs.pop(h.key, None)
+        self.event(job.name, "Normal", "TaskRestarting",
+                   f"{h.task_type.value}-{h.index} restart #{rc}")
+        self._start_task(job, h.task_type, h.index, hs)
+        nh = hs.get(h.key)
+        if nh:
+            nh.restart_count = rc
+
+    # ------------------------------------------------------------------
+    def _check_terminated(self, job: TorchJob, hs: dict) -> bool:
+        rp = job.run_policy
+        failed = [h for h in hs.values() if h.phase == TaskPhase.FAILED]
+        # permanent (non-retryable) task failure fails the job
+        for h in failed:
+            if not self._would_retry(job, h):
+                self._fail_job(job, hs, "PermanentTaskFailure",
+                               f"task {h.key} exit {h.exit_code}")
+                return True
+        # backoff limit exhausted (job.go:115-135): a retryable failure
+        # with no restart budget left fails the job
+        if failed and job.status.restart_count >= rp.backoff_limit:
+            self._fail_job(job, hs, "BackoffLimitExceeded",
+                           f"restarts {job.status.restart_count}")
+            return True
+        if rp.active_deadline_seconds is not None and \
+                job.status.start_time is not None and \
+                time.time() - job.status.start_time > rp.active_deadline_seconds:
+            self._fail_job(job, hs, "DeadlineExceeded", "")
+            return True
+        return False
+
+    def _would_retry(self, job: TorchJob, h: TaskHandle) -> bool:
+        policy = job.tasks[h.task_type].restart_policy
+        if policy in (RestartPolicy.ALWAYS, RestartPolicy.ON_FAILURE):
+            return True
+        if policy == RestartPolicy.ON_EXIT_CODE:
+            return fo.exit_code_retryable(h.exit_code, h.reason)
+        return False
+
+    def _fail_job(self, job: TorchJob, hs: dict, reason: str, msg: str):
+        job.status.set_condition(JobConditionType.FAILED, reason, msg)
+        job.status.completion_time = time.time()
+        self._cleanup(job, kill_all=True)
+        if self.metrics:
+            self.metrics.failed()
+        self.event(job.name, "Warning", reason, msg)
+
+    # ------------------------------------------------------------------
+    def _update_status(self, job: TorchJob, hs: dict):
+        """State machine parity with train/job.go:99-207."""
+        from torch_on_k8s_amd.controlplane.api import TaskStatus
+        job.status.tasks = {}
+        for t in job.tasks:
+            st = TaskStatus()
+            for h in hs.values():
+                if h.task_type != t:
+                    continue
+                if h.phase == TaskPhase.RUNNING:
+                    st.active += 1
+                elif h.phase == TaskPhase.SUCCEEDED:
+                    st.succeeded += 1
+                elif h.phase == TaskPhase.FAILED:
+                    st.failed += 1
+            job.status.tasks[t] = st
+
+        master = job.status.tasks.get(TaskType.MASTER)
+        workers = job.status.tasks.get(TaskType.WORKER)
+        aim = job.status.tasks.get(TaskType.AIMASTER)
+
+        # Running detection: master (or aimaster) active; gang jobs
+        # become Running at MinMember running tasks
+        active_total = sum(s.active for s in job.status.tasks.values())
+        running = (master and master.active) or \
+                  (master is None and aim and aim.active) or \
+                  (master is None and workers and workers.active)
+        if self.gang is not None and not running:
+            running = self.gang.min_member_running(job, active_total)
+        if running and job.status.phase != JobConditionType.RUNNING:
+            job.status.set_condition(JobConditionType.RUNNING, "JobRunning")
+            if job.status.start_time is None:
+                job.status.start_time = time.time()
+            if self.metrics:
+                self.metrics.all_tasks_delay(job.name)
+
+        # Succeeded: master done (+ all workers done if workers exist)
+        done = False
+        if master is not None:
+            done = master.succeeded >= 1
+            if done and workers is not None:
+                done = workers.succeeded >= job.tasks[TaskType.WORKER].replicas
+        elif workers is not None:
+            done = workers.succeeded >= job.tasks[TaskType.WORKER].replicas
+        if done:
+            job.status.set_condition(JobConditionType.SUCCEEDED, "JobSucceeded")
+            job.status.completion_time = time.time()
+            if self.metrics:
+                self.metrics.succeeded()
+            self._on_success(job, hs)
+
+    def _on_success(self, job: TorchJob, hs: dict):
+        # model packaging (job.go:462-508): output dir -> ModelVersion
+        if self.model_registry is not None and job.model_name:
+            mv = self.model_registry.create_version_for_job(job)
+            job.status.model_version = mv.version if mv else None
+        self._cleanup(job, kill_all=(job.run_policy.clean_task_policy
+                                     != CleanPodPolicy.NONE))
+        if self.gang is not None:
+            self.gang.delete_pod_group(job.name)
+
+    def _cleanup(self, job: TorchJob, kill_all: bool):
+        hs = self.handles.get(job.name, {})
+        for h in list(hs.values()):
+            if kill_all and not h.finished:
+                self.runtime.kill(h)
+            if h.gpu_slots:
+                self.node.release(h.gpu_slots)
+                h.gpu_slots = ()
+        self.node.release_owner(None) if False else None
+        if self.gang is not None and job.deleted:
+            self.gang.delete_pod_group(job.name)
+
+    def _maybe_ttl_cleanup(self, job: TorchJob):
+        ttl = job.run_policy.ttl_seconds_after_finished
+        if ttl is None or job.status.completion_time is None:
+            return
+        if time.time() - job.status.completion_time >= ttl:
+            job.deleted = True
+            self.reconcile(job)
+
+    # ------------------------------------------------------------------
+    def reconcile_all(self):
+        for job in list(self.jobs.values()):
+            self.reconcile(job)
+
+    def tenant_gpu_usage(self, tenant: str) -> int:
+        used = 0
+        for job in self.jobs.values():
+            if (job.scheduling.queue or job.namespace) != tenant:
+                continue
+            for h in self.handles.get(job.name, {}).values():
+                used += len(h.gpu_slots)
+        return used

@@ -1,0 +1,189 @@
+"""Task runtimes: where the reference creates pods via the k8s API
+(controllers/common/pod.go:503-637), this framework starts one local
+process per task on the 8xMI355X node, with GPU slots assigned through
+HIP_VISIBLE_DEVICES and the torch env contract injected
+(SetClusterSpec parity, torchjob_controller.go:314-449).
+
+FakeRuntime is the envtest analog: tests flip task phases
+by hand, no process ever runs.
+"""
+from __future__ import annotations
+
+import os
+import signal
+import subprocess
+import sys
+import time
+from dataclasses import dataclass, field
+
+from torch_on_k8s_amd.controlplane.api import (TaskPhase, TaskType, TorchJob,
+                                               DEFAULT_MASTER_PORT)
+
+
+@dataclass
+class TaskHandle:
+    job_name: str
+    task_type: TaskType
+    index: int
+    phase: TaskPhase = TaskPhase.PENDING
+    exit_code: int | None = None
+    reason: str = ""
+    gpu_slots: tuple = ()
+    generation: int = 1
+    annotations: dict = field(default_factory=dict)
+    start_time: float = field(default_factory=time.time)
+    restart_count: int = 0
+    proc: object = None  # subprocess.Popen for the local runtime
+
+    @property
+    def key(self):
+        return (self.job_name, self.task_type, self.index)
+
+    @property
+    def finished(self):
+        return self.phase in (TaskPhase.SUCCEEDED, TaskPhase.FAILED)
+
+
+def task_name(job: str, t: TaskType, index: int) -> str:
+    """'<job>-<tasktype>-<index>' (reference utils.go:75-77)."""
+    return f"{job}-{t.value}-{index}"
+
+
+def cluster_env(job: TorchJob, t: TaskType, index: int,
+                master_port: int | None = None) -> dict:
+    """The torch env contract (SetClusterSpec parity,
+    torchjob_controller.go:332-445): master rank 0, workers index+1;
+    WORLD_SIZE excludes the AIMaster; master addr is localhost on the
+    single-node design (the reference's TorchLocalMasterAddr gate,
+    features.go:31-63, is the natural choice here)."""
+    world = job.total_replicas(include_aimaster=False)
+    if t == TaskType.MASTER:
+        rank = 0
+    elif t == TaskType.WORKER:
+        # +1 iff a master task exists (reference :339-348)
+        rank = index + (1 if TaskType.MASTER in job.tasks else 0)
+    else:
+        rank = -1  # AIMaster is not part of the process group
+    env = {
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(master_port or DEFAULT_MASTER_PORT),
+        "WORLD_SIZE": str(job.annotations.get("world-size", world)),
+        "RANK": str(rank),
+        "PYTHONUNBUFFERED": "1",  # reference torchjob_controller.go:394-445
+        "TOK_JOB_NAME": job.name,
+        "TOK_TASK_TYPE": t.value,
+        "TOK_TASK_INDEX": str(index),
+        "TOK_GENERATION": str(job.generation),
+    }
+    return env
+
+
+class Runtime:
+    """Abstract task runtime."""
+
+    def start_task(self, job: TorchJob, t: TaskType, index: int,
+                   gpu_slots: tuple, extra_env: dict) -> TaskHandle:
+        raise NotImplementedError
+
+    def poll(self, h: TaskHandle) -> TaskHandle:
+        raise NotImplementedError
+
+    def kill(self, h: TaskHandle, grace: bool = True) -> None:
+        raise NotImplementedError
+
+
+class FakeRuntime(Runtime):
+    """envtest analog: the test drives phases."""
+
+    def __init__(self):
+        self.tasks: dict = {}
+        self.started: list = []
+        self.killed: list = []
+        # default behavior: tasks move PENDING->RUNNING on first poll
+        self.auto_run = True
+
+    def start_task(self, job, t, index, gpu_slots, extra_env):
+        h = TaskHandle(job.name, t, index, gpu_slots=tuple(gpu_slots),
+                       generation=job.generation)
+        h.env = dict(cluster_env(job, t, index), **extra_env)
+        self.tasks[h.key] = h
+        self.started.append(h.key)
+        return h
+
+    def poll(self, h):
+        if self.auto_run and h.phase == TaskPhase.PENDING:
+            h.phase = TaskPhase.RUNNING
+        return h
+
+    def kill(self, h, grace=True):
+        self.killed.append(h.key)
+        if not h.finished:
+            h.phase = TaskPhase.FAILED
+            h.exit_code = 137 if not grace else 143
+            h.reason = "Killed"
+
+    # --- test helpers ------------------------------------------------
+    def set_phase(self, key, phase, exit_code=None, reason=""):
+        h = self.tasks[key]
+        h.phase = phase
+        h.exit_code = exit_code
+        h.reason = reason
+
+
+class LocalProcessRuntime(Runtime):
+    """One OS process per task ('kubelet' analog). GPU isolation via
+    HIP_VISIBLE_DEVICES over the node's GPU slots."""
+
+    def __init__(self, workdir: str, python: str | None = None):
+        self.workdir = workdir
+        self.python = python or sys.executable
+        os.makedirs(workdir, exist_ok=True)
+
+    def start_task(self, job, t, index, gpu_slots, extra_env):
+        h = TaskHandle(job.name, t, index, gpu_slots=tuple(gpu_slots),
+                       generation=job.generation)
+        spec = job.tasks[t]
+        env = dict(os.environ)
+        env.update(cluster_env(job, t, index))
+        env.update(spec.env)
+        env.update(extra_env)
+        if gpu_slots:
+            env["HIP_VISIBLE_DEVICES"] = ",".join(str(s) for s in gpu_slots)
+            env["LOCAL_RANK"] = "0"  # each process sees exactly its GPUs
+        logdir = os.path.join(self.workdir, job.name)
+        os.makedirs(logdir, exist_ok=True)
+        argv = spec.command or [
+            self.python, "-m", "torch_on_k8s_amd.entrypoint"]
+        logf = open(os.path.join(
+            logdir, f"{task_name(job.name, t, index)}.log"), "ab")
+        h.proc = subprocess.Popen(argv, env=env, stdout=logf, stderr=logf,
+                                  start_new_session=True)
+        h.phase = TaskPhase.RUNNING
+        return h
+
+    def poll(self, h):
+        if h.proc is None or h.finished:
+            return h
+        rc = h.proc.poll()
+        if rc is not None:
+            h.exit_code = rc
+            h.phase = TaskPhase.SUCCEEDED if rc == 0 else TaskPhase.FAILED
+        return h
+
+    def kill(self, h, grace=True):
+        if h.proc is None or h.proc.poll() is not None:
+            return
+        try:
+            pgid = os.getpgid(h.proc.pid)
+            os.killpg(pgid, signal.SIGTERM if grace else signal.SIGKILL)
+        except ProcessLookupError:
+            pass
+
+    def wait(self, h, timeout=60):
+        if h.proc is not None:
+            try:
+                h.proc.wait(timeout=timeout)
+            except subprocess.TimeoutExpired:
+                self.kill(h, grace=False)
+                h.proc.wait(timeout=10)
+        return self.poll(h)
