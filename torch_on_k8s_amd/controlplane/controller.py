@@ -134,6 +134,10 @@ class JobController:
         for h in hs.values():
             self.runtime.poll(h)
 
+        # state-file bridge for process runtimes (the "API server" channel
+        # between controller and the rank-0 checkpoint agent)
+        self._sync_state_files(job)
+
         # release GPU slots of finished tasks
         for h in hs.values():
             if h.finished and h.gpu_slots:
@@ -340,7 +344,14 @@ class JobController:
     def _on_success(self, job: TorchJob, hs: dict):
         # model packaging (job.go:462-508): output dir -> ModelVersion
         if self.model_registry is not None and job.model_name:
-            mv = self.model_registry.create_version_for_job(job)
+            import os
+            src = None
+            workdir = getattr(self.runtime, "workdir", None)
+            if workdir:
+                cand = os.path.join(workdir, job.name, "output")
+                if os.path.isdir(cand):
+                    src = cand
+            mv = self.model_registry.create_version_for_job(job, src_dir=src)
             job.status.model_version = mv.version if mv else None
         self._cleanup(job, kill_all=(job.run_policy.clean_task_policy
                                      != CleanPodPolicy.NONE))
@@ -368,6 +379,39 @@ class JobController:
             self.reconcile(job)
 
     # ------------------------------------------------------------------
+    def _sync_state_files(self, job: TorchJob):
+        """With a LocalProcessRuntime, annotations travel through per-job
+        state files: controller writes job.json, the rank-0 checkpoint
+        agent writes agent.json (entrypoint.py). Single writer per file;
+        atomic replace on both sides."""
+        import json
+        import os
+        workdir = getattr(self.runtime, "workdir", None)
+        if not workdir:
+            return
+        jobdir = os.path.join(workdir, job.name)
+        os.makedirs(jobdir, exist_ok=True)
+        agent = None
+        try:
+            with open(os.path.join(jobdir, "agent.json")) as f:
+                agent = json.load(f)
+        except (OSError, ValueError):
+            pass
+        if agent and agent.get("ckpt-completed-version"):
+            from torch_on_k8s_amd.controlplane.api import ANN_CKPT_COMPLETED
+            job.annotations[ANN_CKPT_COMPLETED] = json.dumps(
+                agent["ckpt-completed-version"])
+        tmp = os.path.join(jobdir, f"job.json.tmp{os.getpid()}")
+        with open(tmp, "w") as f:
+            json.dump({
+                "name": job.name,
+                "generation": job.generation,
+                "annotations": job.annotations,
+                "replicas": {t.value: s.replicas
+                             for t, s in job.tasks.items()},
+            }, f)
+        os.replace(tmp, os.path.join(jobdir, "job.json"))
+
     def reconcile_all(self):
         for job in list(self.jobs.values()):
             self.reconcile(job)

@@ -79,12 +79,12 @@ class ModelRegistry:
             self.models[name] = Model(name)
         return self.models[name]
 
-    def create_version_for_job(self, job) -> ModelVersion | None:
+    def create_version_for_job(self, job, src_dir: str | None = None) -> ModelVersion | None:
         """mv-<job>-<uid5> naming parity (job.go:462-508)."""
         if not job.model_name:
             return None
         version = f"mv-{job.name}-{str(job.uid).zfill(5)[-5:]}"
-        src = self.storage.job_output_dir(job.name)
+        src = src_dir or self.storage.job_output_dir(job.name)
         return self.build_version(job.model_name, version, src,
                                   source_job=job.name)
 

@@ -151,6 +151,7 @@ class LocalProcessRuntime(Runtime):
             env["HIP_VISIBLE_DEVICES"] = ",".join(str(s) for s in gpu_slots)
             env["LOCAL_RANK"] = "0"  # each process sees exactly its GPUs
         logdir = os.path.join(self.workdir, job.name)
+        env.setdefault("TOK_STATE_DIR", logdir)
         os.makedirs(logdir, exist_ok=True)
         argv = spec.command or [
             self.python, "-m", "torch_on_k8s_amd.entrypoint"]

@@ -1,0 +1,130 @@
+"""Training task entrypoint — the process the control plane launches for
+Master/Worker tasks (reference: the opaque user container command the
+operator wires env vars into, torchjob_controller.go:314-449; here the
+data plane is part of the framework).
+
+Env contract (injected by runtime.cluster_env):
+  MASTER_ADDR / MASTER_PORT / RANK / WORLD_SIZE / PYTHONUNBUFFERED
+  TOK_JOB_NAME / TOK_TASK_TYPE / TOK_TASK_INDEX / TOK_GENERATION
+Trainer configuration:
+  TOK_TRAINER_CONFIG  JSON dict of TrainerConfig fields
+  TOK_TRAIN_STEPS     total steps to run (default 10)
+  TOK_BACKEND         gloo | nccl (default: nccl if GPU else gloo)
+  TOK_STATE_DIR       job state dir: job.json (controller-written),
+                      agent.json (trainer-written), ckpt/, metrics.json,
+                      output/ (model artifact source)
+Exit codes follow the failover contract (controlplane/failover.py):
+  0 success; 143 on SIGTERM after a clean checkpoint (retryable).
+
+Checkpoint-agent duty: the reference's external
+AIMaster watches ckpt-requested-version and checkpoints out-of-band.
+Here rank 0 plays that role: it polls job.json for a checkpoint request,
+writes the checkpoint, and reports completion through agent.json.
+"""
+from __future__ import annotations
+
+import json
+import os
+import signal
+import sys
+import time
+
+
+def _atomic_write(path: str, obj: dict):
+    tmp = path + f".tmp{os.getpid()}"
+    with open(tmp, "w") as f:
+        json.dump(obj, f)
+    os.replace(tmp, path)
+
+
+def _read_json(path: str):
+    try:
+        with open(path) as f:
+            return json.load(f)
+    except (OSError, ValueError):
+        return None
+
+
+def main() -> int:
+    import torch
+    from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
+    from torch_on_k8s_amd.parallel.env import init_distributed, destroy, barrier
+
+    backend = os.environ.get("TOK_BACKEND")
+    steps_total = int(os.environ.get("TOK_TRAIN_STEPS", "10"))
+    state_dir = os.environ.get("TOK_STATE_DIR", "")
+    cfg_overrides = json.loads(os.environ.get("TOK_TRAINER_CONFIG", "{}"))
+
+    ctx = init_distributed(backend=backend)
+    cfg_kw = dict(model="llama3-8b")
+    cfg_kw.update(cfg_overrides)
+    if state_dir and ctx.is_main:
+        os.makedirs(state_dir, exist_ok=True)
+        cfg_kw.setdefault("metrics_path", os.path.join(state_dir, "metrics.json"))
+    cfg = TrainerConfig(**cfg_kw)
+    trainer = Trainer(cfg, ctx)
+
+    ckpt_dir = os.path.join(state_dir, "ckpt") if state_dir else None
+    job_file = os.path.join(state_dir, "job.json") if state_dir else None
+    agent_file = os.path.join(state_dir, "agent.json") if state_dir else None
+
+    # resume (elastic restart with new WORLD_SIZE lands here)
+    if ckpt_dir and os.path.exists(os.path.join(ckpt_dir, "meta.json")):
+        trainer.load_checkpoint(ckpt_dir)
+        print(f"[entrypoint] resumed at step {trainer.step_count}", flush=True)
+
+    stop = {"sig": None}
+
+    def on_term(signum, frame):
+        stop["sig"] = signum
+
+    signal.signal(signal.SIGTERM, on_term)
+
+    def checkpoint_and_ack(version=None):
+        if ckpt_dir:
+            barrier(ctx)
+            trainer.save_checkpoint(ckpt_dir)
+            if ctx.is_main and agent_file and version is not None:
+                _atomic_write(agent_file, {
+                    "ckpt-completed-version": {"version": version,
+                                               "status": "Succeeded"},
+                    "step": trainer.step_count,
+                })
+
+    last_completed = None
+    while trainer.step_count < steps_total:
+        loss = trainer.train_step()
+        if ctx.is_main:
+            print(f"[train] step={trainer.step_count} loss={loss:.4f}",
+                  flush=True)
+        # checkpoint-agent poll (rank 0): serve controller requests
+        if ctx.is_main and job_file:
+            jf = _read_json(job_file)
+            req = (jf or {}).get("annotations", {}).get(
+                "ckpt-requested-version")
+            if isinstance(req, str):
+                try:
+                    req = json.loads(req)
+                except ValueError:
+                    req = None
+            if req and req.get("version") != last_completed:
+                checkpoint_and_ack(req["version"])
+                last_completed = req["version"]
+        if stop["sig"] is not None:
+            checkpoint_and_ack()
+            destroy()
+            return 143  # SIGTERM: clean checkpointed exit, retryable
+
+    # final checkpoint becomes the model artifact (output/ packaged by the
+    # control plane into a ModelVersion on job success)
+    if state_dir and ctx.is_main:
+        out = os.path.join(state_dir, "output")
+        os.makedirs(out, exist_ok=True)
+        trainer.save_checkpoint(os.path.join(out, "final"))
+    barrier(ctx)
+    destroy()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
