@@ -1,0 +1,166 @@
+"""Framework manager daemon — the operator process (reference main.go).
+
+    python -m torch_on_k8s_amd.manager --workdir /var/run/tok \
+        --num-gpus 8 --metrics-addr 8443 \
+        --feature-gates GangScheduling=true,JobCoordinator=true \
+        --quota teamA=8 --quota teamB=4
+
+Watches <workdir>/spool/ for TorchJob YAMLs (the kubectl-apply analog),
+runs the coordinator (100ms schedule loop), the reconcile loop, the
+metric-driven elastic autoscaler pass, and the Prometheus metrics server.
+Job status is published to <workdir>/status/<job>.json; deleting the
+spool file deletes the job.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+from torch_on_k8s_amd.controlplane import features as feat
+from torch_on_k8s_amd.controlplane.controller import (ControllerConfig,
+                                                      JobController)
+from torch_on_k8s_amd.controlplane.coordinator import Coordinator
+from torch_on_k8s_amd.controlplane.elastic import (ElasticScaler,
+                                                   TorchElasticAutoscaler,
+                                                   read_trainer_metrics)
+from torch_on_k8s_amd.controlplane.jobspec import (job_from_yaml,
+                                                   job_status_dict)
+from torch_on_k8s_amd.controlplane.metrics import (JobMetrics,
+                                                   start_metrics_server)
+from torch_on_k8s_amd.controlplane.modelregistry import (ModelRegistry,
+                                                         StorageProvider)
+from torch_on_k8s_amd.controlplane.node import NodeState
+from torch_on_k8s_amd.controlplane.runtime import LocalProcessRuntime
+
+
+class Manager:
+    def __init__(self, workdir: str, num_gpus: int = 8, quotas=None,
+                 gates: feat.FeatureGates | None = None,
+                 sync_period: float = 0.5):
+        self.workdir = workdir
+        self.spool = os.path.join(workdir, "spool")
+        self.status_dir = os.path.join(workdir, "status")
+        for d in (self.spool, self.status_dir):
+            os.makedirs(d, exist_ok=True)
+        self.gates = gates or feat.FeatureGates()
+        self.sync_period = sync_period
+
+        node = NodeState(num_gpus=num_gpus)
+        runtime = LocalProcessRuntime(os.path.join(workdir, "jobs"))
+        storage = StorageProvider(os.path.join(workdir, "models"))
+        self.registry = ModelRegistry(storage)
+        self.metrics = JobMetrics()
+        cfg = ControllerConfig(
+            enable_gang_scheduling=self.gates.enabled(feat.GANG_SCHEDULING),
+            enable_dag_scheduling=self.gates.enabled(feat.DAG_SCHEDULING))
+        self.controller = JobController(
+            node, runtime, cfg, metrics=self.metrics,
+            model_registry=self.registry, elastic=ElasticScaler())
+        if self.gates.enabled(feat.JOB_COORDINATOR):
+            self.coordinator = Coordinator(
+                dequeue_fn=self.controller.reconcile,
+                tenant_usage_fn=self.controller.tenant_gpu_usage,
+                quotas=quotas, default_quota=num_gpus)
+            self.controller.coordinator = self.coordinator
+        else:
+            self.coordinator = None
+        self.autoscaler = TorchElasticAutoscaler(
+            read_trainer_metrics(lambda job: os.path.join(
+                workdir, "jobs", job.name, "metrics.json")))
+        self._spooled: dict[str, str] = {}  # filename -> job name
+
+    # ------------------------------------------------------------------
+    def sync_spool(self):
+        """kubectl-apply/delete analog over the spool directory."""
+        try:
+            files = {f for f in os.listdir(self.spool)
+                     if f.endswith((".yaml", ".yml", ".json"))}
+        except OSError:
+            return
+        for f in sorted(files - set(self._spooled)):
+            path = os.path.join(self.spool, f)
+            try:
+                with open(path) as fh:
+                    job = job_from_yaml(fh.read())
+            except Exception as e:  # malformed spec: surface, skip
+                self.controller.event("-", "Warning", "BadJobSpec",
+                                      f"{f}: {e}")
+                self._spooled[f] = ""
+                continue
+            if job.name in self.controller.jobs:
+                self._spooled[f] = job.name
+                continue
+            self.metrics.job_created_at(job.name)
+            self.controller.create_job(job)
+            self._spooled[f] = job.name
+        for f in set(self._spooled) - files:   # spool file removed
+            name = self._spooled.pop(f)
+            if name:
+                self.controller.delete_job(name)
+
+    def publish_status(self):
+        for name, job in list(self.controller.jobs.items()):
+            path = os.path.join(self.status_dir, f"{name}.json")
+            tmp = path + ".tmp"
+            with open(tmp, "w") as f:
+                json.dump(job_status_dict(job), f, indent=2)
+            os.replace(tmp, path)
+        if self.coordinator is not None:
+            for tenant, q in self.coordinator.queues.items():
+                self.metrics.set_queue_depth(tenant, len(q))
+
+    def autoscale_pass(self):
+        for job in list(self.controller.jobs.values()):
+            if job.elastic is None:
+                continue
+            st = job.status
+            from torch_on_k8s_amd.controlplane.api import JobConditionType
+            if st.phase != JobConditionType.RUNNING:
+                continue
+            self.autoscaler.observe(job)
+            self.autoscaler.decide(job)
+
+    def step(self):
+        self.sync_spool()
+        if self.coordinator is not None:
+            self.coordinator.schedule_once()
+        self.controller.reconcile_all()
+        self.autoscale_pass()
+        self.publish_status()
+
+    def run_forever(self):
+        while True:
+            self.step()
+            time.sleep(self.sync_period)
+
+
+def main():
+    ap = argparse.ArgumentParser(prog="torch-on-k8s-amd-manager")
+    ap.add_argument("--workdir", default="/tmp/torch-on-k8s-amd")
+    ap.add_argument("--num-gpus", type=int,
+                    default=int(os.environ.get("TOK_NUM_GPUS", "8")))
+    ap.add_argument("--metrics-addr", type=int, default=8443)
+    ap.add_argument("--feature-gates", default="")
+    ap.add_argument("--quota", action="append", default=[],
+                    help="tenant=gpus, repeatable")
+    ap.add_argument("--sync-period", type=float, default=0.5)
+    args = ap.parse_args()
+
+    quotas = {}
+    for q in args.quota:
+        k, _, v = q.partition("=")
+        quotas[k] = int(v)
+    gates = feat.FeatureGates.from_flag(args.feature_gates)
+    mgr = Manager(args.workdir, num_gpus=args.num_gpus,
+                  quotas=quotas or None, gates=gates,
+                  sync_period=args.sync_period)
+    start_metrics_server(args.metrics_addr)
+    print(f"[manager] workdir={args.workdir} gpus={args.num_gpus} "
+          f"gates={gates.as_dict()}", flush=True)
+    mgr.run_forever()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,27 +1,27 @@
 // Flash attention (causal, GQA) for MI355X (gfx950) — forward + backward.
 //
-// MI355X-first design (cdna_hip_programming.md §Appendix B):
-//  * MFMA 16x16x32 bf16 tiles, fp32 accumulation in AGPR/VGPR file.
-//  * wave64: a block = 4 waves; each wave owns a 16-row strip of a
-//    64x64 score tile; online softmax (running m,l) entirely in
-//    registers with 16-lane shuffle row-reductions.
-//  * K / V / Q / dO tiles staged in LDS with the XOR swizzle
-//    (byte ^= (row&7)<<4) — row-major [*][128] bf16 tiles read by
-//    ds_read_b128 are otherwise an up-to-16-way bank conflict
-//    (guide §6 Guideline 4).
-//  * V (and dO/Q in backward) additionally staged TRANSPOSED so every
-//    MFMA B-fragment read is one contiguous ds_read_b128.
-//  * Fragment layouts (verified on hardware by mfma_probe):
+// v2 structure (profiling-driven, see profiles/: v1 was 47% of the 8B
+// step at ~100 TF):
+//  * 8-wave blocks (512 threads): fwd/dq tile 128 q rows x 64 kv,
+//    dkdv 128 kv rows x 64 q — tile staging and barriers amortized over
+//    2x the rows of v1.
+//  * async-STAGE split (guide §6 G15 / T14, +17% on attention): the next
+//    KV tile's global loads are issued into registers during the current
+//    tile's MFMA phase; the LDS write lands after the barrier.
+//  * MFMA 16x16x32 bf16; fragment layouts verified on HW by mfma_probe:
 //      A[m][k]: m=lane&15, k=(lane>>4)*8+j
 //      B[k][n]: k=(lane>>4)*8+j, n=lane&15
 //      C/D[m][n]: m=(lane>>4)*4+r, n=lane&15
+//  * LDS XOR swizzle byte^=(row&7)<<4 on every staged tile (row-major
+//    [*][128] bf16 read by ds_read_b128 is otherwise up to 16-way bank
+//    conflict, guide §6 G4).
+//  * V (and Q/dO transposes in backward) staged TRANSPOSED so every
+//    MFMA B-fragment read is one contiguous ds_read_b128.
+//  * online softmax fully in registers; 16-lane shuffle row reduce.
 //
-// Backward uses the standard two-kernel flash scheme with recompute:
-//   dkdv: grid over KV tiles; computes S^T directly (A=K,B=Q) so P^T
-//         needs no lse/Dsum transpose; accumulates dK,dV in registers
-//         across Q tiles and the GQA q-head group.
-//   dq:   grid over Q tiles; recomputes S,P and accumulates dQ.
-// Plus a tiny preprocess kernel: Dsum = rowsum(dO * O).
+// Backward: standard two-kernel flash recompute scheme — dkdv over KV
+// tiles computes S^T directly (A=K,B=Q: no lse/Dsum transpose), dq over
+// Q tiles; plus a Dsum = rowsum(dO*O) preprocess.
 #include "common.h"
 
 namespace {
@@ -29,10 +29,10 @@ namespace {
 using bf16x8v = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-constexpr int BM = 64;   // q rows per block
-constexpr int BN = 64;   // kv rows per tile
-constexpr int NW = 4;    // waves per block
-constexpr int NTHREADS = NW * WAVE;
+constexpr int BN = 64;    // kv rows per LDS tile (fwd/dq); q rows in dkdv
+constexpr int NW = 8;     // waves per block
+constexpr int NTHREADS = NW * WAVE;   // 512
+constexpr int BM = NW * 16;           // 128 rows owned per block (1 strip/wave)
 constexpr float NEG_INF = -INFINITY;
 
 DEVINL int swz(int row, int byte_in_row) {
@@ -45,54 +45,60 @@ DEVINL bf16x8v as_frag(uint4 raw) {
   return c.f;
 }
 
-// ---- cooperative tile staging -----------------------------------------
-// Row-major [64][D] bf16 tile, swizzled; rows beyond row_limit -> 0.
+// ---- two-phase tile staging (async-STAGE split) -----------------------
+// A [64][D] bf16 tile staged by all 512 threads: issue() starts the
+// global loads into registers; write_rm()/write_tr() put them into LDS
+// (row-major swizzled / transposed swizzled). OOB rows load zero.
 template <int D>
-DEVINL void load_tile_rm(char* lds, const bf16_t* src, long row0,
-                         long row_limit, long tok_stride) {
-  constexpr int VPR = D / 8;
-  constexpr int NV = 64 * VPR;
-#pragma unroll 2
-  for (int vi = threadIdx.x; vi < NV; vi += NTHREADS) {
-    const int row = vi / VPR, cv = vi % VPR;
-    uint4 val = {0, 0, 0, 0};
-    if (row0 + row < row_limit)
-      val = *(const uint4*)(src + (row0 + row) * tok_stride + cv * 8);
-    *(uint4*)(lds + row * (D * 2) + swz(row, cv * 16)) = val;
-  }
-}
+struct TileStage {
+  static constexpr int VPR = D / 8;              // 16B vecs per row
+  static constexpr int VPT = 64 * VPR / NTHREADS;  // vecs per thread
+  uint4 vals[VPT];
 
-// Transposed [D][64] bf16 tile (row = d, col = src row), swizzled.
-template <int D>
-DEVINL void load_tile_tr(char* lds, const bf16_t* src, long row0,
-                         long row_limit, long tok_stride) {
-  constexpr int VPR = D / 8;
-  constexpr int NV = 64 * VPR;
-#pragma unroll 2
-  for (int vi = threadIdx.x; vi < NV; vi += NTHREADS) {
-    const int row = vi / VPR, cv = vi % VPR;
-    uint4 val = {0, 0, 0, 0};
-    if (row0 + row < row_limit)
-      val = *(const uint4*)(src + (row0 + row) * tok_stride + cv * 8);
-    const uint16_t* h = (const uint16_t*)&val;
+  DEVINL void issue(const bf16_t* src, long row0, long row_limit,
+                    long tok_stride) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int d = cv * 8 + j;
-      *(uint16_t*)(lds + d * 128 + swz(d, row * 2)) = h[j];
+    for (int i = 0; i < VPT; ++i) {
+      const int vi = threadIdx.x + i * NTHREADS;
+      const int row = vi / VPR, cv = vi % VPR;
+      uint4 val = {0, 0, 0, 0};
+      if (row0 + row < row_limit)
+        val = *(const uint4*)(src + (row0 + row) * tok_stride + cv * 8);
+      vals[i] = val;
     }
   }
-}
 
-// B-fragment read from a swizzled row-major LDS tile with row stride RB
-// bytes: B[k][n] where n = row (16 rows starting at row0), k contiguous.
+  DEVINL void write_rm(char* lds) const {
+#pragma unroll
+    for (int i = 0; i < VPT; ++i) {
+      const int vi = threadIdx.x + i * NTHREADS;
+      const int row = vi / VPR, cv = vi % VPR;
+      *(uint4*)(lds + row * (D * 2) + swz(row, cv * 16)) = vals[i];
+    }
+  }
+
+  DEVINL void write_tr(char* lds) const {
+#pragma unroll
+    for (int i = 0; i < VPT; ++i) {
+      const int vi = threadIdx.x + i * NTHREADS;
+      const int row = vi / VPR, cv = vi % VPR;
+      const uint16_t* h = (const uint16_t*)&vals[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d = cv * 8 + j;
+        *(uint16_t*)(lds + d * 128 + swz(d, row * 2)) = h[j];
+      }
+    }
+  }
+};
+
+// B-fragment read from a swizzled row-major LDS tile, row stride RB bytes.
 template <int RB>
 DEVINL bf16x8v read_bfrag(const char* lds, int row, int col_elem) {
-  return as_frag(
-      *(const uint4*)(lds + row * RB + swz(row, col_elem * 2)));
+  return as_frag(*(const uint4*)(lds + row * RB + swz(row, col_elem * 2)));
 }
 
-// Load an A-fragment set (rows m = lane&15 within a 16-row strip) straight
-// from global memory into registers; OOB rows -> 0.
+// A-fragment set (rows m = lane&15 of a 16-row strip) from global.
 template <int D>
 DEVINL void load_afrags(bf16x8v* frag, const bf16_t* src, long row,
                         long row_limit, long tok_stride, int lane) {
@@ -107,17 +113,16 @@ DEVINL void load_afrags(bf16x8v* frag, const bf16_t* src, long row,
 }
 
 // ========================================================================
-// Forward
+// Forward: grid (ceil(S/128), Hq, B); block = 8 waves, wave = 16 q rows.
 // ========================================================================
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, bf16_t* __restrict__ o,
     float* __restrict__ lse, int B, int S, int Hq, int Hkv, float scale) {
-  constexpr int DC = D / 32;  // MFMA k-chunks
-  constexpr int DT = D / 16;  // output col tiles
-  constexpr int KB = 64 * D * 2;
-  // LDS carve: K tile | VT tile | P scratch (2 KiB per wave)
+  constexpr int DC = D / 32;
+  constexpr int DT = D / 16;
+  constexpr int KB = BN * D * 2;
   __shared__ __attribute__((aligned(16))) char smem[KB + KB + NW * 2048];
   char* k_lds = smem;
   char* vt_lds = smem + KB;
@@ -137,7 +142,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
   bf16_t* op = o + ((long)b * S * q_tok) + (long)hq * D;
   float* lsep = lse + ((long)b * Hq + hq) * S;
 
-  // Q fragments for this wave's 16-row strip
   bf16x8v q_frag[DC];
   const int mrow = m0 + wid * 16 + (lane & 15);
   load_afrags<D>(q_frag, qp, mrow, S, q_tok, lane);
@@ -148,15 +152,24 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
   float m_run[4] = {NEG_INF, NEG_INF, NEG_INF, NEG_INF};
   float l_run[4] = {0.f, 0.f, 0.f, 0.f};
 
+  TileStage<D> k_st, v_st;
   const int n_end = CAUSAL ? min(S, m0 + BM) : S;
+  k_st.issue(kp, 0, S, kv_tok);
+  v_st.issue(vp, 0, S, kv_tok);
+
   for (int n0 = 0; n0 < n_end; n0 += BN) {
-    load_tile_rm<D>(k_lds, kp, n0, S, kv_tok);
-    load_tile_tr<D>(vt_lds, vp, n0, S, kv_tok);
+    // stage tile n0 into LDS (loads were issued last iteration)
+    k_st.write_rm(k_lds);
+    v_st.write_tr(vt_lds);
     __syncthreads();
+    // prefetch tile n0+BN during compute
+    if (n0 + BN < n_end) {
+      k_st.issue(kp, n0 + BN, S, kv_tok);
+      v_st.issue(vp, n0 + BN, S, kv_tok);
+    }
 
     const bool strip_live = !CAUSAL || (n0 <= m0 + wid * 16 + 15);
     if (strip_live) {
-      // S strip: [16 rows x 64 cols] per wave
       f32x4 s_acc[4];
 #pragma unroll
       for (int t = 0; t < 4; ++t) s_acc[t] = {0.f, 0.f, 0.f, 0.f};
@@ -170,8 +183,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
               q_frag[c], bfr, s_acc[t], 0, 0, 0);
         }
 
-      // online softmax on the strip
-      float p[4][4];     // [t][r]
+      float p[4][4];
       float mx[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) mx[r] = NEG_INF;
@@ -224,7 +236,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) o_acc[t][r] *= alpha[r];
 
-      // P -> LDS (per-wave scratch) to convert C-layout to A-layout
+      // P -> per-wave LDS scratch (C-layout -> A-layout)
       char* pw = p_lds + wid * 2048;
 #pragma unroll
       for (int t = 0; t < 4; ++t)
@@ -234,7 +246,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
           const int cl = t * 16 + (lane & 15);
           *(uint16_t*)(pw + rl * 128 + swz(rl, cl * 2)) = f2bfbits(p[t][r]);
         }
-      // PV: O += P * V
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
         bf16x8v pa = as_frag(*(const uint4*)(
@@ -252,7 +263,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
     __syncthreads();
   }
 
-  // epilogue: O = o_acc / l; LSE = m + log(l)
   const int row_base = m0 + wid * 16 + (lane >> 4) * 4;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -277,7 +287,6 @@ __global__ void attn_bwd_pre_kernel(const bf16_t* __restrict__ dout,
                                     const bf16_t* __restrict__ o,
                                     float* __restrict__ dsum, long rows,
                                     int Hq, int S) {
-  // one thread per row; row index = (b*S + s)*Hq + h ; dsum is [B,Hq,S]
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < rows;
        i += (long)gridDim.x * blockDim.x) {
     const bf16_t* dp = dout + i * D;
@@ -293,7 +302,7 @@ __global__ void attn_bwd_pre_kernel(const bf16_t* __restrict__ dout,
       for (int j = 0; j < 8; ++j)
         acc = fmaf(bfbits2f(ah[j]), bfbits2f(bh[j]), acc);
     }
-    const long tok = i / Hq;        // b*S + s
+    const long tok = i / Hq;
     const long h = i - tok * Hq;
     const long b_ = tok / S;
     const long s_ = tok - b_ * S;
@@ -302,8 +311,8 @@ __global__ void attn_bwd_pre_kernel(const bf16_t* __restrict__ dout,
 }
 
 // ========================================================================
-// Backward dK/dV: grid over KV tiles (per kv-head); loops q tiles and the
-// GQA q-head group, accumulating dK/dV in registers.
+// Backward dK/dV: grid (ceil(S/128), Hkv, B); block = 8 waves, wave =
+// 16 kv rows; loops 64-row Q tiles and the GQA q-head group.
 // ========================================================================
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
@@ -314,8 +323,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
     int Hkv, float scale) {
   constexpr int DC = D / 32;
   constexpr int DT = D / 16;
-  constexpr int KB = 64 * D * 2;
-  // Q | QT | dO | dOT | per-wave scratch
+  constexpr int KB = BN * D * 2;
   __shared__ __attribute__((aligned(16))) char smem[4 * KB + NW * 2048];
   char* q_lds = smem;
   char* qt_lds = smem + KB;
@@ -325,7 +333,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int n0 = blockIdx.x * BN;
+  const int n0 = blockIdx.x * BM;       // 128 kv rows per block
   const int hkv = blockIdx.y;
   const int b = blockIdx.z;
   const int rep = Hq / Hkv;
@@ -336,7 +344,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
   bf16_t* dkp = dk + ((long)b * S * kv_tok) + (long)hkv * D;
   bf16_t* dvp = dv + ((long)b * S * kv_tok) + (long)hkv * D;
 
-  // this wave's 16 kv rows: A-fragments of K and V
   const int nrow = n0 + wid * 16 + (lane & 15);
   bf16x8v k_frag[DC], v_frag[DC];
   load_afrags<D>(k_frag, kp, nrow, S, kv_tok, lane);
@@ -350,8 +357,15 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
   }
 
   char* pw = p_lds + wid * 2048;
-  const int m_start = CAUSAL ? (n0 / BM) * BM : 0;
+  const int m_start = CAUSAL ? (n0 / BN) * BN : 0;
 
+  TileStage<D> q_st, do_st;
+  // first tile of the first q-head group; subsequent tiles (including
+  // group transitions) are prefetched inside the loop
+  q_st.issue(q + ((long)b * S * q_tok) + (long)(hkv * rep) * D, m_start, S,
+             q_tok);
+  do_st.issue(dout + ((long)b * S * q_tok) + (long)(hkv * rep) * D, m_start,
+              S, q_tok);
   for (int g = 0; g < rep; ++g) {
     const int hq = hkv * rep + g;
     const bf16_t* qp = q + ((long)b * S * q_tok) + (long)hq * D;
@@ -359,14 +373,25 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
     const float* lsep = lse + ((long)b * Hq + hq) * S;
     const float* dsp = dsum + ((long)b * Hq + hq) * S;
 
-    for (int m0 = m_start; m0 < S; m0 += BM) {
-      load_tile_rm<D>(q_lds, qp, m0, S, q_tok);
-      load_tile_tr<D>(qt_lds, qp, m0, S, q_tok);
-      load_tile_rm<D>(do_lds, dop, m0, S, q_tok);
-      load_tile_tr<D>(dot_lds, dop, m0, S, q_tok);
+    for (int m0 = m_start; m0 < S; m0 += BN) {
+      q_st.write_rm(q_lds);
+      q_st.write_tr(qt_lds);
+      do_st.write_rm(do_lds);
+      do_st.write_tr(dot_lds);
       __syncthreads();
+      // prefetch the next Q/dO tile (wrap into the next q-head group)
+      const int m1 = m0 + BN;
+      if (m1 < S) {
+        q_st.issue(qp, m1, S, q_tok);
+        do_st.issue(dop, m1, S, q_tok);
+      } else if (g + 1 < rep) {
+        const bf16_t* qp2 = q + ((long)b * S * q_tok) + (long)(hq + 1) * D;
+        const bf16_t* dop2 = dout + ((long)b * S * q_tok) + (long)(hq + 1) * D;
+        q_st.issue(qp2, m_start, S, q_tok);
+        do_st.issue(dop2, m_start, S, q_tok);
+      }
 
-      // S^T strip: rows = kv n (this wave's 16), cols = q m (64)
+      // S^T / dP^T strips: rows = 16 kv, cols = 64 q
       f32x4 st[4], dpt[4];
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
@@ -387,7 +412,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
               v_frag[c], dob, dpt[t], 0, 0, 0);
         }
 
-      // P^T = exp(scale*S^T - lse[col]); dS^T = P^T*(dP^T - Dsum[col])
       const int nrow_base = n0 + wid * 16 + (lane >> 4) * 4;
       float pt[4][4], dst[4][4];
 #pragma unroll
@@ -408,7 +432,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
         }
       }
 
-      // dV += P^T * dO  (A = P^T via scratch, B = dOT)
+      // dV += P^T * dO
 #pragma unroll
       for (int t = 0; t < 4; ++t)
 #pragma unroll
@@ -431,7 +455,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
         }
       }
 
-      // dK += dS^T * Q  (A = dS^T via scratch, B = QT); scale folded later
+      // dK += dS^T * Q (scale folded at epilogue)
 #pragma unroll
       for (int t = 0; t < 4; ++t)
 #pragma unroll
@@ -457,7 +481,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
     }
   }
 
-  // epilogue: write dK (scaled), dV
   const int nrow_base = n0 + wid * 16 + (lane >> 4) * 4;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -473,7 +496,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
 }
 
 // ========================================================================
-// Backward dQ: grid over Q tiles; recompute S,P; dQ += scale * dS * K
+// Backward dQ: grid (ceil(S/128), Hq, B); block = 8 waves, wave = 16 q rows.
 // ========================================================================
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
@@ -483,8 +506,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
     bf16_t* __restrict__ dq, int B, int S, int Hq, int Hkv, float scale) {
   constexpr int DC = D / 32;
   constexpr int DT = D / 16;
-  constexpr int KB = 64 * D * 2;
-  // K | KT | V | per-wave scratch
+  constexpr int KB = BN * D * 2;
   __shared__ __attribute__((aligned(16))) char smem[3 * KB + NW * 2048];
   char* k_lds = smem;
   char* kt_lds = smem + KB;
@@ -512,7 +534,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
   load_afrags<D>(q_frag, qp, mrow, S, q_tok, lane);
   load_afrags<D>(do_frag, dop, mrow, S, q_tok, lane);
 
-  // per-row lse/Dsum (C-layout rows)
   const int row_base = m0 + wid * 16 + (lane >> 4) * 4;
   float lse_r[4], ds_r[4];
 #pragma unroll
@@ -528,11 +549,19 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
 
   char* pw = p_lds + wid * 2048;
   const int n_end = CAUSAL ? min(S, m0 + BM) : S;
+  TileStage<D> k_st, v_st;
+  k_st.issue(kp, 0, S, kv_tok);
+  v_st.issue(vp, 0, S, kv_tok);
+
   for (int n0 = 0; n0 < n_end; n0 += BN) {
-    load_tile_rm<D>(k_lds, kp, n0, S, kv_tok);
-    load_tile_tr<D>(kt_lds, kp, n0, S, kv_tok);
-    load_tile_rm<D>(v_lds, vp, n0, S, kv_tok);
+    k_st.write_rm(k_lds);
+    k_st.write_tr(kt_lds);
+    v_st.write_rm(v_lds);
     __syncthreads();
+    if (n0 + BN < n_end) {
+      k_st.issue(kp, n0 + BN, S, kv_tok);
+      v_st.issue(vp, n0 + BN, S, kv_tok);
+    }
 
     const bool strip_live = !CAUSAL || (n0 <= m0 + wid * 16 + 15);
     if (strip_live) {
@@ -556,7 +585,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
               do_frag[c], vb, dp_acc[t], 0, 0, 0);
         }
 
-      // dS = P*(dP - Dsum[row]); P = exp(scale*S - lse[row])
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
         const int col = n0 + t * 16 + (lane & 15);
@@ -573,7 +601,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
           *(uint16_t*)(pw + rl * 128 + swz(rl, cl * 2)) = f2bfbits(dsv);
         }
       }
-      // dQ += dS * K (A = dS via scratch, B = KT)
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
         bf16x8v da = as_frag(*(const uint4*)(
@@ -591,7 +618,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
     __syncthreads();
   }
 
-  // epilogue
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int row = row_base + r;
@@ -642,7 +668,7 @@ hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
   else
     return hipErrorInvalidValue;
 
-  dim3 gkv((S + BN - 1) / BN, Hkv, B);
+  dim3 gkv((S + BM - 1) / BM, Hkv, B);
   dim3 gq((S + BM - 1) / BM, Hq, B);
 #define LAUNCH_BWD(DD, CC)                                                    \
   do {                                                                        \
