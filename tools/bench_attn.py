@@ -6,7 +6,11 @@ attention at the flagship shape, A/B against torch SDPA. Within-process
 interleaved rounds (guide §5.4 rule 24).
 """
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

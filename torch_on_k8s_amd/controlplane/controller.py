@@ -66,12 +66,25 @@ class JobController:
         self.jobs: dict[str, TorchJob] = {}
         self.handles: dict[str, dict] = {}   # job -> {key: TaskHandle}
         self.events: list[Event] = []
+        self._event_last: dict = {}
         self._ports: dict[str, int] = {}
         self._next_port = self.cfg.master_port_range[0]
         self._first_task_ts: dict[str, float] = {}
 
     # ------------------------------------------------------------------
+    EVENT_DEDUP_WINDOW = 5.0  # seconds
+
     def event(self, job: str, etype: str, reason: str, msg: str = ""):
+        """Flow-controlled event recorder (reference
+        utils/flowcontrol/recorder.go:33-122: qps-bounded, dedup by
+        source): identical (job, reason) events within the window are
+        coalesced instead of appended."""
+        now = time.time()
+        key = (job, reason)
+        last = self._event_last.get(key)
+        if last is not None and now - last < self.EVENT_DEDUP_WINDOW:
+            return
+        self._event_last[key] = now
         self.events.append(Event(job, etype, reason, msg))
 
     def create_job(self, job: TorchJob) -> TorchJob:
