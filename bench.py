@@ -36,6 +36,8 @@ def parse_args():
     ap.add_argument("--attn", type=str, default="hip", choices=["hip", "sdpa"])
     ap.add_argument("--no-overlap", action="store_true",
                     help="disable grad-sync/backward overlap (ablation)")
+    ap.add_argument("--hip-graph", action="store_true",
+                    help="capture the training step in a hipGraph")
     return ap.parse_args()
 
 
@@ -58,6 +60,7 @@ def main():
         bucket_mb=args.bucket_mb,
         activation_checkpointing=args.activation_checkpointing,
         overlap_grad_sync=not args.no_overlap,
+        hip_graph=args.hip_graph,
     )
     trainer = Trainer(cfg, ctx)
 

@@ -35,6 +35,8 @@ class TrainerConfig:
     bucket_mb: int = 256
     activation_checkpointing: bool = False
     overlap_grad_sync: bool = True
+    hip_graph: bool = False          # capture the whole step in a hipGraph
+                                     # and replay it (needs grad_clip=0)
     dtype: str = "bf16"             # compute dtype on GPU
     seed: int = 1234
     metrics_path: str | None = None  # structured metrics for the elastic
@@ -69,6 +71,11 @@ class Trainer:
             rank=ctx.rank, seed=cfg.seed)
         self.step_count = 0
         self._t_last = None
+        self._graph = None
+        self._graph_state = None
+        if cfg.hip_graph:
+            assert cfg.grad_clip == 0, "hip_graph requires grad_clip=0"
+            assert self.device.type == "cuda", "hip_graph needs a GPU"
 
     @property
     def module(self) -> LlamaModel:
@@ -81,6 +88,8 @@ class Trainer:
         sync=True returns the host loss (device sync); sync=False returns
         the detached loss tensor without host synchronisation (bench path).
         """
+        if self.cfg.hip_graph:
+            return self._train_step_graph(sync)
         t0 = time.perf_counter()
         self.fb.zero_grads()
         inp, lab = self.data.batch(self.step_count)
@@ -94,6 +103,60 @@ class Trainer:
             coef = torch.clamp(coef, max=1.0)
             scale = scale * coef.item()
         self.opt.step(grad_scale=scale)
+        self.step_count += 1
+        if not sync:
+            return loss.detach()
+        out = loss.detach().float().item()
+        self._t_last = time.perf_counter() - t0
+        if self.cfg.metrics_path and self.ctx.is_main:
+            self._write_metrics(out)
+        return out
+
+    # ---- hipGraph step capture (MI355X: launch-bound inner loop ->
+    # one graph replay per step; the AdamW bias correction reads the
+    # optimizer's on-device step counter so replay stays correct) ------
+    def _capture_graph(self):
+        cfg = self.cfg
+        inp = torch.zeros(cfg.micro_batch, cfg.seq_len, dtype=torch.long,
+                          device=self.device)
+        lab = torch.zeros_like(inp)
+
+        def step_body():
+            self.fb.zero_grads()
+            loss = self.fb(inp, lab)
+            loss.backward()
+            self.fb.finish_grad_sync()
+            self.opt.step()
+            return loss
+
+        # Warm up allocator/RCCL/hook state on a side stream. These are
+        # REAL training steps (real batches) so no model state is wasted.
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):
+                ni, nl = self.data.batch(self.step_count)
+                inp.copy_(ni)
+                lab.copy_(nl)
+                step_body()
+                self.step_count += 1
+        torch.cuda.current_stream().wait_stream(side)
+        # Capture records without executing; buffer contents don't matter.
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            loss = step_body()
+        self._graph = g
+        self._graph_state = (inp, lab, loss)
+
+    def _train_step_graph(self, sync: bool):
+        t0 = time.perf_counter()
+        if self._graph is None:
+            self._capture_graph()
+        inp, lab, loss = self._graph_state
+        ni, nl = self.data.batch(self.step_count)
+        inp.copy_(ni)
+        lab.copy_(nl)
+        self._graph.replay()
         self.step_count += 1
         if not sync:
             return loss.detach()

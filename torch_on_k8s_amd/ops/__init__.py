@@ -123,11 +123,17 @@ def apply_rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.T
 def fused_adamw_(p: torch.Tensor, g: torch.Tensor, m: torch.Tensor,
                  v: torch.Tensor, *, lr: float, beta1: float, beta2: float,
                  eps: float, weight_decay: float, step: int,
-                 grad_scale: float = 1.0) -> None:
-    """In-place AdamW over one flat bucket (p/g bf16, m/v fp32)."""
+                 grad_scale: float = 1.0,
+                 step_dev: torch.Tensor | None = None) -> None:
+    """In-place AdamW over one flat bucket (p/g bf16, m/v fp32).
+
+    step_dev: optional int32 device scalar overriding `step` — used under
+    hipGraph replay, where host-side scalars are frozen into the graph.
+    """
     if p.is_cuda:
         _require_ext("fused_adamw").adamw_(p, g, m, v, lr, beta1, beta2, eps,
-                                           weight_decay, step, grad_scale)
+                                           weight_decay, step, grad_scale,
+                                           step_dev)
         return
     gf = g.float() * grad_scale
     pf = p.float()

@@ -16,7 +16,15 @@ __global__ void adamw_kernel(bf16x8* __restrict__ p, const bf16x8* __restrict__ 
                              f32x4v* __restrict__ m, f32x4v* __restrict__ v,
                              long nvec, float lr, float beta1, float beta2,
                              float eps, float wd, float bc1, float bc2,
-                             float gscale) {
+                             float gscale, const int* __restrict__ step_dev) {
+  if (step_dev != nullptr) {
+    // hipGraph-replay mode: the step counter lives on-device (a host
+    // scalar would be frozen into the captured graph), so the bias
+    // corrections are computed here.
+    const float t = (float)*step_dev;
+    bc1 = 1.f / (1.f - __powf(beta1, t));
+    bc2 = 1.f / (1.f - __powf(beta2, t));
+  }
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
     bf16x8 pv = p[i];
@@ -55,7 +63,8 @@ extern "C" {
 // buckets to 8 elements).
 hipError_t tok_adamw(void* p, const void* g, float* m, float* v, long n,
                      float lr, float beta1, float beta2, float eps, float wd,
-                     int step, float gscale, hipStream_t stream) {
+                     int step, float gscale, const int* step_dev,
+                     hipStream_t stream) {
   const long nvec = n / 8;
   const float bc1 = 1.f / (1.f - powf(beta1, (float)step));
   const float bc2 = 1.f / (1.f - powf(beta2, (float)step));
@@ -64,7 +73,7 @@ hipError_t tok_adamw(void* p, const void* g, float* m, float* v, long n,
   if (grid < 1) grid = 1;
   adamw_kernel<<<(int)grid, BLOCK, 0, stream>>>(
       (bf16x8*)p, (const bf16x8*)g, (f32x4v*)m, (f32x4v*)v, nvec, lr, beta1,
-      beta2, eps, wd, bc1, bc2, gscale);
+      beta2, eps, wd, bc1, bc2, gscale, step_dev);
   return hipGetLastError();
 }
 }

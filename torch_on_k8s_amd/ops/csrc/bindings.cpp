@@ -18,7 +18,8 @@ hipError_t tok_rope(const void* x, void* y, const float* cos_tab,
                     long table_rows, float sign, hipStream_t stream);
 hipError_t tok_adamw(void* p, const void* g, float* m, float* v, long n,
                      float lr, float beta1, float beta2, float eps, float wd,
-                     int step, float gscale, hipStream_t stream);
+                     int step, float gscale, const int* step_dev,
+                     hipStream_t stream);
 hipError_t tok_mfma_probe_16x16x32(const void* A, const void* B, float* D,
                                    hipStream_t stream);
 hipError_t tok_attn_fwd(const void* q, const void* k, const void* v, void* o,
@@ -100,17 +101,22 @@ at::Tensor rope(at::Tensor x, at::Tensor cos_tab, at::Tensor sin_tab,
 
 void adamw_(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double lr,
             double beta1, double beta2, double eps, double wd, long step,
-            double gscale) {
+            double gscale, c10::optional<at::Tensor> step_dev) {
   CHECK_BF16_CUDA(p);
   CHECK_BF16_CUDA(g);
   TORCH_CHECK(m.scalar_type() == at::kFloat && v.scalar_type() == at::kFloat);
   const long n = p.numel();
   TORCH_CHECK(n % 8 == 0, "bucket size must be a multiple of 8");
   TORCH_CHECK(g.numel() == n && m.numel() == n && v.numel() == n);
+  const int* sd = nullptr;
+  if (step_dev.has_value()) {
+    TORCH_CHECK(step_dev->scalar_type() == at::kInt && step_dev->is_cuda());
+    sd = step_dev->data_ptr<int>();
+  }
   TOK_HIP_OK(tok_adamw(p.data_ptr(), g.data_ptr(), m.data_ptr<float>(),
                        v.data_ptr<float>(), n, (float)lr, (float)beta1,
                        (float)beta2, (float)eps, (float)wd, (int)step,
-                       (float)gscale, current_stream()));
+                       (float)gscale, sd, current_stream()));
 }
 
 // q: [B,S,Hq,D], k/v: [B,S,Hkv,D] bf16 contiguous -> (o, lse[B,Hq,S] f32)
@@ -171,5 +177,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16, gfx950)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (bf16, gfx950)");
   mod.def("rope", &rope, "Rotary embedding rotate-half (bf16, gfx950)");
-  mod.def("adamw_", &adamw_, "Fused AdamW on a flat bucket (gfx950)");
+  mod.def("adamw_", &adamw_, "Fused AdamW on a flat bucket (gfx950)",
+          py::arg("p"), py::arg("g"), py::arg("m"), py::arg("v"),
+          py::arg("lr"), py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
+          py::arg("wd"), py::arg("step"), py::arg("gscale"),
+          py::arg("step_dev") = py::none());
 }

@@ -184,16 +184,23 @@ class FlatAdamW:
         self.step_count = 0
         self.exp_avg = []
         self.exp_avg_sq = []
+        dev = fb.buckets[0].flat_param.device if fb.buckets else "cpu"
         for b in fb.buckets:
             self.exp_avg.append(torch.zeros(
                 b.numel, dtype=torch.float32, device=b.flat_param.device))
             self.exp_avg_sq.append(torch.zeros(
                 b.numel, dtype=torch.float32, device=b.flat_param.device))
+        # device-side step counter: under hipGraph replay the host step
+        # is frozen into the capture, so the kernel reads this instead
+        self.step_dev = (torch.zeros((), dtype=torch.int32, device=dev)
+                         if str(dev).startswith("cuda") else None)
 
     @torch.no_grad()
     def step(self, grad_scale: float | None = None, lr: float | None = None):
         from torch_on_k8s_amd import ops
         self.step_count += 1
+        if self.step_dev is not None:
+            self.step_dev += 1  # capturable; source of truth under replay
         if grad_scale is None:
             grad_scale = 1.0 / self.fb.world_size
         for i, b in enumerate(self.fb.buckets):
@@ -203,9 +210,13 @@ class FlatAdamW:
                 lr=(lr if lr is not None else self.lr),
                 beta1=self.betas[0], beta2=self.betas[1], eps=self.eps,
                 weight_decay=self.weight_decay if b.decay else 0.0,
-                step=self.step_count, grad_scale=grad_scale)
+                step=self.step_count, grad_scale=grad_scale,
+                step_dev=self.step_dev)
 
     def state_dict(self):
+        if self.step_dev is not None:
+            # under hipGraph replay the device counter is authoritative
+            self.step_count = int(self.step_dev.item())
         return {
             "step": self.step_count,
             "exp_avg": self.exp_avg,
@@ -214,6 +225,8 @@ class FlatAdamW:
 
     def load_state_dict(self, sd):
         self.step_count = sd["step"]
+        if self.step_dev is not None:
+            self.step_dev.fill_(self.step_count)
         for dst, src in zip(self.exp_avg, sd["exp_avg"]):
             dst.copy_(src)
         for dst, src in zip(self.exp_avg_sq, sd["exp_avg_sq"]):
