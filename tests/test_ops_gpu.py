@@ -31,6 +31,16 @@ def test_mfma_16x16x32_layout():
     assert err < 1e-2, f"MFMA layout mismatch, max err {err}\nD={D.cpu()}\nref={ref}"
 
 
+def test_mfma_32x32x16_layout():
+    torch.manual_seed(1)
+    A = (torch.randn(32, 16) * 0.5).bfloat16().to(dev())
+    B = (torch.randn(16, 32) * 0.5).bfloat16().to(dev())
+    D = ops._C.mfma_probe(A, B)
+    ref = A.float().cpu() @ B.float().cpu()
+    err = (D.cpu() - ref).abs().max().item()
+    assert err < 1e-2, f"32x32x16 layout mismatch, max err {err}"
+
+
 @pytest.mark.parametrize("shape", [(32, 256), (1024, 4096), (33, 1024)])
 def test_rmsnorm_fwd_bwd(shape):
     torch.manual_seed(0)

@@ -113,7 +113,202 @@ DEVINL void load_afrags(bf16x8v* frag, const bf16_t* src, long row,
 }
 
 // ========================================================================
-// Forward: grid (ceil(S/128), Hq, B); block = 8 waves, wave = 16 q rows.
+// Forward v3: grid (ceil(S/256), Hq, B); block = 8 waves, wave = 32 q rows.
+//
+// Swapped QK^T on mfma_f32_32x32x16_bf16: compute S^T = K·Q^T so each
+// lane owns ONE q row (lane&31) and its P values stay in-register —
+// row reduce is an in-lane chain + one half-wave exchange; no P LDS
+// round trip. P is repacked to the PV A-fragment layout with
+// v_cvt_pk_bf16_f32 pairs + permlane32_swap (guide T12):
+//   C-layout kv of reg r (half h): (r&3) + 8*(r>>2) + 4*h
+//   A-layout needs kv = h*8 + j in each 16-slot; one swap fixes two words.
+// Q lives in registers as the MFMA B-operand (B[k][n]: n = lane&31 = its
+// q row, k contiguous per half) — loaded once per wave.
+// ========================================================================
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+DEVINL uint32_t pack_bf16(float lo, float hi) {
+  return (uint32_t)f2bfbits(lo) | ((uint32_t)f2bfbits(hi) << 16);
+}
+
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v, bf16_t* __restrict__ o,
+    float* __restrict__ lse, int B, int S, int Hq, int Hkv, float scale) {
+  constexpr int QC = D / 16;    // Q B-fragments (k-slots of 16)
+  constexpr int DT = D / 32;    // O col tiles of 32
+  constexpr int BM3 = NW * 32;  // 256 q rows per block
+  constexpr int KB = BN * D * 2;
+  __shared__ __attribute__((aligned(16))) char smem[2 * KB];
+  char* k_lds = smem;
+  char* vt_lds = smem + KB;
+
+  const int lane = threadIdx.x & 63;
+  const int hi = lane >> 5;
+  const int wid = threadIdx.x >> 6;
+  const int m0 = blockIdx.x * BM3;
+  const int m0w = m0 + wid * 32;
+  const int hq = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hkv = hq / (Hq / Hkv);
+
+  const long q_tok = (long)Hq * D, kv_tok = (long)Hkv * D;
+  const bf16_t* qp = q + ((long)b * S * q_tok) + (long)hq * D;
+  const bf16_t* kp = k + ((long)b * S * kv_tok) + (long)hkv * D;
+  const bf16_t* vp = v + ((long)b * S * kv_tok) + (long)hkv * D;
+  bf16_t* op = o + ((long)b * S * q_tok) + (long)hq * D;
+  float* lsep = lse + ((long)b * Hq + hq) * S;
+
+  const int qrow = m0w + (lane & 31);
+  bf16x8v q_reg[QC];
+#pragma unroll
+  for (int c = 0; c < QC; ++c) {
+    uint4 raw = {0, 0, 0, 0};
+    if (qrow < S)
+      raw = *(const uint4*)(qp + (long)qrow * q_tok + c * 16 + hi * 8);
+    q_reg[c] = as_frag(raw);
+  }
+
+  f32x16 o_acc[DT];
+#pragma unroll
+  for (int t = 0; t < DT; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) o_acc[t][r] = 0.f;
+  float m_run = NEG_INF, l_run = 0.f;
+
+  TileStage<D> k_st, v_st;
+  const int n_end = CAUSAL ? min(S, m0 + BM3) : S;
+  k_st.issue(kp, 0, S, kv_tok);
+  v_st.issue(vp, 0, S, kv_tok);
+
+  for (int n0 = 0; n0 < n_end; n0 += BN) {
+    k_st.write_rm(k_lds);
+    v_st.write_tr(vt_lds);
+    __syncthreads();
+    if (n0 + BN < n_end) {
+      k_st.issue(kp, n0 + BN, S, kv_tok);
+      v_st.issue(vp, n0 + BN, S, kv_tok);
+    }
+
+    const bool strip_live = !CAUSAL || (n0 <= m0w + 31);
+    if (strip_live) {
+      // S^T strips (two 32-kv subtiles x 32 q cols)
+      f32x16 st[2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) st[ks][r] = 0.f;
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+        for (int c = 0; c < QC; ++c) {
+          bf16x8v ka = read_bfrag<D * 2>(
+              k_lds, ks * 32 + (lane & 31), c * 16 + hi * 8);
+          st[ks] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              ka, q_reg[c], st[ks], 0, 0, 0);
+        }
+
+      // mask + scale + in-lane row max
+      float mx = NEG_INF;
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kvg = n0 + ks * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float s = st[ks][r] * scale;
+          if ((CAUSAL && kvg > qrow) || kvg >= S || qrow >= S) s = NEG_INF;
+          st[ks][r] = s;
+          mx = fmaxf(mx, s);
+        }
+      mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+      const float mn = fmaxf(m_run, mx);
+      const float alpha = (m_run == NEG_INF) ? 0.f : __expf(m_run - mn);
+      m_run = mn;
+      float rsum = 0.f;
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const float pe =
+              (st[ks][r] == NEG_INF) ? 0.f : __expf(st[ks][r] - mn);
+          st[ks][r] = pe;
+          rsum += pe;
+        }
+      rsum += __shfl_xor(rsum, 32, 64);
+      l_run = l_run * alpha + rsum;
+
+      // rescale O by the per-ROW alpha (rows are reg-mapped; alpha is
+      // lane-mapped -> one bpermute gather per reg row)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int rowidx = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const float ar = __shfl(alpha, rowidx, 64);
+#pragma unroll
+        for (int t = 0; t < DT; ++t) o_acc[t][r] *= ar;
+      }
+
+      // P (f32, C-layout) -> bf16 A-fragments via cvt_pk + permlane swap
+      bf16x8v pa[4];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        union { uint32_t w[4]; bf16x8v f; } f0, f1;
+        {
+          auto sA = __builtin_amdgcn_permlane32_swap(
+              pack_bf16(st[ks][0], st[ks][1]),
+              pack_bf16(st[ks][4], st[ks][5]), false, false);
+          auto sB = __builtin_amdgcn_permlane32_swap(
+              pack_bf16(st[ks][2], st[ks][3]),
+              pack_bf16(st[ks][6], st[ks][7]), false, false);
+          f0.w[0] = sA[0]; f0.w[1] = sB[0]; f0.w[2] = sA[1]; f0.w[3] = sB[1];
+        }
+        {
+          auto sC = __builtin_amdgcn_permlane32_swap(
+              pack_bf16(st[ks][8], st[ks][9]),
+              pack_bf16(st[ks][12], st[ks][13]), false, false);
+          auto sD = __builtin_amdgcn_permlane32_swap(
+              pack_bf16(st[ks][10], st[ks][11]),
+              pack_bf16(st[ks][14], st[ks][15]), false, false);
+          f1.w[0] = sC[0]; f1.w[1] = sD[0]; f1.w[2] = sC[1]; f1.w[3] = sD[1];
+        }
+        pa[2 * ks] = f0.f;
+        pa[2 * ks + 1] = f1.f;
+      }
+
+      // O += P * V
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks)
+#pragma unroll
+        for (int t = 0; t < DT; ++t) {
+          bf16x8v vb = read_bfrag<128>(
+              vt_lds, t * 32 + (lane & 31), ks * 16 + hi * 8);
+          o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pa[ks], vb, o_acc[t], 0, 0, 0);
+        }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: O = o_acc / l (per reg row); LSE per lane row
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int rowidx = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const float lv = __shfl(l_run, rowidx, 64);
+    const float inv_l = (lv > 0.f) ? 1.f / lv : 0.f;
+    const int row = m0w + rowidx;
+    if (row >= S) continue;
+#pragma unroll
+    for (int t = 0; t < DT; ++t) {
+      const int col = t * 32 + (lane & 31);
+      op[(long)row * q_tok + col] = f2bf(o_acc[t][r] * inv_l);
+    }
+  }
+  if (hi == 0 && qrow < S)
+    lsep[qrow] = (l_run > 0.f) ? m_run + __logf(l_run) : NEG_INF;
+}
+
+// ========================================================================
+// Forward (16x16 variant, kept for reference/ablation; v3 is dispatched)
 // ========================================================================
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
@@ -637,10 +832,10 @@ extern "C" {
 hipError_t tok_attn_fwd(const void* q, const void* k, const void* v, void* o,
                         float* lse, int B, int S, int Hq, int Hkv, int D,
                         int causal, hipStream_t stream) {
-  dim3 grid((S + BM - 1) / BM, Hq, B);
+  dim3 grid((S + NW * 32 - 1) / (NW * 32), Hq, B);
   const float scale = 1.f / sqrtf((float)D);
 #define LAUNCH_FWD(DD, CC)                                                    \
-  attn_fwd_kernel<DD, CC><<<grid, NTHREADS, 0, stream>>>(                     \
+  attn_fwd_kernel_v3<DD, CC><<<grid, NTHREADS, 0, stream>>>(                  \
       (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (bf16_t*)o, lse, \
       B, S, Hq, Hkv, scale)
   if (D == 128) { if (causal) LAUNCH_FWD(128, true); else LAUNCH_FWD(128, false); }

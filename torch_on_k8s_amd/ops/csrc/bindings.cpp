@@ -22,6 +22,8 @@ hipError_t tok_adamw(void* p, const void* g, float* m, float* v, long n,
                      hipStream_t stream);
 hipError_t tok_mfma_probe_16x16x32(const void* A, const void* B, float* D,
                                    hipStream_t stream);
+hipError_t tok_mfma_probe_32x32x16(const void* A, const void* B, float* D,
+                                   hipStream_t stream);
 hipError_t tok_attn_fwd(const void* q, const void* k, const void* v, void* o,
                         float* lse, int B, int S, int Hq, int Hkv, int D,
                         int causal, hipStream_t stream);
@@ -160,10 +162,18 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
 at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
   CHECK_BF16_CUDA(A);
   CHECK_BF16_CUDA(B);
-  TORCH_CHECK(A.sizes() == at::IntArrayRef({16, 32}) &&
-              B.sizes() == at::IntArrayRef({32, 16}));
-  auto D = at::empty({16, 16}, A.options().dtype(at::kFloat));
-  TOK_HIP_OK(tok_mfma_probe_16x16x32(A.data_ptr(), B.data_ptr(),
+  if (A.size(0) == 16) {
+    TORCH_CHECK(A.sizes() == at::IntArrayRef({16, 32}) &&
+                B.sizes() == at::IntArrayRef({32, 16}));
+    auto D = at::empty({16, 16}, A.options().dtype(at::kFloat));
+    TOK_HIP_OK(tok_mfma_probe_16x16x32(A.data_ptr(), B.data_ptr(),
+                                       D.data_ptr<float>(), current_stream()));
+    return D;
+  }
+  TORCH_CHECK(A.sizes() == at::IntArrayRef({32, 16}) &&
+              B.sizes() == at::IntArrayRef({16, 32}));
+  auto D = at::empty({32, 32}, A.options().dtype(at::kFloat));
+  TOK_HIP_OK(tok_mfma_probe_32x32x16(A.data_ptr(), B.data_ptr(),
                                      D.data_ptr<float>(), current_stream()));
   return D;
 }
