@@ -167,9 +167,15 @@ class JobController:
             if self.elastic.reconcile(self, job, hs):
                 return  # elastic transaction in flight; hold other actions
 
-        # gang admission before any task creation (job.go:214-219)
+        # gang admission before any task creation (job.go:214-219);
+        # preempt lower-priority spot tasks if that unblocks the gang
+        # (SpotTaskSpec semantics, torchjob_types.go:50-61 + the
+        # preempt-protector intent: preemption is a graceful SIGTERM, and
+        # the trainer checkpoints on SIGTERM before exiting)
         if self.gang is not None:
             self.gang.create_pod_group(job)
+            if not hs and not self.gang.can_admit(job):
+                self._preempt_spot_for(job)
             if not hs and not self.gang.can_admit(job):
                 self.event(job.name, "Normal", "GangNotAdmitted",
                            "waiting for gang quota")
@@ -218,6 +224,9 @@ class JobController:
             if need else ()
         extra_env = {"MASTER_PORT": str(self._master_port(job))}
         h = self.runtime.start_task(job, t, idx, slots, extra_env)
+        if spec.spot is not None and \
+                idx >= spec.replicas - spec.spot.num_spot_replicas:
+            h.spot = True
         hs[h.key] = h
         if job.name not in self._first_task_ts:
             self._first_task_ts[job.name] = time.time()
@@ -262,6 +271,46 @@ class JobController:
         nh = hs.get(h.key)
         if nh:
             nh.restart_count = rc
+
+    def _preempt_spot_for(self, job: TorchJob):
+        """Free GPU slots by gracefully killing spot replicas of
+        lower-priority jobs (spot-task priority overlay, pod.go:592-603)."""
+        pg = self.gang.groups.get(job.name)
+        if pg is None:
+            return
+        need = pg.min_gpus - len(self.node.free_slots)
+        if need <= 0:
+            return
+        my_prio = job.scheduling.priority or 0
+        victims = []
+        for other, ohs in self.handles.items():
+            if other == job.name:
+                continue
+            oprio = (self.jobs[other].scheduling.priority or 0) \
+                if other in self.jobs else 0
+            for h in ohs.values():
+                if h.spot and not h.finished and oprio < my_prio:
+                    victims.append((oprio, h, other))
+        victims.sort(key=lambda x: x[0])
+        freed = 0
+        for _, h, other in victims:
+            if freed >= need:
+                break
+            self.runtime.kill(h, grace=True)  # trainer checkpoints on TERM
+            freed += len(h.gpu_slots)
+            if h.gpu_slots:
+                self.node.release(h.gpu_slots)
+                h.gpu_slots = ()
+            # shrink the victim's spot replica count so its controller
+            # does not immediately recreate the preempted replica
+            ospec = self.jobs[other].tasks.get(h.task_type)
+            if ospec is not None and ospec.spot is not None:
+                ospec.replicas = max(0, ospec.replicas - 1)
+                ospec.spot.num_spot_replicas = max(
+                    0, ospec.spot.num_spot_replicas - 1)
+            self.handles[other].pop(h.key, None)
+            self.event(job.name, "Normal", "SpotPreempted",
+                       f"preempted {h.key} of {other}")
 
     # ------------------------------------------------------------------
     def _check_terminated(self, job: TorchJob, hs: dict) -> bool:

@@ -33,6 +33,7 @@ class TaskHandle:
     annotations: dict = field(default_factory=dict)
     start_time: float = field(default_factory=time.time)
     restart_count: int = 0
+    spot: bool = False   # preemptible overflow replica (SpotTaskSpec)
     proc: object = None  # subprocess.Popen for the local runtime
 
     @property

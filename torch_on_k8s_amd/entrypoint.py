@@ -45,7 +45,33 @@ def _read_json(path: str):
         return None
 
 
+def aimaster_main() -> int:
+    """AIMaster task mode: a supervision sidecar (the reference models
+    AIMaster as an external image driving the elastic annotations; here
+    the heavy lifting lives in the manager's autoscaler + the rank-0
+    checkpoint agent, and this process observes/journals training
+    health). Exits 0 on SIGTERM."""
+    import signal as _signal
+    state_dir = os.environ.get("TOK_STATE_DIR", "")
+    stop = {"flag": False}
+    _signal.signal(_signal.SIGTERM, lambda *_: stop.update(flag=True))
+    journal = os.path.join(state_dir, "aimaster.log") if state_dir else None
+    last_step = -1
+    while not stop["flag"]:
+        rec = _read_json(os.path.join(state_dir, "metrics.json")) \
+            if state_dir else None
+        if rec and rec.get("step") != last_step:
+            last_step = rec["step"]
+            if journal:
+                with open(journal, "a") as f:
+                    f.write(json.dumps(rec) + "\n")
+        time.sleep(1.0)
+    return 0
+
+
 def main() -> int:
+    if os.environ.get("TOK_TASK_TYPE") == "aimaster":
+        return aimaster_main()
     import torch
     from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
     from torch_on_k8s_amd.parallel.env import init_distributed, destroy, barrier

@@ -49,17 +49,17 @@ DEVINL bf16x8v as_frag(uint4 raw) {
 // A [64][D] bf16 tile staged by all 512 threads: issue() starts the
 // global loads into registers; write_rm()/write_tr() put them into LDS
 // (row-major swizzled / transposed swizzled). OOB rows load zero.
-template <int D>
+template <int D, int NTH = NTHREADS>
 struct TileStage {
   static constexpr int VPR = D / 8;              // 16B vecs per row
-  static constexpr int VPT = 64 * VPR / NTHREADS;  // vecs per thread
+  static constexpr int VPT = 64 * VPR / NTH;     // vecs per thread
   uint4 vals[VPT];
 
   DEVINL void issue(const bf16_t* src, long row0, long row_limit,
                     long tok_stride) {
 #pragma unroll
     for (int i = 0; i < VPT; ++i) {
-      const int vi = threadIdx.x + i * NTHREADS;
+      const int vi = threadIdx.x + i * NTH;
       const int row = vi / VPR, cv = vi % VPR;
       uint4 val = {0, 0, 0, 0};
       if (row0 + row < row_limit)
@@ -71,7 +71,7 @@ struct TileStage {
   DEVINL void write_rm(char* lds) const {
 #pragma unroll
     for (int i = 0; i < VPT; ++i) {
-      const int vi = threadIdx.x + i * NTHREADS;
+      const int vi = threadIdx.x + i * NTH;
       const int row = vi / VPR, cv = vi % VPR;
       *(uint4*)(lds + row * (D * 2) + swz(row, cv * 16)) = vals[i];
     }
@@ -80,7 +80,7 @@ struct TileStage {
   DEVINL void write_tr(char* lds) const {
 #pragma unroll
     for (int i = 0; i < VPT; ++i) {
-      const int vi = threadIdx.x + i * NTHREADS;
+      const int vi = threadIdx.x + i * NTH;
       const int row = vi / VPR, cv = vi % VPR;
       const uint16_t* h = (const uint16_t*)&vals[i];
 #pragma unroll
@@ -506,29 +506,177 @@ __global__ void attn_bwd_pre_kernel(const bf16_t* __restrict__ dout,
 }
 
 // ========================================================================
-// Backward dK/dV: grid (ceil(S/128), Hkv, B); block = 8 waves, wave =
-// 16 kv rows; loops 64-row Q tiles and the GQA q-head group.
+// Backward dV / dK v3: two kernels, grid (ceil(S/256), Hkv, B), 8 waves,
+// wave = 32 kv rows. Splitting dV and dK keeps each kernel at 2
+// waves/SIMD (a combined kernel needs dk+dv accumulators = 128 regs and
+// drops to 1 wave/SIMD, which measured SLOWER than the 16x16 v2 —
+// occupancy beats the saved recompute here). Swapped-operand structure
+// as the forward: K (and V in the dK kernel) live in registers as MFMA
+// B-operands; P^T / dS^T stay in-register and are repacked with
+// cvt_pk+permlane32_swap.
 // ========================================================================
+DEVINL void repack_pa(const f32x16& pt, bf16x8v& pa0, bf16x8v& pa1) {
+  union { uint32_t w[4]; bf16x8v f; } f0, f1;
+  auto sA = __builtin_amdgcn_permlane32_swap(
+      pack_bf16(pt[0], pt[1]), pack_bf16(pt[4], pt[5]), false, false);
+  auto sB = __builtin_amdgcn_permlane32_swap(
+      pack_bf16(pt[2], pt[3]), pack_bf16(pt[6], pt[7]), false, false);
+  f0.w[0] = sA[0]; f0.w[1] = sB[0]; f0.w[2] = sA[1]; f0.w[3] = sB[1];
+  auto sC = __builtin_amdgcn_permlane32_swap(
+      pack_bf16(pt[8], pt[9]), pack_bf16(pt[12], pt[13]), false, false);
+  auto sD = __builtin_amdgcn_permlane32_swap(
+      pack_bf16(pt[10], pt[11]), pack_bf16(pt[14], pt[15]), false, false);
+  f1.w[0] = sC[0]; f1.w[1] = sD[0]; f1.w[2] = sC[1]; f1.w[3] = sD[1];
+  pa0 = f0.f;
+  pa1 = f1.f;
+}
+
 template <int D, bool CAUSAL>
-__global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
+__global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ dsum,
-    bf16_t* __restrict__ dk, bf16_t* __restrict__ dv, int B, int S, int Hq,
-    int Hkv, float scale) {
-  constexpr int DC = D / 32;
-  constexpr int DT = D / 16;
+    bf16_t* __restrict__ dv, int B, int S, int Hq, int Hkv, float scale) {
+  constexpr int QC = D / 16;
+  constexpr int DT = D / 32;
+  constexpr int BNK = NW * 32;   // 256 kv rows per block
   constexpr int KB = BN * D * 2;
-  __shared__ __attribute__((aligned(16))) char smem[4 * KB + NW * 2048];
+  __shared__ __attribute__((aligned(16))) char smem[2 * KB];  // Q rm | dOT
+  char* q_lds = smem;
+  char* dot_lds = smem + KB;
+
+  const int lane = threadIdx.x & 63;
+  const int hi = lane >> 5;
+  const int wid = threadIdx.x >> 6;
+  const int n0 = blockIdx.x * BNK;
+  const int n0w = n0 + wid * 32;
+  const int hkv = blockIdx.y;
+  const int b = blockIdx.z;
+  const int rep = Hq / Hkv;
+
+  const long q_tok = (long)Hq * D, kv_tok = (long)Hkv * D;
+  const bf16_t* kp = k + ((long)b * S * kv_tok) + (long)hkv * D;
+  bf16_t* dvp = dv + ((long)b * S * kv_tok) + (long)hkv * D;
+
+  const int kvrow = n0w + (lane & 31);
+  bf16x8v k_reg[QC];
+#pragma unroll
+  for (int c = 0; c < QC; ++c) {
+    uint4 kr = {0, 0, 0, 0};
+    if (kvrow < S)
+      kr = *(const uint4*)(kp + (long)kvrow * kv_tok + c * 16 + hi * 8);
+    k_reg[c] = as_frag(kr);
+  }
+
+  f32x16 dv_acc[DT];
+#pragma unroll
+  for (int t = 0; t < DT; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dv_acc[t][r] = 0.f;
+
+  const int m_start = CAUSAL ? (n0 / BN) * BN : 0;
+  TileStage<D> q_st, do_st;
+  q_st.issue(q + ((long)b * S * q_tok) + (long)(hkv * rep) * D, m_start, S,
+             q_tok);
+  do_st.issue(dout + ((long)b * S * q_tok) + (long)(hkv * rep) * D, m_start,
+              S, q_tok);
+
+  for (int g = 0; g < rep; ++g) {
+    const int hq = hkv * rep + g;
+    const bf16_t* qp = q + ((long)b * S * q_tok) + (long)hq * D;
+    const bf16_t* dop = dout + ((long)b * S * q_tok) + (long)hq * D;
+    const float* lsep = lse + ((long)b * Hq + hq) * S;
+
+    for (int m0 = m_start; m0 < S; m0 += BN) {
+      q_st.write_rm(q_lds);
+      do_st.write_tr(dot_lds);
+      __syncthreads();
+      const int m1 = m0 + BN;
+      if (m1 < S) {
+        q_st.issue(qp, m1, S, q_tok);
+        do_st.issue(dop, m1, S, q_tok);
+      } else if (g + 1 < rep) {
+        q_st.issue(qp + D, m_start, S, q_tok);
+        do_st.issue(dop + D, m_start, S, q_tok);
+      }
+
+#pragma unroll
+      for (int qs = 0; qs < 2; ++qs) {
+        f32x16 sv;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) sv[r] = 0.f;
+#pragma unroll
+        for (int c = 0; c < QC; ++c) {
+          bf16x8v qa = read_bfrag<D * 2>(
+              q_lds, qs * 32 + (lane & 31), c * 16 + hi * 8);
+          sv = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              qa, k_reg[c], sv, 0, 0, 0);
+        }
+        const int qg_lane = m0 + qs * 32 + (lane & 31);
+        const float lse_lane = (qg_lane < S) ? lsep[qg_lane] : NEG_INF;
+        f32x16 pt;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int rowidx = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          const int qg = m0 + qs * 32 + rowidx;
+          const float lse_r = __shfl(lse_lane, rowidx, 64);
+          float pv = 0.f;
+          if (qg < S && kvrow < S && (!CAUSAL || qg >= kvrow) &&
+              lse_r != NEG_INF)
+            pv = __expf(sv[r] * scale - lse_r);
+          pt[r] = pv;
+        }
+        bf16x8v pa0, pa1;
+        repack_pa(pt, pa0, pa1);
+#pragma unroll
+        for (int t = 0; t < DT; ++t) {
+          bf16x8v b0 = read_bfrag<128>(
+              dot_lds, t * 32 + (lane & 31), qs * 32 + hi * 8);
+          dv_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pa0, b0, dv_acc[t], 0, 0, 0);
+          bf16x8v b1 = read_bfrag<128>(
+              dot_lds, t * 32 + (lane & 31), qs * 32 + 16 + hi * 8);
+          dv_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pa1, b1, dv_acc[t], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = n0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
+    if (row >= S) continue;
+#pragma unroll
+    for (int t = 0; t < DT; ++t)
+      dvp[(long)row * kv_tok + t * 32 + (lane & 31)] = f2bf(dv_acc[t][r]);
+  }
+}
+
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ dsum,
+    bf16_t* __restrict__ dk, int B, int S, int Hq, int Hkv, float scale) {
+  constexpr int QC = D / 16;
+  constexpr int DT = D / 32;
+  constexpr int BNK = NW * 32;
+  constexpr int KB = BN * D * 2;
+  // Q rm | QT | dO rm | V block tile (staged once; keeping V in
+  // registers alongside K pushed the kernel to 256 VGPR + scratch spill)
+  __shared__ __attribute__((aligned(16))) char smem[3 * KB + BNK * D * 2];
   char* q_lds = smem;
   char* qt_lds = smem + KB;
   char* do_lds = smem + 2 * KB;
-  char* dot_lds = smem + 3 * KB;
-  char* p_lds = smem + 4 * KB;
+  char* v_lds = smem + 3 * KB;
 
   const int lane = threadIdx.x & 63;
+  const int hi = lane >> 5;
   const int wid = threadIdx.x >> 6;
-  const int n0 = blockIdx.x * BM;       // 128 kv rows per block
+  const int n0 = blockIdx.x * BNK;
+  const int n0w = n0 + wid * 32;
   const int hkv = blockIdx.y;
   const int b = blockIdx.z;
   const int rep = Hq / Hkv;
@@ -537,30 +685,36 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
   const bf16_t* kp = k + ((long)b * S * kv_tok) + (long)hkv * D;
   const bf16_t* vp = v + ((long)b * S * kv_tok) + (long)hkv * D;
   bf16_t* dkp = dk + ((long)b * S * kv_tok) + (long)hkv * D;
-  bf16_t* dvp = dv + ((long)b * S * kv_tok) + (long)hkv * D;
 
-  const int nrow = n0 + wid * 16 + (lane & 15);
-  bf16x8v k_frag[DC], v_frag[DC];
-  load_afrags<D>(k_frag, kp, nrow, S, kv_tok, lane);
-  load_afrags<D>(v_frag, vp, nrow, S, kv_tok, lane);
-
-  f32x4 dk_acc[DT], dv_acc[DT];
+  const int kvrow = n0w + (lane & 31);
+  bf16x8v k_reg[QC];
 #pragma unroll
-  for (int t = 0; t < DT; ++t) {
-    dk_acc[t] = {0.f, 0.f, 0.f, 0.f};
-    dv_acc[t] = {0.f, 0.f, 0.f, 0.f};
+  for (int c = 0; c < QC; ++c) {
+    uint4 kr = {0, 0, 0, 0};
+    if (kvrow < S)
+      kr = *(const uint4*)(kp + (long)kvrow * kv_tok + c * 16 + hi * 8);
+    k_reg[c] = as_frag(kr);
+  }
+  {  // stage the block's 256-row V tile (row-major, swizzled) once
+    constexpr int VPR = D / 8;
+#pragma unroll 4
+    for (int vi = threadIdx.x; vi < BNK * VPR; vi += NTHREADS) {
+      const int row = vi / VPR, cv = vi % VPR;
+      uint4 val = {0, 0, 0, 0};
+      if (n0 + row < S)
+        val = *(const uint4*)(vp + (long)(n0 + row) * kv_tok + cv * 8);
+      *(uint4*)(v_lds + row * (D * 2) + swz(row, cv * 16)) = val;
+    }
   }
 
-  char* pw = p_lds + wid * 2048;
+  f32x16 dk_acc[DT];
+#pragma unroll
+  for (int t = 0; t < DT; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dk_acc[t][r] = 0.f;
+
   const int m_start = CAUSAL ? (n0 / BN) * BN : 0;
 
-  TileStage<D> q_st, do_st;
-  // first tile of the first q-head group; subsequent tiles (including
-  // group transitions) are prefetched inside the loop
-  q_st.issue(q + ((long)b * S * q_tok) + (long)(hkv * rep) * D, m_start, S,
-             q_tok);
-  do_st.issue(dout + ((long)b * S * q_tok) + (long)(hkv * rep) * D, m_start,
-              S, q_tok);
   for (int g = 0; g < rep; ++g) {
     const int hq = hkv * rep + g;
     const bf16_t* qp = q + ((long)b * S * q_tok) + (long)hq * D;
@@ -569,129 +723,86 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dkdv_kernel(
     const float* dsp = dsum + ((long)b * Hq + hq) * S;
 
     for (int m0 = m_start; m0 < S; m0 += BN) {
-      q_st.write_rm(q_lds);
-      q_st.write_tr(qt_lds);
-      do_st.write_rm(do_lds);
-      do_st.write_tr(dot_lds);
+      {  // synchronous staging: the prefetch ring costs ~16 VGPRs and
+         // tips this kernel into scratch spill, which is worse
+        TileStage<D> q_st, do_st;
+        q_st.issue(qp, m0, S, q_tok);
+        do_st.issue(dop, m0, S, q_tok);
+        q_st.write_rm(q_lds);
+        q_st.write_tr(qt_lds);
+        do_st.write_rm(do_lds);
+      }
       __syncthreads();
-      // prefetch the next Q/dO tile (wrap into the next q-head group)
-      const int m1 = m0 + BN;
-      if (m1 < S) {
-        q_st.issue(qp, m1, S, q_tok);
-        do_st.issue(dop, m1, S, q_tok);
-      } else if (g + 1 < rep) {
-        const bf16_t* qp2 = q + ((long)b * S * q_tok) + (long)(hq + 1) * D;
-        const bf16_t* dop2 = dout + ((long)b * S * q_tok) + (long)(hq + 1) * D;
-        q_st.issue(qp2, m_start, S, q_tok);
-        do_st.issue(dop2, m_start, S, q_tok);
-      }
 
-      // S^T / dP^T strips: rows = 16 kv, cols = 64 q
-      f32x4 st[4], dpt[4];
 #pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        st[t] = {0.f, 0.f, 0.f, 0.f};
-        dpt[t] = {0.f, 0.f, 0.f, 0.f};
-      }
+      for (int qs = 0; qs < 2; ++qs) {
+        f32x16 sv, dpv;
 #pragma unroll
-      for (int t = 0; t < 4; ++t)
-#pragma unroll
-        for (int c = 0; c < DC; ++c) {
-          bf16x8v qb = read_bfrag<D * 2>(
-              q_lds, t * 16 + (lane & 15), c * 32 + (lane >> 4) * 8);
-          st[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              k_frag[c], qb, st[t], 0, 0, 0);
-          bf16x8v dob = read_bfrag<D * 2>(
-              do_lds, t * 16 + (lane & 15), c * 32 + (lane >> 4) * 8);
-          dpt[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              v_frag[c], dob, dpt[t], 0, 0, 0);
+        for (int r = 0; r < 16; ++r) {
+          sv[r] = 0.f;
+          dpv[r] = 0.f;
         }
-
-      const int nrow_base = n0 + wid * 16 + (lane >> 4) * 4;
-      float pt[4][4], dst[4][4];
 #pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        const int mcol = m0 + t * 16 + (lane & 15);
-        const bool mvalid = mcol < S;
-        const float lse_m = mvalid ? lsep[mcol] : 0.f;
-        const float ds_m = mvalid ? dsp[mcol] : 0.f;
+        for (int c = 0; c < QC; ++c) {
+          bf16x8v qa = read_bfrag<D * 2>(
+              q_lds, qs * 32 + (lane & 31), c * 16 + hi * 8);
+          sv = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              qa, k_reg[c], sv, 0, 0, 0);
+          bf16x8v doa = read_bfrag<D * 2>(
+              do_lds, qs * 32 + (lane & 31), c * 16 + hi * 8);
+          bf16x8v vb = read_bfrag<D * 2>(
+              v_lds, wid * 32 + (lane & 31), c * 16 + hi * 8);
+          dpv = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              doa, vb, dpv, 0, 0, 0);
+        }
+        const int qg_lane = m0 + qs * 32 + (lane & 31);
+        const float lse_lane = (qg_lane < S) ? lsep[qg_lane] : NEG_INF;
+        const float ds_lane = (qg_lane < S) ? dsp[qg_lane] : 0.f;
+        f32x16 dst;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int nr = nrow_base + r;
+        for (int r = 0; r < 16; ++r) {
+          const int rowidx = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          const int qg = m0 + qs * 32 + rowidx;
+          const float lse_r = __shfl(lse_lane, rowidx, 64);
+          const float ds_r = __shfl(ds_lane, rowidx, 64);
           float pv = 0.f;
-          if (mvalid && nr < S && (!CAUSAL || mcol >= nr) &&
-              lse_m != NEG_INF)
-            pv = __expf(st[t][r] * scale - lse_m);
-          pt[t][r] = pv;
-          dst[t][r] = pv * (dpt[t][r] - ds_m);
+          if (qg < S && kvrow < S && (!CAUSAL || qg >= kvrow) &&
+              lse_r != NEG_INF)
+            pv = __expf(sv[r] * scale - lse_r);
+          dst[r] = pv * (dpv[r] - ds_r);
         }
-      }
-
-      // dV += P^T * dO
-#pragma unroll
-      for (int t = 0; t < 4; ++t)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int rl = (lane >> 4) * 4 + r;
-          const int cl = t * 16 + (lane & 15);
-          *(uint16_t*)(pw + rl * 128 + swz(rl, cl * 2)) = f2bfbits(pt[t][r]);
-        }
-#pragma unroll
-      for (int kc = 0; kc < 2; ++kc) {
-        bf16x8v pa = as_frag(*(const uint4*)(
-            pw + (lane & 15) * 128 +
-            swz(lane & 15, (kc * 32 + (lane >> 4) * 8) * 2)));
+        bf16x8v da0, da1;
+        repack_pa(dst, da0, da1);
 #pragma unroll
         for (int t = 0; t < DT; ++t) {
-          bf16x8v db = read_bfrag<128>(
-              dot_lds, t * 16 + (lane & 15), kc * 32 + (lane >> 4) * 8);
-          dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              pa, db, dv_acc[t], 0, 0, 0);
-        }
-      }
-
-      // dK += dS^T * Q (scale folded at epilogue)
-#pragma unroll
-      for (int t = 0; t < 4; ++t)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int rl = (lane >> 4) * 4 + r;
-          const int cl = t * 16 + (lane & 15);
-          *(uint16_t*)(pw + rl * 128 + swz(rl, cl * 2)) = f2bfbits(dst[t][r]);
-        }
-#pragma unroll
-      for (int kc = 0; kc < 2; ++kc) {
-        bf16x8v da = as_frag(*(const uint4*)(
-            pw + (lane & 15) * 128 +
-            swz(lane & 15, (kc * 32 + (lane >> 4) * 8) * 2)));
-#pragma unroll
-        for (int t = 0; t < DT; ++t) {
-          bf16x8v qb = read_bfrag<128>(
-              qt_lds, t * 16 + (lane & 15), kc * 32 + (lane >> 4) * 8);
-          dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              da, qb, dk_acc[t], 0, 0, 0);
+          bf16x8v b0 = read_bfrag<128>(
+              qt_lds, t * 32 + (lane & 31), qs * 32 + hi * 8);
+          dk_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              da0, b0, dk_acc[t], 0, 0, 0);
+          bf16x8v b1 = read_bfrag<128>(
+              qt_lds, t * 32 + (lane & 31), qs * 32 + 16 + hi * 8);
+          dk_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              da1, b1, dk_acc[t], 0, 0, 0);
         }
       }
       __syncthreads();
     }
   }
 
-  const int nrow_base = n0 + wid * 16 + (lane >> 4) * 4;
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int row = nrow_base + r;
+  for (int r = 0; r < 16; ++r) {
+    const int row = n0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
     if (row >= S) continue;
 #pragma unroll
-    for (int t = 0; t < DT; ++t) {
-      const int col = t * 16 + (lane & 15);
-      dkp[(long)row * kv_tok + col] = f2bf(dk_acc[t][r] * scale);
-      dvp[(long)row * kv_tok + col] = f2bf(dv_acc[t][r]);
-    }
+    for (int t = 0; t < DT; ++t)
+      dkp[(long)row * kv_tok + t * 32 + (lane & 31)] =
+          f2bf(dk_acc[t][r] * scale);
   }
 }
 
 // ========================================================================
-// Backward dQ: grid (ceil(S/128), Hq, B); block = 8 waves, wave = 16 q rows.
+// Backward dQ v3: grid (ceil(S/256), Hq, B); 8 waves, wave = 32 q rows.
+// Q/dO in registers (lane owns q row); lse/Dsum are per-lane scalars.
 // ========================================================================
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
@@ -699,18 +810,21 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ dsum,
     bf16_t* __restrict__ dq, int B, int S, int Hq, int Hkv, float scale) {
-  constexpr int DC = D / 32;
-  constexpr int DT = D / 16;
+  constexpr int QC = D / 16;
+  constexpr int DT = D / 32;
+  constexpr int BM3 = NW * 32;
   constexpr int KB = BN * D * 2;
-  __shared__ __attribute__((aligned(16))) char smem[3 * KB + NW * 2048];
+  // K rm | KT | V rm
+  __shared__ __attribute__((aligned(16))) char smem[3 * KB];
   char* k_lds = smem;
   char* kt_lds = smem + KB;
   char* v_lds = smem + 2 * KB;
-  char* p_lds = smem + 3 * KB;
 
   const int lane = threadIdx.x & 63;
+  const int hi = lane >> 5;
   const int wid = threadIdx.x >> 6;
-  const int m0 = blockIdx.x * BM;
+  const int m0 = blockIdx.x * BM3;
+  const int m0w = m0 + wid * 32;
   const int hq = blockIdx.y;
   const int b = blockIdx.z;
   const int hkv = hq / (Hq / Hkv);
@@ -724,27 +838,29 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
   const float* dsp = dsum + ((long)b * Hq + hq) * S;
   bf16_t* dqp = dq + ((long)b * S * q_tok) + (long)hq * D;
 
-  const int mrow = m0 + wid * 16 + (lane & 15);
-  bf16x8v q_frag[DC], do_frag[DC];
-  load_afrags<D>(q_frag, qp, mrow, S, q_tok, lane);
-  load_afrags<D>(do_frag, dop, mrow, S, q_tok, lane);
-
-  const int row_base = m0 + wid * 16 + (lane >> 4) * 4;
-  float lse_r[4], ds_r[4];
+  const int qrow = m0w + (lane & 31);
+  bf16x8v q_reg[QC], do_reg[QC];
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int row = row_base + r;
-    lse_r[r] = (row < S) ? lsep[row] : NEG_INF;
-    ds_r[r] = (row < S) ? dsp[row] : 0.f;
+  for (int c = 0; c < QC; ++c) {
+    uint4 qr = {0, 0, 0, 0}, dor = {0, 0, 0, 0};
+    if (qrow < S) {
+      qr = *(const uint4*)(qp + (long)qrow * q_tok + c * 16 + hi * 8);
+      dor = *(const uint4*)(dop + (long)qrow * q_tok + c * 16 + hi * 8);
+    }
+    q_reg[c] = as_frag(qr);
+    do_reg[c] = as_frag(dor);
   }
+  const float lse_lane = (qrow < S) ? lsep[qrow] : NEG_INF;
+  const float ds_lane = (qrow < S) ? dsp[qrow] : 0.f;
 
-  f32x4 dq_acc[DT];
+  f32x16 dq_acc[DT];
 #pragma unroll
-  for (int t = 0; t < DT; ++t) dq_acc[t] = {0.f, 0.f, 0.f, 0.f};
+  for (int t = 0; t < DT; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dq_acc[t][r] = 0.f;
 
-  char* pw = p_lds + wid * 2048;
-  const int n_end = CAUSAL ? min(S, m0 + BM) : S;
   TileStage<D> k_st, v_st;
+  const int n_end = CAUSAL ? min(S, m0 + BM3) : S;
   k_st.issue(kp, 0, S, kv_tok);
   v_st.issue(vp, 0, S, kv_tok);
 
@@ -758,55 +874,69 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
       v_st.issue(vp, n0 + BN, S, kv_tok);
     }
 
-    const bool strip_live = !CAUSAL || (n0 <= m0 + wid * 16 + 15);
+    const bool strip_live = !CAUSAL || (n0 <= m0w + 31);
     if (strip_live) {
-      f32x4 s_acc[4], dp_acc[4];
 #pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        s_acc[t] = {0.f, 0.f, 0.f, 0.f};
-        dp_acc[t] = {0.f, 0.f, 0.f, 0.f};
-      }
+      for (int ks = 0; ks < 2; ++ks) {   // two 32-kv subtiles
+        f32x16 sv, dpv;
 #pragma unroll
-      for (int t = 0; t < 4; ++t)
+        for (int r = 0; r < 16; ++r) {
+          sv[r] = 0.f;
+          dpv[r] = 0.f;
+        }
 #pragma unroll
-        for (int c = 0; c < DC; ++c) {
-          bf16x8v kb = read_bfrag<D * 2>(
-              k_lds, t * 16 + (lane & 15), c * 32 + (lane >> 4) * 8);
-          s_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              q_frag[c], kb, s_acc[t], 0, 0, 0);
-          bf16x8v vb = read_bfrag<D * 2>(
-              v_lds, t * 16 + (lane & 15), c * 32 + (lane >> 4) * 8);
-          dp_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              do_frag[c], vb, dp_acc[t], 0, 0, 0);
+        for (int c = 0; c < QC; ++c) {
+          bf16x8v ka = read_bfrag<D * 2>(
+              k_lds, ks * 32 + (lane & 31), c * 16 + hi * 8);
+          sv = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              ka, q_reg[c], sv, 0, 0, 0);
+          bf16x8v va = read_bfrag<D * 2>(
+              v_lds, ks * 32 + (lane & 31), c * 16 + hi * 8);
+          dpv = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              va, do_reg[c], dpv, 0, 0, 0);
         }
 
+        f32x16 dst;
 #pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        const int col = n0 + t * 16 + (lane & 15);
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int row = row_base + r;
+        for (int r = 0; r < 16; ++r) {
+          const int kvg = n0 + ks * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
           float pv = 0.f;
-          if (row < S && col < S && (!CAUSAL || col <= row) &&
-              lse_r[r] != NEG_INF)
-            pv = __expf(s_acc[t][r] * scale - lse_r[r]);
-          const float dsv = pv * (dp_acc[t][r] - ds_r[r]);
-          const int rl = (lane >> 4) * 4 + r;
-          const int cl = t * 16 + (lane & 15);
-          *(uint16_t*)(pw + rl * 128 + swz(rl, cl * 2)) = f2bfbits(dsv);
+          if (qrow < S && kvg < S && (!CAUSAL || kvg <= qrow) &&
+              lse_lane != NEG_INF)
+            pv = __expf(sv[r] * scale - lse_lane);
+          dst[r] = pv * (dpv[r] - ds_lane);
         }
-      }
-#pragma unroll
-      for (int kc = 0; kc < 2; ++kc) {
-        bf16x8v da = as_frag(*(const uint4*)(
-            pw + (lane & 15) * 128 +
-            swz(lane & 15, (kc * 32 + (lane >> 4) * 8) * 2)));
+
+        bf16x8v da0, da1;
+        {
+          union { uint32_t w[4]; bf16x8v f; } f0, f1;
+          auto sA = __builtin_amdgcn_permlane32_swap(
+              pack_bf16(dst[0], dst[1]), pack_bf16(dst[4], dst[5]), false,
+              false);
+          auto sB = __builtin_amdgcn_permlane32_swap(
+              pack_bf16(dst[2], dst[3]), pack_bf16(dst[6], dst[7]), false,
+              false);
+          f0.w[0] = sA[0]; f0.w[1] = sB[0]; f0.w[2] = sA[1]; f0.w[3] = sB[1];
+          auto sC = __builtin_amdgcn_permlane32_swap(
+              pack_bf16(dst[8], dst[9]), pack_bf16(dst[12], dst[13]), false,
+              false);
+          auto sD = __builtin_amdgcn_permlane32_swap(
+              pack_bf16(dst[10], dst[11]), pack_bf16(dst[14], dst[15]),
+              false, false);
+          f1.w[0] = sC[0]; f1.w[1] = sD[0]; f1.w[2] = sC[1]; f1.w[3] = sD[1];
+          da0 = f0.f;
+          da1 = f1.f;
+        }
 #pragma unroll
         for (int t = 0; t < DT; ++t) {
-          bf16x8v ktb = read_bfrag<128>(
-              kt_lds, t * 16 + (lane & 15), kc * 32 + (lane >> 4) * 8);
-          dq_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              da, ktb, dq_acc[t], 0, 0, 0);
+          bf16x8v b0 = read_bfrag<128>(
+              kt_lds, t * 32 + (lane & 31), ks * 32 + hi * 8);
+          dq_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              da0, b0, dq_acc[t], 0, 0, 0);
+          bf16x8v b1 = read_bfrag<128>(
+              kt_lds, t * 32 + (lane & 31), ks * 32 + 16 + hi * 8);
+          dq_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              da1, b1, dq_acc[t], 0, 0, 0);
         }
       }
     }
@@ -814,12 +944,12 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
   }
 
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int row = row_base + r;
+  for (int r = 0; r < 16; ++r) {
+    const int row = m0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
     if (row >= S) continue;
 #pragma unroll
     for (int t = 0; t < DT; ++t) {
-      const int col = t * 16 + (lane & 15);
+      const int col = t * 32 + (lane & 31);
       dqp[(long)row * q_tok + col] = f2bf(dq_acc[t][r] * scale);
     }
   }
@@ -863,14 +993,18 @@ hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
   else
     return hipErrorInvalidValue;
 
-  dim3 gkv((S + BM - 1) / BM, Hkv, B);
-  dim3 gq((S + BM - 1) / BM, Hq, B);
+  dim3 gkv((S + NW * 32 - 1) / (NW * 32), Hkv, B);
+  dim3 gq((S + NW * 32 - 1) / (NW * 32), Hq, B);
 #define LAUNCH_BWD(DD, CC)                                                    \
   do {                                                                        \
-    attn_bwd_dkdv_kernel<DD, CC><<<gkv, NTHREADS, 0, stream>>>(               \
+    attn_bwd_dv_kernel<DD, CC><<<gkv, NTHREADS, 0, stream>>>(                 \
         (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,                 \
-        (const bf16_t*)dout, lse, dsum_ws, (bf16_t*)dk, (bf16_t*)dv, B, S,   \
-        Hq, Hkv, scale);                                                      \
+        (const bf16_t*)dout, lse, dsum_ws, (bf16_t*)dv, B, S, Hq, Hkv,       \
+        scale);                                                               \
+    attn_bwd_dk_kernel<DD, CC><<<gkv, NTHREADS, 0, stream>>>(                 \
+        (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,                 \
+        (const bf16_t*)dout, lse, dsum_ws, (bf16_t*)dk, B, S, Hq, Hkv,       \
+        scale);                                                               \
     attn_bwd_dq_kernel<DD, CC><<<gq, NTHREADS, 0, stream>>>(                  \
         (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,                 \
         (const bf16_t*)dout, lse, dsum_ws, (bf16_t*)dq, B, S, Hq, Hkv,       \
