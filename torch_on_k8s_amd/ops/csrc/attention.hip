@@ -140,9 +140,9 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
   constexpr int DT = D / 32;    // O col tiles of 32
   constexpr int BM3 = NW * 32;  // 256 q rows per block
   constexpr int KB = BN * D * 2;
-  __shared__ __attribute__((aligned(16))) char smem[2 * KB];
-  char* k_lds = smem;
-  char* vt_lds = smem + KB;
+  // double-buffered K|VT: one barrier per tile; tile t+1's LDS write
+  // overlaps other waves' compute on tile t
+  __shared__ __attribute__((aligned(16))) char smem[2 * (2 * KB)];
 
   const int lane = threadIdx.x & 63;
   const int hi = lane >> 5;
@@ -181,12 +181,16 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
   const int n_end = CAUSAL ? min(S, m0 + BM3) : S;
   k_st.issue(kp, 0, S, kv_tok);
   v_st.issue(vp, 0, S, kv_tok);
+  k_st.write_rm(smem);
+  v_st.write_tr(smem + KB);
+  __syncthreads();
+  int cur = 0;
 
   for (int n0 = 0; n0 < n_end; n0 += BN) {
-    k_st.write_rm(k_lds);
-    v_st.write_tr(vt_lds);
-    __syncthreads();
-    if (n0 + BN < n_end) {
+    char* k_lds = smem + cur * (2 * KB);
+    char* vt_lds = k_lds + KB;
+    const bool more = n0 + BN < n_end;
+    if (more) {
       k_st.issue(kp, n0 + BN, S, kv_tok);
       v_st.issue(vp, n0 + BN, S, kv_tok);
     }
@@ -199,6 +203,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
       for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
         for (int r = 0; r < 16; ++r) st[ks][r] = 0.f;
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
@@ -208,6 +213,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
           st[ks] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               ka, q_reg[c], st[ks], 0, 0, 0);
         }
+      __builtin_amdgcn_s_setprio(0);
 
       // mask + scale + in-lane row max
       float mx = NEG_INF;
@@ -222,16 +228,27 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
           mx = fmaxf(mx, s);
         }
       mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
-      const float mn = fmaxf(m_run, mx);
-      const float alpha = (m_run == NEG_INF) ? 0.f : __expf(m_run - mn);
-      m_run = mn;
+      // defer-max (guide T13): when every row's max grew by < THR keep
+      // the old running max and skip the O rescale; P is then bounded by
+      // e^THR which the fp32 accumulator tolerates. Nothing of this tile
+      // is pending when the branch is taken (P*V of tile t completes
+      // before tile t+1's decision), so the T13 ordering hazard cannot
+      // occur in this structure.
+      constexpr float DEFER_THR = 8.f;
+      const bool rescale = !__all(mx <= m_run + DEFER_THR);
+      float alpha = 1.f;
+      if (rescale) {
+        const float mn = fmaxf(m_run, mx);
+        alpha = (m_run == NEG_INF) ? 0.f : __expf(m_run - mn);
+        m_run = mn;
+      }
       float rsum = 0.f;
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const float pe =
-              (st[ks][r] == NEG_INF) ? 0.f : __expf(st[ks][r] - mn);
+              (st[ks][r] == NEG_INF) ? 0.f : __expf(st[ks][r] - m_run);
           st[ks][r] = pe;
           rsum += pe;
         }
@@ -239,13 +256,16 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
       l_run = l_run * alpha + rsum;
 
       // rescale O by the per-ROW alpha (rows are reg-mapped; alpha is
-      // lane-mapped -> one bpermute gather per reg row)
+      // lane-mapped -> one bpermute gather per reg row); skipped
+      // entirely on the defer path
+      if (rescale) {
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int rowidx = (r & 3) + 8 * (r >> 2) + 4 * hi;
-        const float ar = __shfl(alpha, rowidx, 64);
+        for (int r = 0; r < 16; ++r) {
+          const int rowidx = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          const float ar = __shfl(alpha, rowidx, 64);
 #pragma unroll
-        for (int t = 0; t < DT; ++t) o_acc[t][r] *= ar;
+          for (int t = 0; t < DT; ++t) o_acc[t][r] *= ar;
+        }
       }
 
       // P (f32, C-layout) -> bf16 A-fragments via cvt_pk + permlane swap
@@ -276,6 +296,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
       }
 
       // O += P * V
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks)
 #pragma unroll
@@ -285,8 +306,15 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
           o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               pa[ks], vb, o_acc[t], 0, 0, 0);
         }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    if (more) {
+      char* nk = smem + (cur ^ 1) * (2 * KB);
+      k_st.write_rm(nk);
+      v_st.write_tr(nk + KB);
     }
     __syncthreads();
+    cur ^= 1;
   }
 
   // epilogue: O = o_acc / l (per reg row); LSE per lane row
@@ -541,9 +569,8 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
   constexpr int DT = D / 32;
   constexpr int BNK = NW * 32;   // 256 kv rows per block
   constexpr int KB = BN * D * 2;
-  __shared__ __attribute__((aligned(16))) char smem[2 * KB];  // Q rm | dOT
-  char* q_lds = smem;
-  char* dot_lds = smem + KB;
+  // double-buffered (Q rm | dOT): one barrier per q tile
+  __shared__ __attribute__((aligned(16))) char smem[2 * (2 * KB)];
 
   const int lane = threadIdx.x & 63;
   const int hi = lane >> 5;
@@ -580,6 +607,10 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
              q_tok);
   do_st.issue(dout + ((long)b * S * q_tok) + (long)(hkv * rep) * D, m_start,
               S, q_tok);
+  q_st.write_rm(smem);
+  do_st.write_tr(smem + KB);
+  __syncthreads();
+  int cur = 0;
 
   for (int g = 0; g < rep; ++g) {
     const int hq = hkv * rep + g;
@@ -588,10 +619,10 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
     const float* lsep = lse + ((long)b * Hq + hq) * S;
 
     for (int m0 = m_start; m0 < S; m0 += BN) {
-      q_st.write_rm(q_lds);
-      do_st.write_tr(dot_lds);
-      __syncthreads();
+      char* q_lds = smem + cur * (2 * KB);
+      char* dot_lds = q_lds + KB;
       const int m1 = m0 + BN;
+      const bool more = m1 < S || g + 1 < rep;
       if (m1 < S) {
         q_st.issue(qp, m1, S, q_tok);
         do_st.issue(dop, m1, S, q_tok);
@@ -640,7 +671,13 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
               pa1, b1, dv_acc[t], 0, 0, 0);
         }
       }
+      if (more) {
+        char* nq = smem + (cur ^ 1) * (2 * KB);
+        q_st.write_rm(nq);
+        do_st.write_tr(nq + KB);
+      }
       __syncthreads();
+      cur ^= 1;
     }
   }
 
@@ -814,11 +851,8 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
   constexpr int DT = D / 32;
   constexpr int BM3 = NW * 32;
   constexpr int KB = BN * D * 2;
-  // K rm | KT | V rm
-  __shared__ __attribute__((aligned(16))) char smem[3 * KB];
-  char* k_lds = smem;
-  char* kt_lds = smem + KB;
-  char* v_lds = smem + 2 * KB;
+  // double-buffered (K rm | KT | V rm): one barrier per kv tile
+  __shared__ __attribute__((aligned(16))) char smem[2 * (3 * KB)];
 
   const int lane = threadIdx.x & 63;
   const int hi = lane >> 5;
@@ -863,13 +897,18 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
   const int n_end = CAUSAL ? min(S, m0 + BM3) : S;
   k_st.issue(kp, 0, S, kv_tok);
   v_st.issue(vp, 0, S, kv_tok);
+  k_st.write_rm(smem);
+  k_st.write_tr(smem + KB);
+  v_st.write_rm(smem + 2 * KB);
+  __syncthreads();
+  int cur = 0;
 
   for (int n0 = 0; n0 < n_end; n0 += BN) {
-    k_st.write_rm(k_lds);
-    k_st.write_tr(kt_lds);
-    v_st.write_rm(v_lds);
-    __syncthreads();
-    if (n0 + BN < n_end) {
+    char* k_lds = smem + cur * (3 * KB);
+    char* kt_lds = k_lds + KB;
+    char* v_lds = k_lds + 2 * KB;
+    const bool more = n0 + BN < n_end;
+    if (more) {
       k_st.issue(kp, n0 + BN, S, kv_tok);
       v_st.issue(vp, n0 + BN, S, kv_tok);
     }
@@ -940,7 +979,14 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
         }
       }
     }
+    if (more) {
+      char* nk = smem + (cur ^ 1) * (3 * KB);
+      k_st.write_rm(nk);
+      k_st.write_tr(nk + KB);
+      v_st.write_rm(nk + 2 * KB);
+    }
     __syncthreads();
+    cur ^= 1;
   }
 
 #pragma unroll
