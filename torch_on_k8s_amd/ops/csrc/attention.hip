@@ -92,6 +92,35 @@ struct TileStage {
   }
 };
 
+// Loads a [D][64] tile (rows = d, cols = 64 tokens) from a
+// pre-transposed [D][S_pad] plane (transpose.hip) into swizzled LDS with
+// plain b128 writes — replaces the in-kernel scalar transpose scatter
+// that measured ~20 LDS bank-conflict cycles per MFMA.
+template <int D, int NTH = NTHREADS>
+struct TileStageT {
+  static constexpr int VPR = 64 / 8;
+  static constexpr int VPT = D * VPR / NTH;
+  uint4 vals[VPT];
+
+  DEVINL void issue(const bf16_t* plane, int col0, long S_pad) {
+#pragma unroll
+    for (int i = 0; i < VPT; ++i) {
+      const int vi = threadIdx.x + i * NTH;
+      const int row = vi / VPR, cv = vi % VPR;
+      vals[i] = *(const uint4*)(plane + (long)row * S_pad + col0 + cv * 8);
+    }
+  }
+
+  DEVINL void write(char* lds) const {
+#pragma unroll
+    for (int i = 0; i < VPT; ++i) {
+      const int vi = threadIdx.x + i * NTH;
+      const int row = vi / VPR, cv = vi % VPR;
+      *(uint4*)(lds + row * 128 + swz(row, cv * 16)) = vals[i];
+    }
+  }
+};
+
 // B-fragment read from a swizzled row-major LDS tile, row stride RB bytes.
 template <int RB>
 DEVINL bf16x8v read_bfrag(const char* lds, int row, int col_elem) {
@@ -134,8 +163,9 @@ DEVINL uint32_t pack_bf16(float lo, float hi) {
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
-    const bf16_t* __restrict__ v, bf16_t* __restrict__ o,
-    float* __restrict__ lse, int B, int S, int Hq, int Hkv, float scale) {
+    const bf16_t* __restrict__ vt, bf16_t* __restrict__ o,
+    float* __restrict__ lse, int B, int S, int Hq, int Hkv, int S_pad,
+    float scale) {
   constexpr int QC = D / 16;    // Q B-fragments (k-slots of 16)
   constexpr int DT = D / 32;    // O col tiles of 32
   constexpr int BM3 = NW * 32;  // 256 q rows per block
@@ -156,7 +186,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
   const long q_tok = (long)Hq * D, kv_tok = (long)Hkv * D;
   const bf16_t* qp = q + ((long)b * S * q_tok) + (long)hq * D;
   const bf16_t* kp = k + ((long)b * S * kv_tok) + (long)hkv * D;
-  const bf16_t* vp = v + ((long)b * S * kv_tok) + (long)hkv * D;
+  const bf16_t* vtp = vt + ((long)b * Hkv + hkv) * D * (long)S_pad;
   bf16_t* op = o + ((long)b * S * q_tok) + (long)hq * D;
   float* lsep = lse + ((long)b * Hq + hq) * S;
 
@@ -177,12 +207,13 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
     for (int r = 0; r < 16; ++r) o_acc[t][r] = 0.f;
   float m_run = NEG_INF, l_run = 0.f;
 
-  TileStage<D> k_st, v_st;
+  TileStage<D> k_st;
+  TileStageT<D> v_st;
   const int n_end = CAUSAL ? min(S, m0 + BM3) : S;
   k_st.issue(kp, 0, S, kv_tok);
-  v_st.issue(vp, 0, S, kv_tok);
+  v_st.issue(vtp, 0, S_pad);
   k_st.write_rm(smem);
-  v_st.write_tr(smem + KB);
+  v_st.write(smem + KB);
   __syncthreads();
   int cur = 0;
 
@@ -192,7 +223,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
     const bool more = n0 + BN < n_end;
     if (more) {
       k_st.issue(kp, n0 + BN, S, kv_tok);
-      v_st.issue(vp, n0 + BN, S, kv_tok);
+      v_st.issue(vtp, n0 + BN, S_pad);
     }
 
     const bool strip_live = !CAUSAL || (n0 <= m0w + 31);
@@ -311,7 +342,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
     if (more) {
       char* nk = smem + (cur ^ 1) * (2 * KB);
       k_st.write_rm(nk);
-      v_st.write_tr(nk + KB);
+      v_st.write(nk + KB);
     }
     __syncthreads();
     cur ^= 1;
@@ -562,9 +593,10 @@ DEVINL void repack_pa(const f32x16& pt, bf16x8v& pa0, bf16x8v& pa1) {
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
-    const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
+    const bf16_t* __restrict__ dot_t, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ dsum,
-    bf16_t* __restrict__ dv, int B, int S, int Hq, int Hkv, float scale) {
+    bf16_t* __restrict__ dv, int B, int S, int Hq, int Hkv, int S_pad,
+    float scale) {
   constexpr int QC = D / 16;
   constexpr int DT = D / 32;
   constexpr int BNK = NW * 32;   // 256 kv rows per block
@@ -602,20 +634,22 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
     for (int r = 0; r < 16; ++r) dv_acc[t][r] = 0.f;
 
   const int m_start = CAUSAL ? (n0 / BN) * BN : 0;
-  TileStage<D> q_st, do_st;
+  const long dplane = (long)D * S_pad;
+  TileStage<D> q_st;
+  TileStageT<D> do_st;
+  const bf16_t* dot0 = dot_t + ((long)b * Hq + hkv * rep) * dplane;
   q_st.issue(q + ((long)b * S * q_tok) + (long)(hkv * rep) * D, m_start, S,
              q_tok);
-  do_st.issue(dout + ((long)b * S * q_tok) + (long)(hkv * rep) * D, m_start,
-              S, q_tok);
+  do_st.issue(dot0, m_start, S_pad);
   q_st.write_rm(smem);
-  do_st.write_tr(smem + KB);
+  do_st.write(smem + KB);
   __syncthreads();
   int cur = 0;
 
   for (int g = 0; g < rep; ++g) {
     const int hq = hkv * rep + g;
     const bf16_t* qp = q + ((long)b * S * q_tok) + (long)hq * D;
-    const bf16_t* dop = dout + ((long)b * S * q_tok) + (long)hq * D;
+    const bf16_t* dotp = dot_t + ((long)b * Hq + hq) * dplane;
     const float* lsep = lse + ((long)b * Hq + hq) * S;
 
     for (int m0 = m_start; m0 < S; m0 += BN) {
@@ -625,10 +659,10 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
       const bool more = m1 < S || g + 1 < rep;
       if (m1 < S) {
         q_st.issue(qp, m1, S, q_tok);
-        do_st.issue(dop, m1, S, q_tok);
+        do_st.issue(dotp, m1, S_pad);
       } else if (g + 1 < rep) {
         q_st.issue(qp + D, m_start, S, q_tok);
-        do_st.issue(dop + D, m_start, S, q_tok);
+        do_st.issue(dotp + dplane, m_start, S_pad);
       }
 
 #pragma unroll
@@ -674,7 +708,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
       if (more) {
         char* nq = smem + (cur ^ 1) * (2 * KB);
         q_st.write_rm(nq);
-        do_st.write_tr(nq + KB);
+        do_st.write(nq + KB);
       }
       __syncthreads();
       cur ^= 1;
@@ -695,8 +729,9 @@ template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
-    const float* __restrict__ lse, const float* __restrict__ dsum,
-    bf16_t* __restrict__ dk, int B, int S, int Hq, int Hkv, float scale) {
+    const bf16_t* __restrict__ q_t, const float* __restrict__ lse,
+    const float* __restrict__ dsum, bf16_t* __restrict__ dk, int B, int S,
+    int Hq, int Hkv, int S_pad, float scale) {
   constexpr int QC = D / 16;
   constexpr int DT = D / 32;
   constexpr int BNK = NW * 32;
@@ -759,14 +794,17 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
     const float* lsep = lse + ((long)b * Hq + hq) * S;
     const float* dsp = dsum + ((long)b * Hq + hq) * S;
 
+    const bf16_t* qtp = q_t + ((long)b * Hq + hq) * (long)D * S_pad;
     for (int m0 = m_start; m0 < S; m0 += BN) {
       {  // synchronous staging: the prefetch ring costs ~16 VGPRs and
          // tips this kernel into scratch spill, which is worse
         TileStage<D> q_st, do_st;
+        TileStageT<D> qt_st;
         q_st.issue(qp, m0, S, q_tok);
         do_st.issue(dop, m0, S, q_tok);
+        qt_st.issue(qtp, m0, S_pad);
         q_st.write_rm(q_lds);
-        q_st.write_tr(qt_lds);
+        qt_st.write(qt_lds);
         do_st.write_rm(do_lds);
       }
       __syncthreads();
@@ -845,8 +883,9 @@ template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
-    const float* __restrict__ lse, const float* __restrict__ dsum,
-    bf16_t* __restrict__ dq, int B, int S, int Hq, int Hkv, float scale) {
+    const bf16_t* __restrict__ k_t, const float* __restrict__ lse,
+    const float* __restrict__ dsum, bf16_t* __restrict__ dq, int B, int S,
+    int Hq, int Hkv, int S_pad, float scale) {
   constexpr int QC = D / 16;
   constexpr int DT = D / 32;
   constexpr int BM3 = NW * 32;
@@ -868,6 +907,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
   const bf16_t* kp = k + ((long)b * S * kv_tok) + (long)hkv * D;
   const bf16_t* vp = v + ((long)b * S * kv_tok) + (long)hkv * D;
   const bf16_t* dop = dout + ((long)b * S * q_tok) + (long)hq * D;
+  const bf16_t* ktp = k_t + ((long)b * Hkv + hkv) * (long)D * S_pad;
   const float* lsep = lse + ((long)b * Hq + hq) * S;
   const float* dsp = dsum + ((long)b * Hq + hq) * S;
   bf16_t* dqp = dq + ((long)b * S * q_tok) + (long)hq * D;
@@ -894,11 +934,13 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
     for (int r = 0; r < 16; ++r) dq_acc[t][r] = 0.f;
 
   TileStage<D> k_st, v_st;
+  TileStageT<D> kt_st;
   const int n_end = CAUSAL ? min(S, m0 + BM3) : S;
   k_st.issue(kp, 0, S, kv_tok);
   v_st.issue(vp, 0, S, kv_tok);
+  kt_st.issue(ktp, 0, S_pad);
   k_st.write_rm(smem);
-  k_st.write_tr(smem + KB);
+  kt_st.write(smem + KB);
   v_st.write_rm(smem + 2 * KB);
   __syncthreads();
   int cur = 0;
@@ -911,6 +953,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
     if (more) {
       k_st.issue(kp, n0 + BN, S, kv_tok);
       v_st.issue(vp, n0 + BN, S, kv_tok);
+      kt_st.issue(ktp, n0 + BN, S_pad);
     }
 
     const bool strip_live = !CAUSAL || (n0 <= m0w + 31);
@@ -982,7 +1025,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
     if (more) {
       char* nk = smem + (cur ^ 1) * (3 * KB);
       k_st.write_rm(nk);
-      k_st.write_tr(nk + KB);
+      kt_st.write(nk + KB);
       v_st.write_rm(nk + 2 * KB);
     }
     __syncthreads();
@@ -1005,15 +1048,16 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
 
 extern "C" {
 
-hipError_t tok_attn_fwd(const void* q, const void* k, const void* v, void* o,
+// vt: pre-transposed V plane [B, Hkv, D, S_pad] (tok_transpose_head)
+hipError_t tok_attn_fwd(const void* q, const void* k, const void* vt, void* o,
                         float* lse, int B, int S, int Hq, int Hkv, int D,
-                        int causal, hipStream_t stream) {
+                        int S_pad, int causal, hipStream_t stream) {
   dim3 grid((S + NW * 32 - 1) / (NW * 32), Hq, B);
   const float scale = 1.f / sqrtf((float)D);
 #define LAUNCH_FWD(DD, CC)                                                    \
   attn_fwd_kernel_v3<DD, CC><<<grid, NTHREADS, 0, stream>>>(                  \
-      (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (bf16_t*)o, lse, \
-      B, S, Hq, Hkv, scale)
+      (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)vt, (bf16_t*)o,     \
+      lse, B, S, Hq, Hkv, S_pad, scale)
   if (D == 128) { if (causal) LAUNCH_FWD(128, true); else LAUNCH_FWD(128, false); }
   else if (D == 64) { if (causal) LAUNCH_FWD(64, true); else LAUNCH_FWD(64, false); }
   else return hipErrorInvalidValue;
@@ -1021,10 +1065,12 @@ hipError_t tok_attn_fwd(const void* q, const void* k, const void* v, void* o,
   return hipGetLastError();
 }
 
+// q_t/k_t/dot_t: pre-transposed [.., D, S_pad] planes (tok_transpose_head)
 hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
-                        const void* o, const void* dout, const float* lse,
+                        const void* o, const void* dout, const void* q_t,
+                        const void* k_t, const void* dot_t, const float* lse,
                         float* dsum_ws, void* dq, void* dk, void* dv, int B,
-                        int S, int Hq, int Hkv, int D, int causal,
+                        int S, int Hq, int Hkv, int D, int S_pad, int causal,
                         hipStream_t stream) {
   const float scale = 1.f / sqrtf((float)D);
   const long rows = (long)B * S * Hq;
@@ -1044,17 +1090,17 @@ hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
 #define LAUNCH_BWD(DD, CC)                                                    \
   do {                                                                        \
     attn_bwd_dv_kernel<DD, CC><<<gkv, NTHREADS, 0, stream>>>(                 \
-        (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,                 \
+        (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)dot_t,             \
         (const bf16_t*)dout, lse, dsum_ws, (bf16_t*)dv, B, S, Hq, Hkv,       \
-        scale);                                                               \
+        S_pad, scale);                                                        \
     attn_bwd_dk_kernel<DD, CC><<<gkv, NTHREADS, 0, stream>>>(                 \
         (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,                 \
-        (const bf16_t*)dout, lse, dsum_ws, (bf16_t*)dk, B, S, Hq, Hkv,       \
-        scale);                                                               \
+        (const bf16_t*)dout, (const bf16_t*)q_t, lse, dsum_ws, (bf16_t*)dk,  \
+        B, S, Hq, Hkv, S_pad, scale);                                         \
     attn_bwd_dq_kernel<DD, CC><<<gq, NTHREADS, 0, stream>>>(                  \
         (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,                 \
-        (const bf16_t*)dout, lse, dsum_ws, (bf16_t*)dq, B, S, Hq, Hkv,       \
-        scale);                                                               \
+        (const bf16_t*)dout, (const bf16_t*)k_t, lse, dsum_ws, (bf16_t*)dq,  \
+        B, S, Hq, Hkv, S_pad, scale);                                         \
   } while (0)
   if (D == 128) { if (causal) LAUNCH_BWD(128, true); else LAUNCH_BWD(128, false); }
   else { if (causal) LAUNCH_BWD(64, true); else LAUNCH_BWD(64, false); }

@@ -24,14 +24,17 @@ hipError_t tok_mfma_probe_16x16x32(const void* A, const void* B, float* D,
                                    hipStream_t stream);
 hipError_t tok_mfma_probe_32x32x16(const void* A, const void* B, float* D,
                                    hipStream_t stream);
-hipError_t tok_attn_fwd(const void* q, const void* k, const void* v, void* o,
+hipError_t tok_attn_fwd(const void* q, const void* k, const void* vt, void* o,
                         float* lse, int B, int S, int Hq, int Hkv, int D,
-                        int causal, hipStream_t stream);
+                        int S_pad, int causal, hipStream_t stream);
 hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
-                        const void* o, const void* dout, const float* lse,
+                        const void* o, const void* dout, const void* q_t,
+                        const void* k_t, const void* dot_t, const float* lse,
                         float* dsum_ws, void* dq, void* dk, void* dv, int B,
-                        int S, int Hq, int Hkv, int D, int causal,
+                        int S, int Hq, int Hkv, int D, int S_pad, int causal,
                         hipStream_t stream);
+hipError_t tok_transpose_head(const void* in, void* out, int B, int S, int H,
+                              int D, int S_pad, hipStream_t stream);
 }
 
 namespace {
@@ -134,9 +137,13 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   TORCH_CHECK(k.sizes() == v.sizes() && k.size(0) == B && k.size(1) == S);
   auto o = at::empty_like(q);
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
-  TOK_HIP_OK(tok_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+  const int S_pad = (S + 63) / 64 * 64;
+  auto vt = at::empty({B, Hkv, D, S_pad}, q.options());
+  TOK_HIP_OK(tok_transpose_head(v.data_ptr(), vt.data_ptr(), B, S, Hkv, D,
+                                S_pad, current_stream()));
+  TOK_HIP_OK(tok_attn_fwd(q.data_ptr(), k.data_ptr(), vt.data_ptr(),
                           o.data_ptr(), lse.data_ptr<float>(), B, S, Hq, Hkv,
-                          D, causal ? 1 : 0, current_stream()));
+                          D, S_pad, causal ? 1 : 0, current_stream()));
   return {o, lse};
 }
 
@@ -151,11 +158,22 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
   auto dsum = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  const int S_pad = (S + 63) / 64 * 64;
+  auto qt = at::empty({B, Hq, D, S_pad}, q.options());
+  auto kt = at::empty({B, Hkv, D, S_pad}, q.options());
+  auto dot = at::empty({B, Hq, D, S_pad}, q.options());
+  TOK_HIP_OK(tok_transpose_head(q.data_ptr(), qt.data_ptr(), B, S, Hq, D,
+                                S_pad, current_stream()));
+  TOK_HIP_OK(tok_transpose_head(k.data_ptr(), kt.data_ptr(), B, S, Hkv, D,
+                                S_pad, current_stream()));
+  TOK_HIP_OK(tok_transpose_head(dout.data_ptr(), dot.data_ptr(), B, S, Hq, D,
+                                S_pad, current_stream()));
   TOK_HIP_OK(tok_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                          o.data_ptr(), dout.data_ptr(),
+                          o.data_ptr(), dout.data_ptr(), qt.data_ptr(),
+                          kt.data_ptr(), dot.data_ptr(),
                           lse.data_ptr<float>(), dsum.data_ptr<float>(),
                           dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), B, S,
-                          Hq, Hkv, D, causal ? 1 : 0, current_stream()));
+                          Hq, Hkv, D, S_pad, causal ? 1 : 0, current_stream()));
   return {dq, dk, dv};
 }
 
