@@ -116,6 +116,25 @@ def test_fused_adamw_vs_reference():
     assert (vg.cpu() - v_ref).abs().max() < 0.02
 
 
+def test_fused_cross_entropy():
+    torch.manual_seed(0)
+    N, V = 64, 1024
+    logits = (torch.randn(N, V) * 2).bfloat16().float()
+    labels = torch.randint(0, V, (N,))
+
+    lr = logits.clone().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(lr, labels)
+    ref.backward()
+
+    lg = logits.bfloat16().to(dev()).requires_grad_(True)
+    loss = ops.cross_entropy(lg, labels.to(dev()))
+    loss.backward()
+
+    assert abs(loss.item() - ref.item()) < 2e-3
+    gerr = (lg.grad.float().cpu() - lr.grad).abs().max().item()
+    assert gerr < 1e-3, f"CE grad err {gerr}"
+
+
 def test_trainer_single_gpu_step():
     """End-to-end: tiny Llama fwd+bwd+fused AdamW on GPU, loss finite and
     decreasing on a fixed batch."""

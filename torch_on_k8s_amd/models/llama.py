@@ -206,9 +206,7 @@ class LlamaModel(nn.Module):
         logits = self.lm_head(x)
         if labels is None:
             return logits
-        loss = F.cross_entropy(
-            logits.float().view(-1, self.cfg.vocab_size), labels.view(-1))
-        return loss
+        return ops.cross_entropy(logits, labels)
 
     def num_params(self) -> int:
         seen, total = set(), 0

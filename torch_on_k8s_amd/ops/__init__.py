@@ -199,7 +199,37 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return attention_ref(q, k, v, causal)
 
 
+# --------------------------------------------------------------------------
+# Fused cross-entropy (mean reduction over all tokens)
+# --------------------------------------------------------------------------
+class _FusedCEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels):
+        loss, lse = _require_ext("cross_entropy").ce_fwd(logits, labels)
+        ctx.save_for_backward(logits, labels, lse)
+        return loss.mean()
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        logits, labels, lse = ctx.saved_tensors
+        # device-scalar grad scale: stays correct under hipGraph replay
+        gscale = (grad_out.float() / logits.numel() * logits.size(-1)) \
+            .reshape(())
+        dlogits = _C.ce_bwd(logits, labels, lse, gscale.contiguous())
+        return dlogits, None
+
+
+def cross_entropy(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    """Mean next-token CE. GPU bf16: fused single-pass kernels (one logits
+    read fwd, one read + bf16 grad write bwd). CPU: eager fp32."""
+    if logits.is_cuda and logits.dtype == torch.bfloat16:
+        return _FusedCEFn.apply(logits.contiguous().view(-1, logits.shape[-1]),
+                                labels.reshape(-1).int())
+    return torch.nn.functional.cross_entropy(
+        logits.float().view(-1, logits.shape[-1]), labels.reshape(-1))
+
+
 __all__ = [
     "rmsnorm", "rmsnorm_ref", "apply_rope", "rope_ref", "fused_adamw_",
-    "attention", "attention_ref", "hip_ext_available",
+    "attention", "attention_ref", "cross_entropy", "hip_ext_available",
 ]

@@ -36,7 +36,11 @@ constexpr int BM = NW * 16;           // 128 rows owned per block (1 strip/wave)
 constexpr float NEG_INF = -INFINITY;
 
 DEVINL int swz(int row, int byte_in_row) {
-  return byte_in_row ^ ((row & 7) << 4);
+  // (row&15)<<4 spreads a 16-lane ds_read_b128 group over 16 16-B slots
+  // (conflict-free when the group's rows are distinct mod 16, guide §6
+  // G4). For 128-B-row tiles the XOR crosses into the neighbour row's
+  // bytes; write and read use the same map, so it stays a bijection.
+  return byte_in_row ^ ((row & 15) << 4);
 }
 
 DEVINL bf16x8v as_frag(uint4 raw) {
@@ -246,18 +250,32 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
         }
       __builtin_amdgcn_s_setprio(0);
 
-      // mask + scale + in-lane row max
+      // mask + scale + in-lane row max; interior tiles (fully below the
+      // diagonal and inside S) skip the per-element mask entirely
+      const bool needs_mask = (n0 + BN > S) ||
+                              (CAUSAL && (n0 + BN - 1 > m0w));
       float mx = NEG_INF;
+      if (needs_mask) {
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
+        for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int kvg = n0 + ks * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          float s = st[ks][r] * scale;
-          if ((CAUSAL && kvg > qrow) || kvg >= S || qrow >= S) s = NEG_INF;
-          st[ks][r] = s;
-          mx = fmaxf(mx, s);
-        }
+          for (int r = 0; r < 16; ++r) {
+            const int kvg = n0 + ks * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+            float s = st[ks][r] * scale;
+            if ((CAUSAL && kvg > qrow) || kvg >= S || qrow >= S) s = NEG_INF;
+            st[ks][r] = s;
+            mx = fmaxf(mx, s);
+          }
+      } else {
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const float s = st[ks][r] * scale;
+            st[ks][r] = s;
+            mx = fmaxf(mx, s);
+          }
+      }
       mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
       // defer-max (guide T13): when every row's max grew by < THR keep
       // the old running max and skip the O rescale; P is then bounded by

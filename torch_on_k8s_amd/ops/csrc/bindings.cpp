@@ -35,6 +35,11 @@ hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
                         hipStream_t stream);
 hipError_t tok_transpose_head(const void* in, void* out, int B, int S, int H,
                               int D, int S_pad, hipStream_t stream);
+hipError_t tok_ce_fwd(const void* logits, const int* labels, float* loss,
+                      float* lse, long rows, int V, hipStream_t stream);
+hipError_t tok_ce_bwd(const void* logits, const int* labels, const float* lse,
+                      const float* gscale, void* dlogits, long rows, int V,
+                      hipStream_t stream);
 }
 
 namespace {
@@ -177,6 +182,34 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {dq, dk, dv};
 }
 
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels) {
+  CHECK_BF16_CUDA(logits);
+  TORCH_CHECK(labels.scalar_type() == at::kInt && labels.is_cuda());
+  const long V = logits.size(-1);
+  TORCH_CHECK(V % 8 == 0, "vocab must be a multiple of 8");
+  const long rows = logits.numel() / V;
+  TORCH_CHECK(labels.numel() == rows);
+  auto loss = at::empty({rows}, logits.options().dtype(at::kFloat));
+  auto lse = at::empty({rows}, logits.options().dtype(at::kFloat));
+  TOK_HIP_OK(tok_ce_fwd(logits.data_ptr(), labels.data_ptr<int>(),
+                        loss.data_ptr<float>(), lse.data_ptr<float>(), rows,
+                        (int)V, current_stream()));
+  return {loss, lse};
+}
+
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor lse,
+                  at::Tensor gscale) {
+  CHECK_BF16_CUDA(logits);
+  TORCH_CHECK(gscale.scalar_type() == at::kFloat && gscale.is_cuda());
+  const long V = logits.size(-1);
+  const long rows = logits.numel() / V;
+  auto dlogits = at::empty_like(logits);
+  TOK_HIP_OK(tok_ce_bwd(logits.data_ptr(), labels.data_ptr<int>(),
+                        lse.data_ptr<float>(), gscale.data_ptr<float>(),
+                        dlogits.data_ptr(), rows, (int)V, current_stream()));
+  return dlogits;
+}
+
 at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
   CHECK_BF16_CUDA(A);
   CHECK_BF16_CUDA(B);
@@ -202,6 +235,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 bf16 layout probe");
   mod.def("attn_fwd", &attn_fwd, "Flash attention forward (bf16, gfx950)");
   mod.def("attn_bwd", &attn_bwd, "Flash attention backward (bf16, gfx950)");
+  mod.def("ce_fwd", &ce_fwd, "Fused cross-entropy forward (bf16, gfx950)");
+  mod.def("ce_bwd", &ce_bwd, "Fused cross-entropy backward (bf16, gfx950)");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16, gfx950)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (bf16, gfx950)");
   mod.def("rope", &rope, "Rotary embedding rotate-half (bf16, gfx950)");
