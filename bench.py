@@ -19,8 +19,13 @@ import os
 import sys
 import time
 
-import torch
-import torch.distributed as dist
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+from torch_on_k8s_amd.tunable import setup_tunableop  # noqa: E402
+
+setup_tunableop()  # load committed hipBLASLt tuning results (if any)
+
+import torch  # noqa: E402
+import torch.distributed as dist  # noqa: E402
 
 
 def parse_args():

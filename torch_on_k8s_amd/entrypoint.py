@@ -72,6 +72,8 @@ def aimaster_main() -> int:
 def main() -> int:
     if os.environ.get("TOK_TASK_TYPE") == "aimaster":
         return aimaster_main()
+    from torch_on_k8s_amd.tunable import setup_tunableop
+    setup_tunableop()
     import torch
     from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
     from torch_on_k8s_amd.parallel.env import init_distributed, destroy, barrier

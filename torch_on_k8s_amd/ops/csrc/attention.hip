@@ -35,12 +35,14 @@ constexpr int NTHREADS = NW * WAVE;   // 512
 constexpr int BM = NW * 16;           // 128 rows owned per block (1 strip/wave)
 constexpr float NEG_INF = -INFINITY;
 
-DEVINL int swz(int row, int byte_in_row) {
-  // (row&15)<<4 spreads a 16-lane ds_read_b128 group over 16 16-B slots
-  // (conflict-free when the group's rows are distinct mod 16, guide §6
-  // G4). For 128-B-row tiles the XOR crosses into the neighbour row's
-  // bytes; write and read use the same map, so it stays a bijection.
-  return byte_in_row ^ ((row & 15) << 4);
+// Swizzled tile offset: (row*RB + col) ^ ((row&15)<<4) spreads a 16-lane
+// ds_read_b128 group over 16 16-B slots (conflict-free when the group's
+// rows are distinct mod 16, guide §6 G4). The XOR is applied to the FULL
+// tile offset: for 128-B rows bit 7 of the mask swaps row parity, which
+// keeps the mapping a bijection INSIDE the tile (XOR-ing only the
+// byte-in-row would escape the last rows' allocation).
+DEVINL int swz_off(int row, int row_bytes, int byte_in_row) {
+  return (row * row_bytes + byte_in_row) ^ ((row & 15) << 4);
 }
 
 DEVINL bf16x8v as_frag(uint4 raw) {
@@ -77,7 +79,7 @@ struct TileStage {
     for (int i = 0; i < VPT; ++i) {
       const int vi = threadIdx.x + i * NTH;
       const int row = vi / VPR, cv = vi % VPR;
-      *(uint4*)(lds + row * (D * 2) + swz(row, cv * 16)) = vals[i];
+      *(uint4*)(lds + swz_off(row, D * 2, cv * 16)) = vals[i];
     }
   }
 
@@ -90,7 +92,7 @@ struct TileStage {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int d = cv * 8 + j;
-        *(uint16_t*)(lds + d * 128 + swz(d, row * 2)) = h[j];
+        *(uint16_t*)(lds + swz_off(d, 128, row * 2)) = h[j];
       }
     }
   }
@@ -120,7 +122,7 @@ struct TileStageT {
     for (int i = 0; i < VPT; ++i) {
       const int vi = threadIdx.x + i * NTH;
       const int row = vi / VPR, cv = vi % VPR;
-      *(uint4*)(lds + row * 128 + swz(row, cv * 16)) = vals[i];
+      *(uint4*)(lds + swz_off(row, 128, cv * 16)) = vals[i];
     }
   }
 };
@@ -128,7 +130,7 @@ struct TileStageT {
 // B-fragment read from a swizzled row-major LDS tile, row stride RB bytes.
 template <int RB>
 DEVINL bf16x8v read_bfrag(const char* lds, int row, int col_elem) {
-  return as_frag(*(const uint4*)(lds + row * RB + swz(row, col_elem * 2)));
+  return as_frag(*(const uint4*)(lds + swz_off(row, RB, col_elem * 2)));
 }
 
 // A-fragment set (rows m = lane&15 of a 16-row strip) from global.
@@ -385,173 +387,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel_v3(
 }
 
 // ========================================================================
-// Forward (16x16 variant, kept for reference/ablation; v3 is dispatched)
-// ========================================================================
-template <int D, bool CAUSAL>
-__global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
-    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
-    const bf16_t* __restrict__ v, bf16_t* __restrict__ o,
-    float* __restrict__ lse, int B, int S, int Hq, int Hkv, float scale) {
-  constexpr int DC = D / 32;
-  constexpr int DT = D / 16;
-  constexpr int KB = BN * D * 2;
-  __shared__ __attribute__((aligned(16))) char smem[KB + KB + NW * 2048];
-  char* k_lds = smem;
-  char* vt_lds = smem + KB;
-  char* p_lds = smem + 2 * KB;
-
-  const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
-  const int m0 = blockIdx.x * BM;
-  const int hq = blockIdx.y;
-  const int b = blockIdx.z;
-  const int hkv = hq / (Hq / Hkv);
-
-  const long q_tok = (long)Hq * D, kv_tok = (long)Hkv * D;
-  const bf16_t* qp = q + ((long)b * S * q_tok) + (long)hq * D;
-  const bf16_t* kp = k + ((long)b * S * kv_tok) + (long)hkv * D;
-  const bf16_t* vp = v + ((long)b * S * kv_tok) + (long)hkv * D;
-  bf16_t* op = o + ((long)b * S * q_tok) + (long)hq * D;
-  float* lsep = lse + ((long)b * Hq + hq) * S;
-
-  bf16x8v q_frag[DC];
-  const int mrow = m0 + wid * 16 + (lane & 15);
-  load_afrags<D>(q_frag, qp, mrow, S, q_tok, lane);
-
-  f32x4 o_acc[DT];
-#pragma unroll
-  for (int t = 0; t < DT; ++t) o_acc[t] = {0.f, 0.f, 0.f, 0.f};
-  float m_run[4] = {NEG_INF, NEG_INF, NEG_INF, NEG_INF};
-  float l_run[4] = {0.f, 0.f, 0.f, 0.f};
-
-  TileStage<D> k_st, v_st;
-  const int n_end = CAUSAL ? min(S, m0 + BM) : S;
-  k_st.issue(kp, 0, S, kv_tok);
-  v_st.issue(vp, 0, S, kv_tok);
-
-  for (int n0 = 0; n0 < n_end; n0 += BN) {
-    // stage tile n0 into LDS (loads were issued last iteration)
-    k_st.write_rm(k_lds);
-    v_st.write_tr(vt_lds);
-    __syncthreads();
-    // prefetch tile n0+BN during compute
-    if (n0 + BN < n_end) {
-      k_st.issue(kp, n0 + BN, S, kv_tok);
-      v_st.issue(vp, n0 + BN, S, kv_tok);
-    }
-
-    const bool strip_live = !CAUSAL || (n0 <= m0 + wid * 16 + 15);
-    if (strip_live) {
-      f32x4 s_acc[4];
-#pragma unroll
-      for (int t = 0; t < 4; ++t) s_acc[t] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int t = 0; t < 4; ++t)
-#pragma unroll
-        for (int c = 0; c < DC; ++c) {
-          bf16x8v bfr = read_bfrag<D * 2>(
-              k_lds, t * 16 + (lane & 15), c * 32 + (lane >> 4) * 8);
-          s_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              q_frag[c], bfr, s_acc[t], 0, 0, 0);
-        }
-
-      float p[4][4];
-      float mx[4];
-#pragma unroll
-      for (int r = 0; r < 4; ++r) mx[r] = NEG_INF;
-      const int row_base = m0 + wid * 16 + (lane >> 4) * 4;
-#pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        const int col = n0 + t * 16 + (lane & 15);
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int row = row_base + r;
-          float s = s_acc[t][r] * scale;
-          if ((CAUSAL && col > row) || col >= S || row >= S) s = NEG_INF;
-          p[t][r] = s;
-          mx[r] = fmaxf(mx[r], s);
-        }
-      }
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          mx[r] = fmaxf(mx[r], __shfl_xor(mx[r], off, 64));
-
-      float alpha[4];
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const float mn = fmaxf(m_run[r], mx[r]);
-        alpha[r] = (m_run[r] == NEG_INF) ? 0.f : __expf(m_run[r] - mn);
-        m_run[r] = mn;
-      }
-      float rowsum[4] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int t = 0; t < 4; ++t)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const float pe =
-              (p[t][r] == NEG_INF) ? 0.f : __expf(p[t][r] - m_run[r]);
-          p[t][r] = pe;
-          rowsum[r] += pe;
-        }
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          rowsum[r] += __shfl_xor(rowsum[r], off, 64);
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        l_run[r] = l_run[r] * alpha[r] + rowsum[r];
-#pragma unroll
-      for (int t = 0; t < DT; ++t)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) o_acc[t][r] *= alpha[r];
-
-      // P -> per-wave LDS scratch (C-layout -> A-layout)
-      char* pw = p_lds + wid * 2048;
-#pragma unroll
-      for (int t = 0; t < 4; ++t)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int rl = (lane >> 4) * 4 + r;
-          const int cl = t * 16 + (lane & 15);
-          *(uint16_t*)(pw + rl * 128 + swz(rl, cl * 2)) = f2bfbits(p[t][r]);
-        }
-#pragma unroll
-      for (int kc = 0; kc < 2; ++kc) {
-        bf16x8v pa = as_frag(*(const uint4*)(
-            pw + (lane & 15) * 128 +
-            swz(lane & 15, (kc * 32 + (lane >> 4) * 8) * 2)));
-#pragma unroll
-        for (int t = 0; t < DT; ++t) {
-          bf16x8v vb = read_bfrag<128>(
-              vt_lds, t * 16 + (lane & 15), kc * 32 + (lane >> 4) * 8);
-          o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              pa, vb, o_acc[t], 0, 0, 0);
-        }
-      }
-    }
-    __syncthreads();
-  }
-
-  const int row_base = m0 + wid * 16 + (lane >> 4) * 4;
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int row = row_base + r;
-    if (row >= S) continue;
-    const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
-#pragma unroll
-    for (int t = 0; t < DT; ++t) {
-      const int col = t * 16 + (lane & 15);
-      op[(long)row * q_tok + col] = f2bf(o_acc[t][r] * inv_l);
-    }
-    if ((lane & 15) == 0)
-      lsep[row] = (l_run[r] > 0.f) ? m_run[r] + __logf(l_run[r]) : NEG_INF;
-  }
-}
-
-// ========================================================================
 // Backward preprocess: Dsum[b,h,m] = sum_d dO*O (fp32)
 // ========================================================================
 template <int D>
@@ -793,7 +628,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
       uint4 val = {0, 0, 0, 0};
       if (n0 + row < S)
         val = *(const uint4*)(vp + (long)(n0 + row) * kv_tok + cv * 8);
-      *(uint4*)(v_lds + row * (D * 2) + swz(row, cv * 16)) = val;
+      *(uint4*)(v_lds + swz_off(row, D * 2, cv * 16)) = val;
     }
   }
 

@@ -42,7 +42,11 @@ __global__ void ce_fwd_kernel(const bf16x8* __restrict__ logits,
       const float mo = __shfl_xor(m, off, WAVE);
       const float so = __shfl_xor(s, off, WAVE);
       const float mn = fmaxf(m, mo);
-      s = s * __expf(m - mn) + so * __expf(mo - mn);
+      // guard: a lane with no elements has m = -inf (exp(-inf - -inf)
+      // would be NaN)
+      const float sa = (m == -INFINITY) ? 0.f : s * __expf(m - mn);
+      const float sb = (mo == -INFINITY) ? 0.f : so * __expf(mo - mn);
+      s = sa + sb;
       m = mn;
     }
     const int wid = threadIdx.x / WAVE;
@@ -56,7 +60,10 @@ __global__ void ce_fwd_kernel(const bf16x8* __restrict__ logits,
 #pragma unroll 4
       for (int i = 1; i < BLOCK / WAVE; ++i) {
         const float mn = fmaxf(M, red_m[i]);
-        S = S * __expf(M - mn) + red_s[i] * __expf(red_m[i] - mn);
+        const float sa = (M == -INFINITY) ? 0.f : S * __expf(M - mn);
+        const float sb = (red_m[i] == -INFINITY)
+                             ? 0.f : red_s[i] * __expf(red_m[i] - mn);
+        S = sa + sb;
         M = mn;
       }
       const float l = M + __logf(S);
