@@ -83,19 +83,6 @@ struct TileStage {
     }
   }
 
-  DEVINL void write_tr(char* lds) const {
-#pragma unroll
-    for (int i = 0; i < VPT; ++i) {
-      const int vi = threadIdx.x + i * NTH;
-      const int row = vi / VPR, cv = vi % VPR;
-      const uint16_t* h = (const uint16_t*)&vals[i];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int d = cv * 8 + j;
-        *(uint16_t*)(lds + swz_off(d, 128, row * 2)) = h[j];
-      }
-    }
-  }
 };
 
 // Loads a [D][64] tile (rows = d, cols = 64 tokens) from a
