@@ -109,16 +109,34 @@ def main() -> int:
     signal.signal(signal.SIGTERM, on_term)
 
     def checkpoint_and_ack(version=None):
-        if ckpt_dir:
-            barrier(ctx)
+        # No barrier here: only rank 0 runs this (from its agent poll or
+        # its own SIGTERM) and DP replicas are identical after a step, so
+        # rank 0's tensors alone are a consistent checkpoint. A barrier
+        # would interleave with other ranks' in-flight collectives.
+        if ckpt_dir and ctx.is_main:
             trainer.save_checkpoint(ckpt_dir)
-            if ctx.is_main and agent_file and version is not None:
+            if agent_file and version is not None:
                 _atomic_write(agent_file, {
                     "ckpt-completed-version": {"version": version,
                                                "status": "Succeeded"},
                     "step": trainer.step_count,
                 })
 
+    try:
+        return _train_loop(trainer, ctx, steps_total, state_dir, job_file,
+                           agent_file, stop, checkpoint_and_ack, destroy,
+                           barrier)
+    except RuntimeError as e:
+        # A peer restarting (elastic scale / preemption) tears down the
+        # process group mid-collective; classify as retryable so the
+        # controller restarts us into the new rendezvous (exit-code
+        # contract, controlplane/failover.py).
+        print(f"[entrypoint] collective aborted: {e}", flush=True)
+        return 143
+
+
+def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
+                stop, checkpoint_and_ack, destroy, barrier):
     last_completed = None
     while trainer.step_count < steps_total:
         loss = trainer.train_step()
