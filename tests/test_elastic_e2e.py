@@ -1,0 +1,98 @@
+"""Elastic scale-out end-to-end with REAL processes (CPU, gloo):
+master + 1 worker training; mid-run scale to 2 workers runs the 2-stage
+checkpoint transaction through the state-file bridge (controller
+requests -> rank-0 agent checkpoints + acks -> controller restarts the
+gang at the new WORLD_SIZE) and the job completes with all 3 ranks.
+
+This is BASELINE config 4's shape ("elastic min->max mid-run scale-up")
+at CPU scale.
+"""
+import json
+import os
+import time
+
+import pytest
+
+from torch_on_k8s_amd.controlplane.api import (ANN_CKPT_REQUESTED,
+                                               ElasticPolicy,
+                                               JobConditionType, TaskSpec,
+                                               TaskType, TorchJob)
+from torch_on_k8s_amd.controlplane.controller import (ControllerConfig,
+                                                      JobController)
+from torch_on_k8s_amd.controlplane.elastic import ElasticScaler
+from torch_on_k8s_amd.controlplane.node import NodeState
+from torch_on_k8s_amd.controlplane.runtime import LocalProcessRuntime
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def task_env(steps):
+    return {
+        "TOK_BACKEND": "gloo",
+        "TOK_TRAIN_STEPS": str(steps),
+        "TOK_TRAINER_CONFIG": json.dumps(
+            {"model": "llama-tiny", "micro_batch": 1, "seq_len": 32,
+             "lr": 1e-3}),
+        "PYTHONPATH": ROOT,
+    }
+
+
+@pytest.mark.timeout(420)
+def test_elastic_scale_out_e2e(tmp_path):
+    node = NodeState(num_gpus=0)
+    rt = LocalProcessRuntime(str(tmp_path / "work"))
+    ctl = JobController(node, rt,
+                        ControllerConfig(enable_gang_scheduling=False),
+                        elastic=ElasticScaler())
+    steps = 60
+    job = TorchJob(
+        name="elastic-e2e",
+        tasks={
+            TaskType.MASTER: TaskSpec(replicas=1, gpus_per_task=0,
+                                      env=task_env(steps)),
+            TaskType.WORKER: TaskSpec(replicas=1, gpus_per_task=0,
+                                      env=task_env(steps)),
+        },
+        elastic=ElasticPolicy(min_replicas=1, max_replicas=2),
+    )
+    ctl.create_job(job)
+
+    # wait until training is underway
+    mpath = tmp_path / "work" / "elastic-e2e" / "metrics.json"
+    t0 = time.time()
+    while time.time() - t0 < 120:
+        ctl.reconcile(job)
+        if mpath.exists() and json.load(open(mpath))["step"] >= 3:
+            break
+        time.sleep(0.3)
+    assert mpath.exists(), "training never started"
+    assert json.load(open(mpath))["world_size"] == 2
+
+    # scale out 1 -> 2 workers (generation bump -> checkpoint transaction)
+    ElasticScaler.scale(job, 2)
+    saw_ckpt_request = False
+    t0 = time.time()
+    while time.time() - t0 < 240:
+        ctl.reconcile(job)
+        saw_ckpt_request = saw_ckpt_request or \
+            ANN_CKPT_REQUESTED in job.annotations
+        if job.status.phase in (JobConditionType.SUCCEEDED,
+                                JobConditionType.FAILED):
+            break
+        time.sleep(0.3)
+
+    if job.status.phase != JobConditionType.SUCCEEDED:
+        logdir = tmp_path / "work" / "elastic-e2e"
+        logs = "\n".join(f"== {p.name}\n{p.read_text()[-1500:]}"
+                         for p in logdir.glob("*.log"))
+        raise AssertionError(
+            f"phase={job.status.phase} events="
+            f"{[(e.reason, e.message) for e in ctl.events]}\n{logs}")
+
+    assert saw_ckpt_request, "checkpoint transaction never ran"
+    final = json.load(open(mpath))
+    assert final["step"] == steps
+    assert final["world_size"] == 3  # master + 2 workers after scale
+    # the restarted gang resumed from the checkpoint, not step 0
+    agent = json.load(open(tmp_path / "work" / "elastic-e2e" / "agent.json"))
+    assert agent["ckpt-completed-version"]["version"] == job.generation

@@ -57,10 +57,22 @@ class Trainer:
 
         mcfg = get_config(cfg.model, **cfg.model_overrides)
         self.model_cfg = mcfg
-        model = LlamaModel(mcfg, activation_checkpointing=cfg.activation_checkpointing)
         if self.device.type == "cuda" and cfg.dtype == "bf16":
-            model = model.to(torch.bfloat16)
-        model = model.to(self.device)
+            # construct + random-init directly on the GPU in bf16 (a CPU
+            # fp32 init of 8B params costs ~30s and 32 GB of host RAM)
+            prev_dtype = torch.get_default_dtype()
+            try:
+                torch.set_default_dtype(torch.bfloat16)
+                with torch.device(self.device):
+                    model = LlamaModel(
+                        mcfg,
+                        activation_checkpointing=cfg.activation_checkpointing)
+            finally:
+                torch.set_default_dtype(prev_dtype)
+        else:
+            model = LlamaModel(
+                mcfg, activation_checkpointing=cfg.activation_checkpointing)
+            model = model.to(self.device)
         self.fb = FlatBucketModel(
             model, bucket_mb=cfg.bucket_mb, overlap=cfg.overlap_grad_sync)
         self.opt = FlatAdamW(
