@@ -34,7 +34,7 @@ def parse_args():
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--model", type=str, default="llama3-8b")
-    ap.add_argument("--micro-batch", type=int, default=4)
+    ap.add_argument("--micro-batch", type=int, default=8)
     ap.add_argument("--seq-len", type=int, default=4096)
     ap.add_argument("--bucket-mb", type=int, default=256)
     ap.add_argument("--activation-checkpointing", action="store_true")

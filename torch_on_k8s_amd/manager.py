@@ -33,6 +33,9 @@ from torch_on_k8s_amd.controlplane.modelregistry import (ModelRegistry,
                                                          StorageProvider)
 from torch_on_k8s_amd.controlplane.node import NodeState
 from torch_on_k8s_amd.controlplane.runtime import LocalProcessRuntime
+from torch_on_k8s_amd.utils.logging import get_logger
+
+log = get_logger("manager")
 
 
 class Manager:
@@ -157,8 +160,8 @@ def main():
                   quotas=quotas or None, gates=gates,
                   sync_period=args.sync_period)
     start_metrics_server(args.metrics_addr)
-    print(f"[manager] workdir={args.workdir} gpus={args.num_gpus} "
-          f"gates={gates.as_dict()}", flush=True)
+    log.info("workdir=%s gpus=%d gates=%s", args.workdir, args.num_gpus,
+             gates.as_dict())
     mgr.run_forever()
 
 
