@@ -3,8 +3,7 @@ a task's processes are not created until every upstream task has its full
 replica count at (or past) the required phase."""
 from __future__ import annotations
 
-from torch_on_k8s_amd.controlplane.api import (TaskPhase, TorchJob,
-                                               phase_reached)
+from torch_on_k8s_amd.controlplane.api import TorchJob, phase_reached
 
 
 def dag_condition_ready(job: TorchJob, task_type, task_handles: dict) -> bool:

@@ -13,7 +13,7 @@ from __future__ import annotations
 import time
 
 try:
-    from prometheus_client import (Counter, Gauge, Histogram, REGISTRY,
+    from prometheus_client import (Counter, Gauge, Histogram,
                                    start_http_server)
     HAVE_PROM = True
 except ImportError:  # pragma: no cover

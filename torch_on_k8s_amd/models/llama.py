@@ -15,8 +15,7 @@ metric). Design choices for MI355X:
 """
 from __future__ import annotations
 
-import math
-from dataclasses import dataclass, field, asdict
+from dataclasses import dataclass, asdict
 
 import torch
 import torch.nn as nn
