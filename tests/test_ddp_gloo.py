@@ -71,6 +71,68 @@ def test_two_rank_training_stays_in_sync():
     assert all(l == l for l in rec["losses"])  # no NaN
 
 
+GRAD_WORKER = r"""
+import json, os, sys
+sys.path.insert(0, os.environ["TOK_ROOT"])
+import torch
+from torch_on_k8s_amd.parallel.env import init_distributed, destroy
+from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
+
+ctx = init_distributed(backend="gloo")
+cfg = TrainerConfig(model="llama-tiny", micro_batch=2, seq_len=32)
+tr = Trainer(cfg, ctx)
+tr.fb.zero_grads()
+inp, lab = tr.data.batch(0)
+loss = tr.fb(inp, lab)
+loss.backward()
+tr.fb.finish_grad_sync()
+if ctx.rank == 0:
+    torch.save([b.flat_grad.clone() for b in tr.fb.buckets],
+               os.environ["TOK_GRAD_OUT"])
+destroy()
+"""
+
+
+def test_two_rank_grads_match_single_process_sum(tmp_path):
+    """The RCCL/gloo bucket all-reduce must produce exactly the SUM of
+    per-rank gradients (the 1/world average is folded into AdamW)."""
+    gpath = str(tmp_path / "grads.pt")
+    procs = []
+    env0 = dict(os.environ, TOK_ROOT=ROOT, TOK_GRAD_OUT=gpath,
+                MASTER_ADDR="127.0.0.1", MASTER_PORT="29713",
+                WORLD_SIZE="2")
+    for r in range(2):
+        env = dict(env0, RANK=str(r), LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen(
+            [sys.executable, "-c", GRAD_WORKER], env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
+    for p in procs:
+        out, err = p.communicate(timeout=300)
+        assert p.returncode == 0, f"worker failed:\n{out}\n{err}"
+
+    # single-process reference: same init, both ranks' batches, summed
+    from torch_on_k8s_amd.engine.data import SyntheticTokens
+    from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
+    from torch_on_k8s_amd.parallel.env import DistContext
+    tr = Trainer(TrainerConfig(model="llama-tiny", micro_batch=2,
+                               seq_len=32), DistContext())
+    tr.fb.zero_grads()
+    for rank in range(2):
+        data = SyntheticTokens(tr.model_cfg.vocab_size, 2, 32,
+                               torch.device("cpu"), rank=rank,
+                               seed=tr.cfg.seed)
+        inp, lab = data.batch(0)
+        loss = tr.fb(inp, lab)
+        loss.backward()
+    ref = [b.flat_grad for b in tr.fb.buckets]
+
+    got = torch.load(gpath, weights_only=True)
+    assert len(got) == len(ref)
+    for g, r in zip(got, ref):
+        assert torch.allclose(g, r, atol=1e-5), \
+            (g - r).abs().max()
+
+
 def test_allreduce_bucket_math():
     """Direct check: flat-bucket all-reduce averages match manual DDP."""
     # single-process simulation of the averaging math
