@@ -1,0 +1,47 @@
+"""Trainer features: gradient accumulation + LR schedule."""
+import torch
+
+from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
+from torch_on_k8s_amd.parallel.env import DistContext
+
+
+def test_grad_accum_sums_micro_grads():
+    torch.manual_seed(0)
+    ctx = DistContext()
+    tr = Trainer(TrainerConfig(model="llama-tiny", micro_batch=1,
+                               seq_len=32, grad_accum_steps=2), ctx)
+    # manual: two micro backwards accumulated
+    tr.fb.zero_grads()
+    for micro in range(2):
+        inp, lab = tr.data.batch(micro)
+        tr.fb(inp, lab).backward()
+    ref = [b.flat_grad.clone() for b in tr.fb.buckets]
+
+    # trainer step does the same micro sequence (step_count=0 -> batches 0,1)
+    torch.manual_seed(0)
+    tr2 = Trainer(TrainerConfig(model="llama-tiny", micro_batch=1,
+                                seq_len=32, grad_accum_steps=2), ctx)
+    tr2.train_step()
+    # grads were consumed by the optimizer but flat_grad still holds the
+    # accumulated sum (scale folded into the kernel)
+    for b, r in zip(tr2.fb.buckets, ref):
+        assert torch.allclose(b.flat_grad, r, atol=1e-6)
+    assert tr2.tokens_per_step() == 2 * 32  # accum counted in throughput
+
+
+def test_lr_schedule_warmup_cosine():
+    ctx = DistContext()
+    tr = Trainer(TrainerConfig(model="llama-tiny", micro_batch=1, seq_len=32,
+                               lr=1.0, lr_warmup_steps=10,
+                               lr_decay_steps=110, lr_min_ratio=0.1), ctx)
+    tr.step_count = 0
+    assert abs(tr.current_lr() - 0.1) < 1e-6          # step 1 of warmup
+    tr.step_count = 4
+    assert abs(tr.current_lr() - 0.5) < 1e-6
+    tr.step_count = 9
+    assert abs(tr.current_lr() - 1.0) < 1e-6          # warmup done
+    tr.step_count = 200
+    assert abs(tr.current_lr() - 0.1) < 1e-6          # decayed to floor
+    # midpoint of cosine: (floor + lr)/2
+    tr.step_count = 59                                 # t = 50/100
+    assert abs(tr.current_lr() - 0.55) < 1e-2

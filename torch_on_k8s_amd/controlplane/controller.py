@@ -428,7 +428,6 @@ class JobController:
             if h.gpu_slots:
                 self.node.release(h.gpu_slots)
                 h.gpu_slots = ()
-        self.node.release_owner(None) if False else None
         if self.gang is not None and job.deleted:
             self.gang.delete_pod_group(job.name)
 

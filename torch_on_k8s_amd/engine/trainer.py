@@ -28,7 +28,11 @@ class TrainerConfig:
     model_overrides: dict = field(default_factory=dict)
     micro_batch: int = 2
     seq_len: int = 4096
+    grad_accum_steps: int = 1       # micro-steps per optimizer step
     lr: float = 3e-4
+    lr_warmup_steps: int = 0        # linear warmup
+    lr_decay_steps: int = 0         # cosine decay horizon (0 = constant)
+    lr_min_ratio: float = 0.1       # floor as a fraction of lr
     weight_decay: float = 0.1
     betas: tuple = (0.9, 0.95)
     grad_clip: float = 0.0          # 0 disables (extra HBM pass when on)
@@ -103,18 +107,24 @@ class Trainer:
         if self.cfg.hip_graph:
             return self._train_step_graph(sync)
         t0 = time.perf_counter()
+        accum = max(1, self.cfg.grad_accum_steps)
         self.fb.zero_grads()
-        inp, lab = self.data.batch(self.step_count)
-        loss = self.fb(inp, lab)
-        loss.backward()
+        loss = None
+        for micro in range(accum):
+            # all-reduce (and its backward overlap) fires only on the
+            # final micro-step; earlier ones just accumulate locally
+            self.fb.set_accumulate(micro < accum - 1)
+            inp, lab = self.data.batch(self.step_count * accum + micro)
+            loss = self.fb(inp, lab)
+            loss.backward()
         self.fb.finish_grad_sync()
-        scale = 1.0 / self.fb.world_size
+        scale = 1.0 / (self.fb.world_size * accum)
         if self.cfg.grad_clip > 0:
-            norm = self.fb.grad_norm()
+            norm = self.fb.grad_norm() / accum
             coef = self.cfg.grad_clip / (norm + 1e-6)
             coef = torch.clamp(coef, max=1.0)
             scale = scale * coef.item()
-        self.opt.step(grad_scale=scale)
+        self.opt.step(grad_scale=scale, lr=self.current_lr())
         self.step_count += 1
         if not sync:
             return loss.detach()
@@ -178,8 +188,24 @@ class Trainer:
             self._write_metrics(out)
         return out
 
+    def current_lr(self) -> float:
+        """Linear warmup then cosine decay to lr_min_ratio*lr."""
+        cfg = self.cfg
+        step = self.step_count + 1
+        lr = cfg.lr
+        if cfg.lr_warmup_steps and step < cfg.lr_warmup_steps:
+            return lr * step / cfg.lr_warmup_steps
+        if cfg.lr_decay_steps:
+            import math
+            t = min(1.0, (step - cfg.lr_warmup_steps) /
+                    max(1, cfg.lr_decay_steps - cfg.lr_warmup_steps))
+            floor = lr * cfg.lr_min_ratio
+            return floor + (lr - floor) * 0.5 * (1 + math.cos(math.pi * t))
+        return lr
+
     def tokens_per_step(self) -> int:
-        return self.cfg.micro_batch * self.cfg.seq_len * self.fb.world_size
+        return self.cfg.micro_batch * self.cfg.seq_len * \
+            self.fb.world_size * max(1, self.cfg.grad_accum_steps)
 
     def _write_metrics(self, loss: float):
         """Structured metrics endpoint for the elastic autoscaler

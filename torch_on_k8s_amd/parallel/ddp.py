@@ -112,6 +112,10 @@ class FlatBucketModel:
                     .view(s.param.shape)
             b.pending = len(b.segs)
 
+        # gradient accumulation: while True, backward only accumulates
+        # into the flat grads; the all-reduce fires on the final
+        # micro-step (set_accumulate(False) before it)
+        self.accumulate_only = False
         self._hooks = []
         if self.overlap and self.world_size > 1:
             for b in self.buckets:
@@ -124,9 +128,14 @@ class FlatBucketModel:
         def hook(_param):
             bucket.pending -= 1
             if bucket.pending == 0:
-                bucket.work = dist.all_reduce(
-                    bucket.flat_grad, group=self.group, async_op=True)
+                bucket.pending = len(bucket.segs)
+                if not self.accumulate_only:
+                    bucket.work = dist.all_reduce(
+                        bucket.flat_grad, group=self.group, async_op=True)
         return hook
+
+    def set_accumulate(self, accumulate: bool):
+        self.accumulate_only = accumulate
 
     # -- training-step API --------------------------------------------
     def zero_grads(self):
