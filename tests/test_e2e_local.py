@@ -128,3 +128,29 @@ def test_worker_process_failure_recovers(tmp_path):
     assert job.status.phase == JobConditionType.SUCCEEDED
     final = json.load(open(mpath))
     assert final["step"] == 30
+
+
+@pytest.mark.timeout(120)
+def test_arbitrary_command_task(tmp_path):
+    """Parity with the reference's opaque-container model: a TaskSpec may
+    run ANY command, not just the framework entrypoint."""
+    node = NodeState(num_gpus=0)
+    rt = LocalProcessRuntime(str(tmp_path / "work"))
+    ctl = JobController(node, rt,
+                        ControllerConfig(enable_gang_scheduling=False))
+    import sys as _sys
+    job = TorchJob(
+        name="custom-cmd",
+        tasks={
+            TaskType.MASTER: TaskSpec(
+                replicas=1, gpus_per_task=0,
+                command=[_sys.executable, "-c",
+                         "import os; print('RANK', os.environ['RANK']); "
+                         "open(os.environ['TOK_STATE_DIR'] + '/ok', 'w')"
+                         ".write('done')"]),
+        },
+    )
+    ctl.create_job(job)
+    drive(ctl, job, timeout=60)
+    assert job.status.phase == JobConditionType.SUCCEEDED
+    assert (tmp_path / "work" / "custom-cmd" / "ok").read_text() == "done"
