@@ -153,3 +153,45 @@ def test_trainer_single_gpu_step():
         tr.step_count = 1
     assert all(l == l for l in losses)
     assert losses[-1] < losses[0], losses
+
+
+def test_layernorm_gpu():
+    torch.manual_seed(0)
+    x = torch.randn(256, 1024).bfloat16().float()
+    w = torch.randn(1024).bfloat16().float()
+    b = torch.randn(1024).bfloat16().float()
+    dy = torch.randn(256, 1024).bfloat16().float()
+
+    xr = x.clone().requires_grad_(True)
+    wr = w.clone().requires_grad_(True)
+    br = b.clone().requires_grad_(True)
+    yr = torch.nn.functional.layer_norm(xr, (1024,), wr, br, 1e-5)
+    yr.backward(dy)
+
+    xg = x.bfloat16().to(dev()).requires_grad_(True)
+    wg = w.bfloat16().to(dev()).requires_grad_(True)
+    bg = b.bfloat16().to(dev()).requires_grad_(True)
+    yg = ops.layernorm(xg, wg, bg, 1e-5)
+    yg.backward(dy.bfloat16().to(dev()))
+
+    assert (yg.float().cpu() - yr.detach()).abs().max() < 0.05
+    assert (xg.grad.float().cpu() - xr.grad).abs().max() < 0.05
+    assert (wg.grad.float().cpu() - wr.grad).abs().max() < \
+        0.05 * max(1.0, wr.grad.abs().max().item())
+    assert (bg.grad.float().cpu() - br.grad).abs().max() < \
+        0.05 * max(1.0, br.grad.abs().max().item())
+
+
+def test_gpt2_trainer_gpu_step():
+    from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
+    from torch_on_k8s_amd.parallel.env import DistContext
+    torch.manual_seed(0)
+    tr = Trainer(TrainerConfig(model="gpt2-tiny", micro_batch=2, seq_len=128,
+                               lr=1e-3), DistContext(device=dev()))
+    losses = []
+    for _ in range(5):
+        tr.step_count = 0
+        losses.append(tr.train_step())
+        tr.step_count = 1
+    assert all(l == l for l in losses)
+    assert losses[-1] < losses[0], losses

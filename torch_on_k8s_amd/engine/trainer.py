@@ -17,7 +17,7 @@ from dataclasses import dataclass, field, asdict
 import torch
 
 from torch_on_k8s_amd.engine.data import SyntheticTokens
-from torch_on_k8s_amd.models.llama import LlamaModel, get_config
+from torch_on_k8s_amd.models.registry import build_model, get_model_config
 from torch_on_k8s_amd.parallel.ddp import FlatBucketModel, FlatAdamW
 from torch_on_k8s_amd.parallel.env import DistContext
 
@@ -59,7 +59,7 @@ class Trainer:
         self.device = ctx.device
         torch.manual_seed(cfg.seed)  # same init on every rank
 
-        mcfg = get_config(cfg.model, **cfg.model_overrides)
+        mcfg = get_model_config(cfg.model, **cfg.model_overrides)
         self.model_cfg = mcfg
         if self.device.type == "cuda" and cfg.dtype == "bf16":
             # construct + random-init directly on the GPU in bf16 (a CPU
@@ -68,13 +68,13 @@ class Trainer:
             try:
                 torch.set_default_dtype(torch.bfloat16)
                 with torch.device(self.device):
-                    model = LlamaModel(
+                    model = build_model(
                         mcfg,
                         activation_checkpointing=cfg.activation_checkpointing)
             finally:
                 torch.set_default_dtype(prev_dtype)
         else:
-            model = LlamaModel(
+            model = build_model(
                 mcfg, activation_checkpointing=cfg.activation_checkpointing)
             model = model.to(self.device)
         self.fb = FlatBucketModel(
@@ -94,7 +94,7 @@ class Trainer:
             assert self.device.type == "cuda", "hip_graph needs a GPU"
 
     @property
-    def module(self) -> LlamaModel:
+    def module(self):
         return self.fb.module
 
     def train_step(self, sync: bool = True):

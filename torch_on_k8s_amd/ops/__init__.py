@@ -200,6 +200,51 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
 
 # --------------------------------------------------------------------------
+# LayerNorm (GPT-family normalization)
+# --------------------------------------------------------------------------
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, eps):
+        if x.is_cuda:
+            y, mu, rstd = _require_ext("layernorm").layernorm_fwd(x, w, b, eps)
+        else:
+            xf = x.float()
+            mu = xf.mean(-1)
+            var = xf.var(-1, unbiased=False)
+            rstd = torch.rsqrt(var + eps)
+            y = (((xf - mu.unsqueeze(-1)) * rstd.unsqueeze(-1)) * w.float() +
+                 b.float()).to(x.dtype)
+            mu, rstd = mu.reshape(-1), rstd.reshape(-1)
+        ctx.save_for_backward(x, w, mu, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, mu, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if x.is_cuda:
+            dx, dw, db = _C.layernorm_bwd(x, w, dy, mu, rstd)
+        else:
+            H = x.shape[-1]
+            xf, wf, dyf = x.float(), w.float(), dy.float()
+            muv = mu.view(*x.shape[:-1], 1)
+            rs = rstd.view(*x.shape[:-1], 1)
+            xhat = (xf - muv) * rs
+            dyw = dyf * wf
+            a1 = dyw.mean(-1, keepdim=True)
+            a2 = (dyw * xhat).mean(-1, keepdim=True)
+            dx = (rs * (dyw - a1 - xhat * a2)).to(x.dtype)
+            dw = (dyf * xhat).reshape(-1, H).sum(0)
+            db = dyf.reshape(-1, H).sum(0)
+        return dx, dw.to(w.dtype), db.to(w.dtype), None
+
+
+def layernorm(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
+              eps: float = 1e-5) -> torch.Tensor:
+    return _LayerNormFn.apply(x.contiguous(), w, b, eps)
+
+
+# --------------------------------------------------------------------------
 # Fused cross-entropy (mean reduction over all tokens)
 # --------------------------------------------------------------------------
 class _FusedCEFn(torch.autograd.Function):
@@ -231,5 +276,6 @@ def cross_entropy(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
 
 __all__ = [
     "rmsnorm", "rmsnorm_ref", "apply_rope", "rope_ref", "fused_adamw_",
-    "attention", "attention_ref", "cross_entropy", "hip_ext_available",
+    "attention", "attention_ref", "cross_entropy", "layernorm",
+    "hip_ext_available",
 ]

@@ -35,6 +35,13 @@ hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
                         hipStream_t stream);
 hipError_t tok_transpose_head(const void* in, void* out, int B, int S, int H,
                               int D, int S_pad, hipStream_t stream);
+hipError_t tok_layernorm_fwd(const void* x, const void* w, const void* b,
+                             void* y, float* mu, float* rstd, long nrows,
+                             int H, float eps, hipStream_t stream);
+hipError_t tok_layernorm_bwd(const void* x, const void* w, const void* dy,
+                             const float* mu, const float* rstd, void* dx,
+                             float* dw, float* db, long nrows, int H,
+                             hipStream_t stream);
 hipError_t tok_ce_fwd(const void* logits, const int* labels, float* loss,
                       float* lse, long rows, int V, hipStream_t stream);
 hipError_t tok_ce_bwd(const void* logits, const int* labels, const float* lse,
@@ -182,6 +189,41 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {dq, dk, dv};
 }
 
+std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w,
+                                      at::Tensor b, double eps) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(w);
+  CHECK_BF16_CUDA(b);
+  const long H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "hidden dim must be a multiple of 8");
+  const long nrows = x.numel() / H;
+  auto y = at::empty_like(x);
+  auto mu = at::empty({nrows}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({nrows}, x.options().dtype(at::kFloat));
+  TOK_HIP_OK(tok_layernorm_fwd(x.data_ptr(), w.data_ptr(), b.data_ptr(),
+                               y.data_ptr(), mu.data_ptr<float>(),
+                               rstd.data_ptr<float>(), nrows, (int)H,
+                               (float)eps, current_stream()));
+  return {y, mu, rstd};
+}
+
+std::vector<at::Tensor> layernorm_bwd(at::Tensor x, at::Tensor w,
+                                      at::Tensor dy, at::Tensor mu,
+                                      at::Tensor rstd) {
+  CHECK_BF16_CUDA(x);
+  const long H = x.size(-1);
+  const long nrows = x.numel() / H;
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto db = at::zeros({H}, x.options().dtype(at::kFloat));
+  TOK_HIP_OK(tok_layernorm_bwd(x.data_ptr(), w.data_ptr(), dy.data_ptr(),
+                               mu.data_ptr<float>(), rstd.data_ptr<float>(),
+                               dx.data_ptr(), dw.data_ptr<float>(),
+                               db.data_ptr<float>(), nrows, (int)H,
+                               current_stream()));
+  return {dx, dw, db};
+}
+
 std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels) {
   CHECK_BF16_CUDA(logits);
   TORCH_CHECK(labels.scalar_type() == at::kInt && labels.is_cuda());
@@ -236,6 +278,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_fwd", &attn_fwd, "Flash attention forward (bf16, gfx950)");
   mod.def("attn_bwd", &attn_bwd, "Flash attention backward (bf16, gfx950)");
   mod.def("ce_fwd", &ce_fwd, "Fused cross-entropy forward (bf16, gfx950)");
+  mod.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (bf16, gfx950)");
+  mod.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (bf16, gfx950)");
   mod.def("ce_bwd", &ce_bwd, "Fused cross-entropy backward (bf16, gfx950)");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16, gfx950)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (bf16, gfx950)");
