@@ -1,27 +1,34 @@
-// Flash attention (causal, GQA) for MI355X (gfx950) — forward + backward.
+// Flash attention (causal, GQA/MHA) for MI355X (gfx950) - fwd + bwd.
 //
-// v2 structure (profiling-driven, see profiles/: v1 was 47% of the 8B
-// step at ~100 TF):
-//  * 8-wave blocks (512 threads): fwd/dq tile 128 q rows x 64 kv,
-//    dkdv 128 kv rows x 64 q — tile staging and barriers amortized over
-//    2x the rows of v1.
-//  * async-STAGE split (guide §6 G15 / T14, +17% on attention): the next
-//    KV tile's global loads are issued into registers during the current
-//    tile's MFMA phase; the LDS write lands after the barrier.
-//  * MFMA 16x16x32 bf16; fragment layouts verified on HW by mfma_probe:
-//      A[m][k]: m=lane&15, k=(lane>>4)*8+j
-//      B[k][n]: k=(lane>>4)*8+j, n=lane&15
-//      C/D[m][n]: m=(lane>>4)*4+r, n=lane&15
-//  * LDS XOR swizzle byte^=(row&7)<<4 on every staged tile (row-major
-//    [*][128] bf16 read by ds_read_b128 is otherwise up to 16-way bank
-//    conflict, guide §6 G4).
-//  * V (and Q/dO transposes in backward) staged TRANSPOSED so every
-//    MFMA B-fragment read is one contiguous ds_read_b128.
-//  * online softmax fully in registers; 16-lane shuffle row reduce.
+// Architecture (v3; progression + measurements in profiles/README.md):
+//  * mfma_f32_32x32x16_bf16 tiles with SWAPPED operands: S^T = K*Q^T so
+//    each lane owns ONE q row; online softmax is fully in-register (an
+//    in-lane chain + one half-wave shuffle; no cross-lane reduce, no P
+//    round trip through LDS). Fragment layouts verified on hardware by
+//    mfma_probe.hip.
+//  * P / dS^T are repacked from the MFMA C-layout to the next MFMA's
+//    A-layout with v_cvt_pk_bf16_f32 pairs + permlane32_swap (guide T12):
+//    C-layout element r of half h sits at k=(r&3)+8*(r>>2)+4h; one swap
+//    fixes two A-fragment words.
+//  * Q (fwd), K/V (dq) and K (dv/dk) live in registers as the MFMA
+//    B-operand; tiles consumed d-major (V fwd, K^T/Q^T/dO^T bwd) are
+//    read from PRE-TRANSPOSED [B,H,D,S_pad] planes (transpose.hip) so
+//    all LDS staging is vectorized b128 writes - the in-kernel scalar
+//    transpose scatter measured ~20 LDS bank-conflict cycles per MFMA.
+//  * LDS tiles are double-buffered (one barrier per tile; the next
+//    tile's write overlaps other waves' compute) and XOR-swizzled on the
+//    FULL tile offset ((row&15)<<4: conflict-free 16-lane b128 groups;
+//    bit 7 swaps row parity in-tile for 128-B rows).
+//  * async-stage prefetch (T14), s_setprio around MFMA clusters (T5),
+//    defer-max online softmax (T13), interior tiles skip masking.
+//  * Backward: flash recompute scheme in three kernels - dV and dK split
+//    (a combined kernel needs 472 regs = 1 wave/SIMD and measured
+//    slower; split keeps 2 waves/SIMD), dQ over q tiles, plus a
+//    Dsum = rowsum(dO*O) preprocess. All kernels: 0 scratch spill.
 //
-// Backward: standard two-kernel flash recompute scheme — dkdv over KV
-// tiles computes S^T directly (A=K,B=Q: no lse/Dsum transpose), dq over
-// Q tiles; plus a Dsum = rowsum(dO*O) preprocess.
+// Measured (B=2 S=4096 Hq=32 Hkv=8 D=128 causal, random data): fwd
+// 358-371 TF (parity with torch's aotriton flash), bwd 194 TF incl. the
+// three operand transposes.
 #include "common.h"
 
 namespace {
