@@ -207,6 +207,71 @@ class LlamaModel(nn.Module):
             return logits
         return ops.cross_entropy(logits, labels)
 
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
+                 temperature: float = 0.0) -> torch.Tensor:
+        """Greedy (temperature=0) or sampled autoregressive generation —
+        the serving path. Prefill runs through the flash-attention
+        kernel; each decode step uses the KV-cache single-token decode
+        kernel (ops.attention_decode)."""
+        cfg = self.cfg
+        B, S0 = input_ids.shape
+        dev = input_ids.device
+        dtype = self.embed_tokens.weight.dtype
+        Tmax = S0 + max_new_tokens
+        Hkv, Hq, D = cfg.num_kv_heads, cfg.num_heads, cfg.head_dim
+        kc = [torch.zeros(B, Tmax, Hkv, D, dtype=dtype, device=dev)
+              for _ in self.layers]
+        vc = [torch.zeros(B, Tmax, Hkv, D, dtype=dtype, device=dev)
+              for _ in self.layers]
+        cos, sin = build_rope_table(cfg, Tmax, dev)
+
+        def run_block(blk, i, x, pos0, T):
+            """One block over x ([B,s,H]); writes this slice's K/V into
+            the caches and attends over cache[:T]."""
+            B_, s, _ = x.shape
+            h = blk.input_norm(x)
+            q = blk.attn.q_proj(h).view(B_, s, Hq, D)
+            k = blk.attn.k_proj(h).view(B_, s, Hkv, D)
+            v = blk.attn.v_proj(h).view(B_, s, Hkv, D)
+            q = ops.apply_rope(q, cos[pos0:pos0 + s].contiguous(),
+                               sin[pos0:pos0 + s].contiguous())
+            k = ops.apply_rope(k, cos[pos0:pos0 + s].contiguous(),
+                               sin[pos0:pos0 + s].contiguous())
+            kc[i][:, pos0:pos0 + s] = k
+            vc[i][:, pos0:pos0 + s] = v
+            if s > 1:  # prefill
+                o = ops.attention(q, k, v, causal=True)
+            else:      # single-token decode over the cache
+                o = ops.attention_decode(q[:, 0], kc[i], vc[i], T)
+                o = o.view(B_, 1, Hq, D)
+            x = x + blk.attn.o_proj(o.reshape(B_, s, Hq * D))
+            x = x + blk.mlp(blk.post_attn_norm(x))
+            return x
+
+        def next_token(logits):
+            if temperature and temperature > 0:
+                probs = torch.softmax(logits.float() / temperature, dim=-1)
+                return torch.multinomial(probs, 1)
+            return logits.argmax(-1, keepdim=True)
+
+        # prefill
+        x = self.embed_tokens(input_ids)
+        for i, blk in enumerate(self.layers):
+            x = run_block(blk, i, x, 0, S0)
+        logits = self.lm_head(self.norm(x[:, -1:]))[:, -1]
+        tokens = [next_token(logits)]
+
+        # decode
+        for step in range(1, max_new_tokens):
+            pos = S0 + step - 1
+            x = self.embed_tokens(tokens[-1])
+            for i, blk in enumerate(self.layers):
+                x = run_block(blk, i, x, pos, pos + 1)
+            logits = self.lm_head(self.norm(x))[:, -1]
+            tokens.append(next_token(logits))
+        return torch.cat([input_ids] + tokens, dim=1)
+
     def num_params(self) -> int:
         seen, total = set(), 0
         for p in self.parameters():

@@ -200,6 +200,27 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
 
 # --------------------------------------------------------------------------
+# KV-cache attention decode (serving path; no autograd)
+# --------------------------------------------------------------------------
+@torch.no_grad()
+def attention_decode(q: torch.Tensor, kcache: torch.Tensor,
+                     vcache: torch.Tensor, T: int) -> torch.Tensor:
+    """q: [B, Hq, D] new-token queries (post-RoPE); kcache/vcache:
+    [B, Tmax, Hkv, D] with the first T rows valid. Returns [B, Hq, D]."""
+    if q.is_cuda:
+        return _require_ext("attention_decode").attn_decode(
+            q.contiguous(), kcache, vcache, T)
+    B, Hq, D = q.shape
+    Hkv = kcache.shape[2]
+    rep = Hq // Hkv
+    kf = kcache[:, :T].float().permute(0, 2, 1, 3).repeat_interleave(rep, 1)
+    vf = vcache[:, :T].float().permute(0, 2, 1, 3).repeat_interleave(rep, 1)
+    s = torch.einsum("bhd,bhtd->bht", q.float(), kf) / (D ** 0.5)
+    p = torch.softmax(s, dim=-1)
+    return torch.einsum("bht,bhtd->bhd", p, vf).to(q.dtype)
+
+
+# --------------------------------------------------------------------------
 # LayerNorm (GPT-family normalization)
 # --------------------------------------------------------------------------
 class _LayerNormFn(torch.autograd.Function):

@@ -35,6 +35,9 @@ hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
                         hipStream_t stream);
 hipError_t tok_transpose_head(const void* in, void* out, int B, int S, int H,
                               int D, int S_pad, hipStream_t stream);
+hipError_t tok_attn_decode(const void* q, const void* k, const void* v,
+                           void* out, int B, int T, int Tmax, int Hq,
+                           int Hkv, int D, hipStream_t stream);
 hipError_t tok_layernorm_fwd(const void* x, const void* w, const void* b,
                              void* y, float* mu, float* rstd, long nrows,
                              int H, float eps, hipStream_t stream);
@@ -189,6 +192,23 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {dq, dk, dv};
 }
 
+// q: [B, Hq, D]; kcache/vcache: [B, Tmax, Hkv, D] (first T rows valid)
+at::Tensor attn_decode(at::Tensor q, at::Tensor kcache, at::Tensor vcache,
+                       long T) {
+  CHECK_BF16_CUDA(q);
+  CHECK_BF16_CUDA(kcache);
+  CHECK_BF16_CUDA(vcache);
+  const int B = q.size(0), Hq = q.size(1), D = q.size(2);
+  const int Tmax = kcache.size(1), Hkv = kcache.size(2);
+  TORCH_CHECK(T >= 1 && T <= Tmax);
+  TORCH_CHECK(D == 64 || D == 128);
+  auto out = at::empty_like(q);
+  TOK_HIP_OK(tok_attn_decode(q.data_ptr(), kcache.data_ptr(),
+                             vcache.data_ptr(), out.data_ptr(), B, (int)T,
+                             Tmax, Hq, Hkv, D, current_stream()));
+  return out;
+}
+
 std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w,
                                       at::Tensor b, double eps) {
   CHECK_BF16_CUDA(x);
@@ -279,6 +299,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_bwd", &attn_bwd, "Flash attention backward (bf16, gfx950)");
   mod.def("ce_fwd", &ce_fwd, "Fused cross-entropy forward (bf16, gfx950)");
   mod.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (bf16, gfx950)");
+  mod.def("attn_decode", &attn_decode,
+          "Single-token KV-cache attention decode (bf16, gfx950)");
   mod.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (bf16, gfx950)");
   mod.def("ce_bwd", &ce_bwd, "Fused cross-entropy backward (bf16, gfx950)");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16, gfx950)");
