@@ -209,7 +209,8 @@ class LlamaModel(nn.Module):
 
     @torch.no_grad()
     def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
-                 temperature: float = 0.0) -> torch.Tensor:
+                 temperature: float = 0.0,
+                 use_graph: bool | None = None) -> torch.Tensor:
         """Greedy (temperature=0) or sampled autoregressive generation —
         the serving path. Prefill runs through the flash-attention
         kernel; each decode step uses the KV-cache single-token decode
@@ -262,14 +263,68 @@ class LlamaModel(nn.Module):
         logits = self.lm_head(self.norm(x[:, -1:]))[:, -1]
         tokens = [next_token(logits)]
 
-        # decode
-        for step in range(1, max_new_tokens):
-            pos = S0 + step - 1
-            x = self.embed_tokens(tokens[-1])
+        # Greedy GPU decode is launch-bound (hundreds of tiny kernels per
+        # token); capture one whole decode step in a hipGraph and replay
+        # it per token. Position/length live on-device (index_copy_/
+        # index_select + the decode kernel's T_dev) so replay stays
+        # correct; the graph is self-feeding (argmax writes the token
+        # buffer the next replay embeds).
+        if use_graph is None:
+            use_graph = (dev.type == "cuda" and not temperature and
+                         max_new_tokens >= 8)
+
+        if not use_graph:
+            for step in range(1, max_new_tokens):
+                pos = S0 + step - 1
+                x = self.embed_tokens(tokens[-1])
+                for i, blk in enumerate(self.layers):
+                    x = run_block(blk, i, x, pos, pos + 1)
+                logits = self.lm_head(self.norm(x))[:, -1]
+                tokens.append(next_token(logits))
+            return torch.cat([input_ids] + tokens, dim=1)
+
+        token_buf = tokens[0].clone()                       # [B, 1]
+        pos_long = torch.tensor([S0], dtype=torch.long, device=dev)
+        t32 = torch.zeros((), dtype=torch.int32, device=dev)
+
+        def decode_step():
+            t32.copy_((pos_long[0] + 1).to(torch.int32))
+            x = self.embed_tokens(token_buf)
+            cs = torch.index_select(cos, 0, pos_long).contiguous()
+            sn = torch.index_select(sin, 0, pos_long).contiguous()
             for i, blk in enumerate(self.layers):
-                x = run_block(blk, i, x, pos, pos + 1)
+                h = blk.input_norm(x)
+                q = blk.attn.q_proj(h).view(B, 1, Hq, D)
+                k = blk.attn.k_proj(h).view(B, 1, Hkv, D)
+                v = blk.attn.v_proj(h).view(B, 1, Hkv, D)
+                q = ops.apply_rope(q, cs, sn)
+                k = ops.apply_rope(k, cs, sn)
+                kc[i].index_copy_(1, pos_long, k)
+                vc[i].index_copy_(1, pos_long, v)
+                o = ops.attention_decode(q[:, 0], kc[i], vc[i], 0, T_dev=t32)
+                x = x + blk.attn.o_proj(o.view(B, 1, Hq * D))
+                x = x + blk.mlp(blk.post_attn_norm(x))
             logits = self.lm_head(self.norm(x))[:, -1]
-            tokens.append(next_token(logits))
+            nxt = logits.argmax(-1, keepdim=True)
+            token_buf.copy_(nxt)       # self-feed the next replay
+            pos_long.add_(1)
+
+        graph = None
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        n_warm = min(2, max_new_tokens - 1)
+        with torch.cuda.stream(side):
+            for _ in range(n_warm):    # real decode steps (allocator/rng warm)
+                decode_step()
+                tokens.append(token_buf.clone())
+        torch.cuda.current_stream().wait_stream(side)
+        if max_new_tokens - 1 > n_warm:
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                decode_step()          # capture only; not executed
+            for _ in range(max_new_tokens - 1 - n_warm):
+                graph.replay()
+                tokens.append(token_buf.clone())
         return torch.cat([input_ids] + tokens, dim=1)
 
     def num_params(self) -> int:

@@ -204,12 +204,14 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 # --------------------------------------------------------------------------
 @torch.no_grad()
 def attention_decode(q: torch.Tensor, kcache: torch.Tensor,
-                     vcache: torch.Tensor, T: int) -> torch.Tensor:
+                     vcache: torch.Tensor, T: int,
+                     T_dev: torch.Tensor | None = None) -> torch.Tensor:
     """q: [B, Hq, D] new-token queries (post-RoPE); kcache/vcache:
-    [B, Tmax, Hkv, D] with the first T rows valid. Returns [B, Hq, D]."""
+    [B, Tmax, Hkv, D] with the first T rows valid. Returns [B, Hq, D].
+    T_dev (int32 device scalar) overrides T under hipGraph replay."""
     if q.is_cuda:
         return _require_ext("attention_decode").attn_decode(
-            q.contiguous(), kcache, vcache, T)
+            q.contiguous(), kcache, vcache, T, T_dev)
     B, Hq, D = q.shape
     Hkv = kcache.shape[2]
     rep = Hq // Hkv

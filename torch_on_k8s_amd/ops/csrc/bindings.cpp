@@ -36,8 +36,9 @@ hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
 hipError_t tok_transpose_head(const void* in, void* out, int B, int S, int H,
                               int D, int S_pad, hipStream_t stream);
 hipError_t tok_attn_decode(const void* q, const void* k, const void* v,
-                           void* out, int B, int T, int Tmax, int Hq,
-                           int Hkv, int D, hipStream_t stream);
+                           void* out, int B, int T, const int* T_dev,
+                           int Tmax, int Hq, int Hkv, int D,
+                           hipStream_t stream);
 hipError_t tok_layernorm_fwd(const void* x, const void* w, const void* b,
                              void* y, float* mu, float* rstd, long nrows,
                              int H, float eps, hipStream_t stream);
@@ -194,18 +195,23 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
 
 // q: [B, Hq, D]; kcache/vcache: [B, Tmax, Hkv, D] (first T rows valid)
 at::Tensor attn_decode(at::Tensor q, at::Tensor kcache, at::Tensor vcache,
-                       long T) {
+                       long T, c10::optional<at::Tensor> T_dev) {
   CHECK_BF16_CUDA(q);
   CHECK_BF16_CUDA(kcache);
   CHECK_BF16_CUDA(vcache);
   const int B = q.size(0), Hq = q.size(1), D = q.size(2);
   const int Tmax = kcache.size(1), Hkv = kcache.size(2);
-  TORCH_CHECK(T >= 1 && T <= Tmax);
+  TORCH_CHECK(T_dev.has_value() || (T >= 1 && T <= Tmax));
   TORCH_CHECK(D == 64 || D == 128);
+  const int* td = nullptr;
+  if (T_dev.has_value()) {
+    TORCH_CHECK(T_dev->scalar_type() == at::kInt && T_dev->is_cuda());
+    td = T_dev->data_ptr<int>();
+  }
   auto out = at::empty_like(q);
   TOK_HIP_OK(tok_attn_decode(q.data_ptr(), kcache.data_ptr(),
                              vcache.data_ptr(), out.data_ptr(), B, (int)T,
-                             Tmax, Hq, Hkv, D, current_stream()));
+                             td, Tmax, Hq, Hkv, D, current_stream()));
   return out;
 }
 
@@ -300,7 +306,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ce_fwd", &ce_fwd, "Fused cross-entropy forward (bf16, gfx950)");
   mod.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (bf16, gfx950)");
   mod.def("attn_decode", &attn_decode,
-          "Single-token KV-cache attention decode (bf16, gfx950)");
+          "Single-token KV-cache attention decode (bf16, gfx950)",
+          py::arg("q"), py::arg("kcache"), py::arg("vcache"), py::arg("T"),
+          py::arg("T_dev") = py::none());
   mod.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (bf16, gfx950)");
   mod.def("ce_bwd", &ce_bwd, "Fused cross-entropy backward (bf16, gfx950)");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16, gfx950)");

@@ -19,7 +19,8 @@ template <int D>
 __global__ __launch_bounds__(NTH_DEC) void attn_decode_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, bf16_t* __restrict__ out, int B, int T,
-    int Tmax, int Hq, int Hkv, float scale) {
+    const int* __restrict__ T_dev, int Tmax, int Hq, int Hkv, float scale) {
+  if (T_dev != nullptr) T = *T_dev;  // hipGraph replay: T lives on-device
   constexpr int EPL = D / WAVE;  // elems per lane (2 for D=128, 1 for 64)
   // per-wave partial state: o slab [NW][D] f32 + (m, l) pairs
   __shared__ __attribute__((aligned(16))) float o_slab[NW_DEC][D];
@@ -100,18 +101,19 @@ __global__ __launch_bounds__(NTH_DEC) void attn_decode_kernel(
 extern "C" {
 
 hipError_t tok_attn_decode(const void* q, const void* k, const void* v,
-                           void* out, int B, int T, int Tmax, int Hq,
-                           int Hkv, int D, hipStream_t stream) {
+                           void* out, int B, int T, const int* T_dev,
+                           int Tmax, int Hq, int Hkv, int D,
+                           hipStream_t stream) {
   dim3 grid(Hq, B);
   const float scale = 1.f / sqrtf((float)D);
   if (D == 128)
     attn_decode_kernel<128><<<grid, NTH_DEC, 0, stream>>>(
         (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (bf16_t*)out,
-        B, T, Tmax, Hq, Hkv, scale);
+        B, T, T_dev, Tmax, Hq, Hkv, scale);
   else if (D == 64)
     attn_decode_kernel<64><<<grid, NTH_DEC, 0, stream>>>(
         (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (bf16_t*)out,
-        B, T, Tmax, Hq, Hkv, scale);
+        B, T, T_dev, Tmax, Hq, Hkv, scale);
   else
     return hipErrorInvalidValue;
   return hipGetLastError();
