@@ -19,6 +19,7 @@ import os
 import time
 
 from torch_on_k8s_amd.controlplane import features as feat
+from torch_on_k8s_amd.controlplane.api import JobConditionType
 from torch_on_k8s_amd.controlplane.controller import (ControllerConfig,
                                                       JobController)
 from torch_on_k8s_amd.controlplane.coordinator import Coordinator
@@ -118,9 +119,7 @@ class Manager:
         for job in list(self.controller.jobs.values()):
             if job.elastic is None:
                 continue
-            st = job.status
-            from torch_on_k8s_amd.controlplane.api import JobConditionType
-            if st.phase != JobConditionType.RUNNING:
+            if job.status.phase != JobConditionType.RUNNING:
                 continue
             self.autoscaler.observe(job)
             self.autoscaler.decide(job)
