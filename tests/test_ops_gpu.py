@@ -195,3 +195,14 @@ def test_gpt2_trainer_gpu_step():
         tr.step_count = 1
     assert all(l == l for l in losses)
     assert losses[-1] < losses[0], losses
+
+
+def test_native_extension_is_in_tree():
+    """The gfx950 extension must be the in-tree build (not a site-packages
+    or JIT-cache copy) so repo snapshots carry the exact code under test."""
+    import os
+    import torch_on_k8s_amd
+    pkg_root = os.path.dirname(os.path.dirname(
+        os.path.abspath(torch_on_k8s_amd.__file__)))
+    assert ops._C.__file__.startswith(pkg_root), ops._C.__file__
+    assert "site-packages" not in ops._C.__file__
