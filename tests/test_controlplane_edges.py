@@ -128,3 +128,58 @@ def test_master_port_stable_across_restart():
     ctl.reconcile(job)  # restart
     port2 = rt.tasks[("pjob", TaskType.MASTER, 0)].env["MASTER_PORT"]
     assert port1 == port2
+
+
+def test_restart_reuses_gpu_slots():
+    """In-place-restart parity: a retryable failure restarts the task on
+    the SAME GPU slots (locality; no gang reshuffle)."""
+    ctl, node, rt = mk_ctl()
+    job = TorchJob(name="slotjob",
+                   tasks={TaskType.WORKER: TaskSpec(replicas=2,
+                                                    gpus_per_task=2)})
+    ctl.create_job(job)
+    ctl.reconcile(job)
+    h = ctl.handles["slotjob"][("slotjob", TaskType.WORKER, 1)]
+    slots_before = h.gpu_slots
+    assert len(slots_before) == 2
+    rt.set_phase(h.key, TaskPhase.FAILED, exit_code=137)
+    ctl.reconcile(job)
+    h2 = ctl.handles["slotjob"][("slotjob", TaskType.WORKER, 1)]
+    assert h2 is not h
+    assert h2.gpu_slots == slots_before
+    assert h2.restart_count == 1
+
+
+def test_manager_spec_update_scales_replicas(tmp_path):
+    """Editing a spooled job YAML (replica change) bumps the generation
+    (OnOwnerUpdate analog -> elastic scale path)."""
+    import os
+    import time as _time
+    from torch_on_k8s_amd.manager import Manager
+    from torch_on_k8s_amd.controlplane import features as feat
+    mgr = Manager(str(tmp_path), num_gpus=8,
+                  gates=feat.FeatureGates({"JobCoordinator": False}))
+    spec = """
+kind: TorchJob
+metadata: {name: upd-job}
+spec:
+  tasks:
+    worker: {replicas: 2, gpusPerTask: 1, command: ["/bin/sleep", "300"]}
+"""
+    path = os.path.join(mgr.spool, "upd.yaml")
+    with open(path, "w") as f:
+        f.write(spec)
+    mgr.step()
+    job = mgr.controller.jobs["upd-job"]
+    gen0 = job.generation
+    assert job.tasks[TaskType.WORKER].replicas == 2
+
+    _time.sleep(0.02)
+    with open(path, "w") as f:
+        f.write(spec.replace("replicas: 2", "replicas: 4"))
+    os.utime(path)  # ensure a fresh mtime
+    mgr.step()
+    assert job.tasks[TaskType.WORKER].replicas == 4
+    assert job.generation == gen0 + 1
+    # cleanup child sleep processes
+    mgr.controller.delete_job("upd-job")

@@ -65,3 +65,45 @@ class TorchJobClient:
             time.sleep(poll)
         raise TimeoutError(f"job {name} did not reach {phases}; "
                            f"last status: {self.get(name)}")
+
+
+def main():
+    """kubectl-style CLI over the manager workdir:
+        python -m torch_on_k8s_amd.client apply job.yaml
+        python -m torch_on_k8s_amd.client get|delete|wait <name>
+        python -m torch_on_k8s_amd.client list
+    """
+    import argparse
+    ap = argparse.ArgumentParser(prog="torch-on-k8s-amd-client")
+    ap.add_argument("--workdir", default="/tmp/torch-on-k8s-amd")
+    sub = ap.add_subparsers(dest="cmd", required=True)
+    p_apply = sub.add_parser("apply")
+    p_apply.add_argument("file")
+    for c in ("get", "delete", "wait"):
+        pc = sub.add_parser(c)
+        pc.add_argument("name")
+    sub.add_parser("list")
+    args = ap.parse_args()
+
+    cli = TorchJobClient(args.workdir)
+    if args.cmd == "apply":
+        with open(args.file) as f:
+            name = cli.apply(f.read())
+        print(f"torchjob/{name} applied")
+    elif args.cmd == "get":
+        st = cli.get(args.name)
+        print(json.dumps(st, indent=2) if st else f"not found: {args.name}")
+    elif args.cmd == "delete":
+        ok = cli.delete(args.name)
+        print(f"torchjob/{args.name} {'deleted' if ok else 'not found'}")
+    elif args.cmd == "wait":
+        st = cli.wait(args.name)
+        print(json.dumps(st, indent=2))
+    elif args.cmd == "list":
+        for n in cli.list():
+            st = cli.get(n) or {}
+            print(f"{n:30s} {st.get('phase')}")
+
+
+if __name__ == "__main__":
+    main()

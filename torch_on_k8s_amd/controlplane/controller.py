@@ -151,9 +151,12 @@ class JobController:
         # between controller and the rank-0 checkpoint agent)
         self._sync_state_files(job)
 
-        # release GPU slots of finished tasks
+        # release GPU slots of SUCCEEDED tasks. Failed tasks keep theirs
+        # until failover decides: a retryable restart reuses the same
+        # slots (in-place locality); permanent failure releases them in
+        # _fail_job/_cleanup.
         for h in hs.values():
-            if h.finished and h.gpu_slots:
+            if h.phase == TaskPhase.SUCCEEDED and h.gpu_slots:
                 self.node.release(h.gpu_slots)
                 h.gpu_slots = ()
 
@@ -258,17 +261,27 @@ class JobController:
             self._restart_task(job, h, hs)
 
     def _restart_task(self, job: TorchJob, h: TaskHandle, hs: dict):
-        """Recreate failover (failover.go:117-127): drop the handle; the
-        next reconcile pass recreates the missing index."""
+        """Restart failover. The reference distinguishes Recreate
+        (delete + reschedule) from Kruise in-place container restart
+        (failover.go:43-48,117-264); on one node both are a process
+        restart, and we keep the IN-PLACE property that matters — the
+        task retains its GPU slots (cache/NUMA locality, no reshuffle of
+        HIP_VISIBLE_DEVICES across the gang)."""
         self.runtime.kill(h)
-        if h.gpu_slots:
-            self.node.release(h.gpu_slots)
+        keep_slots = h.gpu_slots            # slot-affinity restart
         rc = h.restart_count + 1
         hs.pop(h.key, None)
         self.event(job.name, "Normal", "TaskRestarting",
-                   f"{h.task_type.value}-{h.index} restart #{rc}")
-        self._start_task(job, h.task_type, h.index, hs)
-        nh = hs.get(h.key)
+                   f"{h.task_type.value}-{h.index} restart #{rc} "
+                   f"on GPUs {keep_slots}")
+        if keep_slots:
+            extra_env = {"MASTER_PORT": str(self._master_port(job))}
+            nh = self.runtime.start_task(job, h.task_type, h.index,
+                                         keep_slots, extra_env)
+            hs[nh.key] = nh
+        else:
+            self._start_task(job, h.task_type, h.index, hs)
+            nh = hs.get(h.key)
         if nh:
             nh.restart_count = rc
 

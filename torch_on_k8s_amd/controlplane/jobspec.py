@@ -92,8 +92,12 @@ def job_from_yaml(text: str) -> TorchJob:
     return job_from_dict(yaml.safe_load(text))
 
 
-def job_status_dict(job: TorchJob) -> dict:
+def job_status_dict(job: TorchJob, events=None) -> dict:
     return {
+        "events": [
+            {"type": e.type, "reason": e.reason, "message": e.message,
+             "ts": e.ts}
+            for e in (events or [])][-20:],
         "name": job.name,
         "phase": job.status.phase.value if job.status.phase else None,
         "generation": job.generation,

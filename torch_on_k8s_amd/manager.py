@@ -73,7 +73,7 @@ class Manager:
         self.autoscaler = TorchElasticAutoscaler(
             read_trainer_metrics(lambda job: os.path.join(
                 workdir, "jobs", job.name, "metrics.json")))
-        self._spooled: dict[str, str] = {}  # filename -> job name
+        self._spooled: dict = {}  # filename -> (job name, mtime)
 
     # ------------------------------------------------------------------
     def sync_spool(self):
@@ -83,33 +83,59 @@ class Manager:
                      if f.endswith((".yaml", ".yml", ".json"))}
         except OSError:
             return
-        for f in sorted(files - set(self._spooled)):
+        for f in sorted(files):
             path = os.path.join(self.spool, f)
+            try:
+                mtime = os.path.getmtime(path)
+            except OSError:
+                continue
+            prev = self._spooled.get(f)
+            if prev is not None and prev[1] == mtime:
+                continue
             try:
                 with open(path) as fh:
                     job = job_from_yaml(fh.read())
             except Exception as e:  # malformed spec: surface, skip
                 self.controller.event("-", "Warning", "BadJobSpec",
                                       f"{f}: {e}")
-                self._spooled[f] = ""
+                self._spooled[f] = ("", mtime)
                 continue
-            if job.name in self.controller.jobs:
-                self._spooled[f] = job.name
-                continue
-            self.metrics.job_created_at(job.name)
-            self.controller.create_job(job)
-            self._spooled[f] = job.name
+            existing = self.controller.jobs.get(job.name)
+            if existing is None:
+                self.metrics.job_created_at(job.name)
+                self.controller.create_job(job)
+            elif prev is not None:
+                # spec UPDATE (OnOwnerUpdate analog): replica changes go
+                # through the elastic generation bump so running gangs
+                # checkpoint + restart at the new world size
+                self._apply_update(existing, job)
+            self._spooled[f] = (job.name, mtime)
         for f in set(self._spooled) - files:   # spool file removed
-            name = self._spooled.pop(f)
+            name, _ = self._spooled.pop(f)
             if name:
                 self.controller.delete_job(name)
+
+    def _apply_update(self, existing, desired):
+        from torch_on_k8s_amd.controlplane.api import TaskType
+        changed = False
+        for t, spec in desired.tasks.items():
+            cur = existing.tasks.get(t)
+            if cur is not None and cur.replicas != spec.replicas:
+                cur.replicas = spec.replicas
+                changed = True
+        existing.annotations.update(desired.annotations)
+        if changed:
+            existing.generation += 1
+            self.controller.event(existing.name, "Normal", "SpecUpdated",
+                                  f"generation {existing.generation}")
 
     def publish_status(self):
         for name, job in list(self.controller.jobs.items()):
             path = os.path.join(self.status_dir, f"{name}.json")
             tmp = path + ".tmp"
+            evs = [e for e in self.controller.events if e.job == name]
             with open(tmp, "w") as f:
-                json.dump(job_status_dict(job), f, indent=2)
+                json.dump(job_status_dict(job, evs), f, indent=2)
             os.replace(tmp, path)
         if self.coordinator is not None:
             for tenant, q in self.coordinator.queues.items():
