@@ -375,3 +375,16 @@ def test_two_jobs_contending_for_node():
     for _ in range(4):
         ctl.reconcile_all()
     assert ctl.handles[waiting.name]  # tasks now created
+
+
+def test_per_job_podgroup_min_available_scales_gpus():
+    """Reference bug NOT replicated (volcano.go:223-227 TODO): a per-job
+    PodGroup with MinAvailable < replicas must scale MinResources too."""
+    from torch_on_k8s_amd.controlplane.gang import GangScheduler
+    node = NodeState(num_gpus=8)
+    gang = GangScheduler(node, dag_scheduling=False)  # per-job group
+    job = set_defaults(mk_job("pgjob", workers=6, master=False,
+                              scheduling=SchedulingPolicy(min_available=3)))
+    pg = gang.create_pod_group(job)
+    assert pg.min_member == 3
+    assert pg.min_gpus == 3  # scaled with the override, not 6

@@ -152,3 +152,33 @@ def test_allreduce_bucket_math():
     # grads unchanged by optimizer (scale folded into kernel)
     for b, g in zip(fb.buckets, g_before):
         assert torch.equal(b.flat_grad, g)
+
+
+INT_WORKER = r"""
+import os, sys
+sys.path.insert(0, os.environ["TOK_ROOT"])
+import torch
+import torch.distributed as dist
+from torch_on_k8s_amd.parallel.env import init_distributed, destroy
+ctx = init_distributed(backend="gloo")
+# bitwise-exact integer all-reduce (SURVEY §4: collective correctness)
+t = torch.arange(1000, dtype=torch.int64) * (ctx.rank + 1)
+dist.all_reduce(t)
+expect = torch.arange(1000, dtype=torch.int64) * 3  # ranks 1x + 2x
+assert torch.equal(t, expect), "int allreduce mismatch"
+destroy()
+"""
+
+
+def test_int_allreduce_bitwise():
+    procs = []
+    env0 = dict(os.environ, TOK_ROOT=ROOT, MASTER_ADDR="127.0.0.1",
+                MASTER_PORT="29717", WORLD_SIZE="2")
+    for r in range(2):
+        env = dict(env0, RANK=str(r), LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen(
+            [sys.executable, "-c", INT_WORKER], env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
+    for p in procs:
+        out, err = p.communicate(timeout=120)
+        assert p.returncode == 0, f"worker failed:\n{out}\n{err}"
