@@ -91,6 +91,8 @@ class Trainer:
         self._graph_state = None
         if cfg.hip_graph:
             assert cfg.grad_clip == 0, "hip_graph requires grad_clip=0"
+            assert cfg.grad_accum_steps <= 1, \
+                "hip_graph does not support gradient accumulation yet"
             assert self.device.type == "cuda", "hip_graph needs a GPU"
 
     @property
