@@ -61,3 +61,51 @@ def test_chaos_failure_storm():
                                     JobConditionType.FAILED), \
             (job.name, job.status.phase)
     assert len(node.free_slots) == 8  # no leaked GPU slots at the end
+
+
+def test_chaos_with_elastic_scaling():
+    """Failure storms interleaved with random scale() calls: the elastic
+    checkpoint transaction + failover must never wedge or leak slots."""
+    from torch_on_k8s_amd.controlplane.api import ElasticPolicy
+    from torch_on_k8s_amd.controlplane.elastic import ElasticScaler
+
+    rng = random.Random(11)
+    node = NodeState(num_gpus=8)
+    rt = FakeRuntime()
+    ctl = JobController(node, rt, ControllerConfig(),
+                        elastic=ElasticScaler())
+    job = TorchJob(
+        name="chaos-elastic",
+        tasks={TaskType.MASTER: TaskSpec(replicas=1, gpus_per_task=0),
+               TaskType.WORKER: TaskSpec(replicas=2, gpus_per_task=1)},
+        elastic=ElasticPolicy(min_replicas=1, max_replicas=6),
+        run_policy=RunPolicy(backoff_limit=50))
+    ctl.create_job(job)
+
+    for it in range(600):
+        r = rng.random()
+        live = [h for h in rt.tasks.values() if not h.finished]
+        if r < 0.10:
+            ElasticScaler.scale(job, rng.randint(1, 6))
+        elif r < 0.20 and live:
+            h = rng.choice(live)
+            rt.set_phase(h.key, TaskPhase.FAILED,
+                         exit_code=rng.choice([137, 143]))
+        elif r < 0.30:
+            # the data plane completes any pending checkpoint request
+            ElasticScaler.complete_checkpoint(job)
+        ctl.reconcile(job)
+        held = sum(len(h.gpu_slots) for hs in ctl.handles.values()
+                   for h in hs.values())
+        assert held == 8 - len(node.free_slots), f"slot leak at iter {it}"
+        assert job.status.phase != JobConditionType.FAILED, \
+            f"wedged failed at iter {it}"
+
+    # settle: complete checkpoints + reconcile until the gang matches
+    for _ in range(20):
+        ElasticScaler.complete_checkpoint(job)
+        ctl.reconcile(job)
+    workers = [h for h in ctl.handles["chaos-elastic"].values()
+               if h.task_type == TaskType.WORKER]
+    assert len(workers) == job.tasks[TaskType.WORKER].replicas
+    assert all(h.generation == job.generation for h in workers)
