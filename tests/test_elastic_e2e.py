@@ -56,7 +56,16 @@ def test_elastic_scale_out_e2e(tmp_path):
         elastic=ElasticPolicy(min_replicas=1, max_replicas=2),
     )
     ctl.create_job(job)
+    try:
+        _run_scenario(ctl, rt, job, tmp_path, steps)
+    finally:
+        # never leave orphaned gangs holding rendezvous ports for later
+        # tests (a failed run's processes previously caused cross-test
+        # port collisions)
+        ctl.delete_job(job.name)
 
+
+def _run_scenario(ctl, rt, job, tmp_path, steps):
     # wait until training is underway
     mpath = tmp_path / "work" / "elastic-e2e" / "metrics.json"
     t0 = time.time()

@@ -68,7 +68,9 @@ class JobController:
         self.events: list[Event] = []
         self._event_last: dict = {}
         self._ports: dict[str, int] = {}
-        self._next_port = self.cfg.master_port_range[0]
+        import random as _random
+        lo, hi = self.cfg.master_port_range
+        self._next_port = _random.randrange(lo, hi)
         self._first_task_ts: dict[str, float] = {}
 
     # ------------------------------------------------------------------
@@ -112,9 +114,11 @@ class JobController:
             self.metrics.deleted()
 
     def _master_port(self, job: TorchJob) -> int:
-        """Per-job master port from the configured range (the reference
-        picks random host ports from --hostnetwork-port-range,
-        pod.go:531-544; collisions here are avoided by monotonic sweep)."""
+        """Per-job master port: RANDOM start within the configured range
+        (parity with the reference's random host-port selection,
+        pod.go:531-544) then a monotonic sweep, so concurrent jobs -- and
+        stale sockets of earlier controllers/orphaned gangs -- don't
+        collide. The port is sticky per job (stable across restarts)."""
         if job.name not in self._ports:
             lo, hi = self.cfg.master_port_range
             self._ports[job.name] = lo + (self._next_port - lo) % (hi - lo)
@@ -268,6 +272,11 @@ class JobController:
         task retains its GPU slots (cache/NUMA locality, no reshuffle of
         HIP_VISIBLE_DEVICES across the gang)."""
         self.runtime.kill(h)
+        # wait for the old process to release its sockets before the
+        # replacement binds the same master port (runtimes without a
+        # wait(), e.g. FakeRuntime, finish synchronously)
+        if hasattr(self.runtime, "wait"):
+            self.runtime.wait(h, timeout=60)
         keep_slots = h.gpu_slots            # slot-affinity restart
         rc = h.restart_count + 1
         hs.pop(h.key, None)

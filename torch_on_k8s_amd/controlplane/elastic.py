@@ -116,17 +116,31 @@ class ElasticScaler:
             (1 if TaskType.MASTER in job.tasks else 0)
         job.annotations[ANN_WORLD_SIZE] = str(new_world)
 
-        # kill victims (scale-in) and stale-generation tasks; the normal
-        # reconcile pass recreates missing indices with the new env
-        # (restartStalePod analog, elastic_scale.go:303-397)
+        # Kill victims (scale-in) and stale-generation tasks, then WAIT
+        # for them to exit before the normal pass recreates anything:
+        # the new gang reuses the job's master port, and racing a dying
+        # master's rendezvous store hangs or mis-wires the new ranks
+        # (restartStalePod analog, elastic_scale.go:303-397; the
+        # reference gets the equivalent ordering from Kruise CRR
+        # completion before patching the generation label).
+        pending = False
         for h in list(hs.values()):
             victim = (h.task_type == TaskType.WORKER and
                       h.index >= desired.replicas)
-            if victim or h.generation != job.generation:
+            if not (victim or h.generation != job.generation):
+                continue
+            if not h.finished:
                 ctl.runtime.kill(h)
+                ctl.runtime.poll(h)
+            if h.finished:
                 if h.gpu_slots:
                     ctl.node.release(h.gpu_slots)
+                    h.gpu_slots = ()
                 hs.pop(h.key, None)
+            else:
+                pending = True
+        if pending:
+            return True  # victims still terminating; hold recreation
         ctl.event(job.name, "Normal", "ScaleExecuted",
                   f"world_size={new_world} gen={job.generation}")
         return False  # let the normal pass recreate tasks now
