@@ -388,3 +388,27 @@ def test_per_job_podgroup_min_available_scales_gpus():
     pg = gang.create_pod_group(job)
     assert pg.min_member == 3
     assert pg.min_gpus == 3  # scaled with the override, not 6
+
+
+def test_wrr_statistical_fairness():
+    """WRR picks must be proportional to queue weights over many rounds
+    (weight = pending tasks, policy.go:224-230 semantics)."""
+    from torch_on_k8s_amd.controlplane.coordinator import (
+        Queue, QueueUnit, WeightedRoundRobinSelector)
+    qa, qb, qc = Queue("a"), Queue("b"), Queue("c")
+    # weights: a = 6, b = 3, c = 1 (pending tasks incl. master)
+    qa.add(QueueUnit(set_defaults(mk_job(
+        "wa", workers=5, scheduling=SchedulingPolicy(queue="a")))))
+    qb.add(QueueUnit(set_defaults(mk_job(
+        "wb", workers=2, scheduling=SchedulingPolicy(queue="b")))))
+    qc.add(QueueUnit(set_defaults(mk_job(
+        "wc", workers=0, scheduling=SchedulingPolicy(queue="c")))))
+    sel = WeightedRoundRobinSelector()
+    picks = {"a": 0, "b": 0, "c": 0}
+    rounds = 1000
+    for _ in range(rounds):
+        picks[sel.next([qa, qb, qc]).tenant] += 1
+    # expected proportions 6:3:1 within 5% absolute
+    assert abs(picks["a"] / rounds - 0.6) < 0.05, picks
+    assert abs(picks["b"] / rounds - 0.3) < 0.05, picks
+    assert abs(picks["c"] / rounds - 0.1) < 0.05, picks
