@@ -109,13 +109,13 @@ def main() -> int:
     signal.signal(signal.SIGTERM, on_term)
 
     def checkpoint_and_ack(version=None):
-        # No barrier here: only rank 0 runs this (from its agent poll or
-        # its own SIGTERM) and DP replicas are identical after a step, so
-        # rank 0's tensors alone are a consistent checkpoint. A barrier
-        # would interleave with other ranks' in-flight collectives.
-        if ckpt_dir and ctx.is_main:
+        # Called by EVERY rank at the same step (the train loop
+        # broadcasts the decision from rank 0), so the sharded save's
+        # internal barriers are collective-safe. World 1 degenerates to
+        # a plain rank-0 save.
+        if ckpt_dir:
             trainer.save_checkpoint(ckpt_dir)
-            if agent_file and version is not None:
+            if ctx.is_main and agent_file and version is not None:
                 _atomic_write(agent_file, {
                     "ckpt-completed-version": {"version": version,
                                                "status": "Succeeded"},
@@ -135,38 +135,55 @@ def main() -> int:
         return 143
 
 
+def _read_ckpt_request(job_file):
+    req = (_read_json(job_file) or {}).get("annotations", {}).get(
+        "ckpt-requested-version")
+    if isinstance(req, str):
+        try:
+            req = json.loads(req)
+        except ValueError:
+            req = None
+    return req
+
+
 def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
                 stop, checkpoint_and_ack, destroy, barrier):
+    import torch
+    import torch.distributed as dist
     last_completed = None
+    # coordination word broadcast from rank 0 each step so every rank
+    # takes checkpoint/stop decisions at the SAME step (sharded saves
+    # are collective): [requested ckpt version or 0, stop flag]
+    coord = torch.zeros(2, dtype=torch.long, device=ctx.device)
     while trainer.step_count < steps_total:
         loss = trainer.train_step()
         if ctx.is_main:
             print(f"[train] step={trainer.step_count} loss={loss:.4f}",
                   flush=True)
-        # checkpoint-agent poll (rank 0): serve controller requests
-        if ctx.is_main and job_file:
-            jf = _read_json(job_file)
-            req = (jf or {}).get("annotations", {}).get(
-                "ckpt-requested-version")
-            if isinstance(req, str):
-                try:
-                    req = json.loads(req)
-                except ValueError:
-                    req = None
-            if req and req.get("version") != last_completed:
-                checkpoint_and_ack(req["version"])
-                last_completed = req["version"]
-        if stop["sig"] is not None:
+        if ctx.is_main:
+            req = _read_ckpt_request(job_file) if job_file else None
+            coord[0] = req["version"] if req else 0
+            coord[1] = 1 if stop["sig"] is not None else 0
+        if ctx.is_distributed and dist.is_initialized():
+            dist.broadcast(coord, src=0)
+        reqv = int(coord[0])
+        stop_now = bool(int(coord[1])) or \
+            (not ctx.is_distributed and stop["sig"] is not None)
+        if reqv and reqv != last_completed:
+            checkpoint_and_ack(reqv)
+            last_completed = reqv
+        if stop_now:
             checkpoint_and_ack()
             destroy()
             return 143  # SIGTERM: clean checkpointed exit, retryable
 
     # final checkpoint becomes the model artifact (output/ packaged by the
-    # control plane into a ModelVersion on job success)
+    # control plane into a ModelVersion on job success). Explicitly
+    # UNsharded: only rank 0 calls it, so it must not contain barriers.
     if state_dir and ctx.is_main:
         out = os.path.join(state_dir, "output")
         os.makedirs(out, exist_ok=True)
-        trainer.save_checkpoint(os.path.join(out, "final"))
+        trainer.save_checkpoint(os.path.join(out, "final"), sharded=False)
     barrier(ctx)
     destroy()
     return 0
