@@ -19,8 +19,12 @@ from torch_on_k8s_amd.controlplane.runtime import FakeRuntime
 EXIT_CODES = [0, 1, 137, 138, 139, 143, 2, 130]
 
 
-def test_chaos_failure_storm():
-    rng = random.Random(7)
+import pytest
+
+
+@pytest.mark.parametrize("seed", [7, 23, 101, 2024])
+def test_chaos_failure_storm(seed):
+    rng = random.Random(seed)
     node = NodeState(num_gpus=8)
     rt = FakeRuntime()
     ctl = JobController(node, rt, ControllerConfig())

@@ -93,6 +93,8 @@ class Trainer:
             assert cfg.grad_clip == 0, "hip_graph requires grad_clip=0"
             assert cfg.grad_accum_steps <= 1, \
                 "hip_graph does not support gradient accumulation yet"
+            assert not cfg.lr_warmup_steps and not cfg.lr_decay_steps, \
+                "hip_graph freezes the captured LR; schedules need eager"
             assert self.device.type == "cuda", "hip_graph needs a GPU"
 
     @property
