@@ -287,7 +287,7 @@ def test_autoscaler_doubles_then_reverts():
     assert act is not None and job.tasks[TaskType.WORKER].replicas == 4
     gen_after_scale = job.generation
 
-    # 4 replicas: latency 0.4 (per-replica improves: 0.4*4 < 1.0*2) -> 8
+    # 4 replicas: latency 0.4 (per-replica improves: 0.4/4 < 1.0/2) -> 8
     feed["obs"] = [Observation(3, 0.4, 250, 5.0),
                    Observation(4, 0.4, 250, 5.0)]
     auto.observe(job)
@@ -295,9 +295,9 @@ def test_autoscaler_doubles_then_reverts():
     act = auto.decide(job)
     assert job.tasks[TaskType.WORKER].replicas == 8
 
-    # 8 replicas: latency 0.39 (0.39*8 > 0.4*4) -> revert to 4, stop
-    feed["obs"] = [Observation(5, 0.39, 260, 5.0),
-                   Observation(6, 0.39, 260, 5.0)]
+    # 8 replicas: latency 0.85 (0.85/8 > 0.4/4) -> revert to 4, stop
+    feed["obs"] = [Observation(5, 0.85, 260, 5.0),
+                   Observation(6, 0.85, 260, 5.0)]
     auto.observe(job)
     auto.observe(job)
     act = auto.decide(job)

@@ -240,7 +240,11 @@ class TorchElasticAutoscaler:
             return True  # first scale decision: allow growth
         cur_lat = sum(o.latency for o in cur) / len(cur)
         prev_lat = sum(o.latency for o in prev) / len(prev)
-        return cur_lat * st.replicas < prev_lat * st.last_replicas
+        # lastLat/lastN > curLat/curN  <=>  cur_lat*lastN < prev_lat*curN
+        # (latency per replica must drop; with DP more replicas process
+        # proportionally more data per step, so flat step latency at 2x
+        # replicas IS an improvement)
+        return cur_lat * st.last_replicas < prev_lat * st.replicas
 
 
 def read_trainer_metrics(metrics_path_fn):

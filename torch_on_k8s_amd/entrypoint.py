@@ -13,6 +13,12 @@ Trainer configuration:
   TOK_STATE_DIR       job state dir: job.json (controller-written),
                       agent.json (trainer-written), ckpt/, metrics.json,
                       output/ (model artifact source)
+  TOK_BENCH_STEPS /   benchmark mode: after TOK_BENCH_WARMUP untimed
+  TOK_BENCH_WARMUP    steps, time exactly TOK_BENCH_STEPS steps bracketed
+                      by barrier+synchronize on both sides; rank 0 writes
+                      <state_dir>/bench.json (read by `bench.py
+                      --via-manager` for the gang-scheduled headline
+                      number). TOK_TRAIN_STEPS must equal warmup+steps.
 Exit codes follow the failover contract (controlplane/failover.py):
   0 success; 143 on SIGTERM after a clean checkpoint (retryable).
 
@@ -151,11 +157,48 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
     import torch
     import torch.distributed as dist
     last_completed = None
+    # benchmark instrumentation (TOK_BENCH_*): timed region bracketed by
+    # barrier + device sync on both sides, MAX-elapsed over ranks
+    bench_steps = int(os.environ.get("TOK_BENCH_STEPS", "0"))
+    bench_warmup = int(os.environ.get("TOK_BENCH_WARMUP", "0"))
+    bench_t0 = None
+
+    def _barrier_sync():
+        barrier(ctx)
+        if ctx.device.type == "cuda":
+            torch.cuda.synchronize()
     # coordination word broadcast from rank 0 each step so every rank
     # takes checkpoint/stop decisions at the SAME step (sharded saves
     # are collective): [requested ckpt version or 0, stop flag]
     coord = torch.zeros(2, dtype=torch.long, device=ctx.device)
     while trainer.step_count < steps_total:
+        if bench_steps:
+            # benchmark mode: every timed step is EXACTLY a bare-bench
+            # step (sync=False, no per-step host sync / coordination
+            # collectives inside the timed region — those are
+            # control-plane conveniences, not step work)
+            if trainer.step_count == bench_warmup:
+                _barrier_sync()
+                bench_t0 = time.perf_counter()
+            loss = trainer.train_step(sync=False)
+            if bench_t0 is not None and \
+                    trainer.step_count == bench_warmup + bench_steps:
+                _barrier_sync()
+                elapsed = time.perf_counter() - bench_t0
+                if ctx.is_distributed and dist.is_initialized():
+                    t = torch.tensor([elapsed], dtype=torch.float64,
+                                     device=ctx.device)
+                    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+                    elapsed = t.item()
+                if ctx.is_main and state_dir:
+                    _atomic_write(os.path.join(state_dir, "bench.json"), {
+                        "elapsed_s": elapsed,
+                        "steps": bench_steps,
+                        "warmup": bench_warmup,
+                        "world_size": ctx.world_size,
+                        "loss": float(loss.detach().float().item()),
+                    })
+            continue
         loss = trainer.train_step()
         if ctx.is_main:
             print(f"[train] step={trainer.step_count} loss={loss:.4f}",
@@ -180,6 +223,9 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
     # final checkpoint becomes the model artifact (output/ packaged by the
     # control plane into a ModelVersion on job success). Explicitly
     # UNsharded: only rank 0 calls it, so it must not contain barriers.
+    # Benchmark jobs skip it (they measure steps, not packaging).
+    if bench_steps:
+        state_dir = None
     if state_dir and ctx.is_main:
         out = os.path.join(state_dir, "output")
         os.makedirs(out, exist_ok=True)

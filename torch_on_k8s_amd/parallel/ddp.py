@@ -150,13 +150,17 @@ class FlatBucketModel:
         if self.world_size <= 1:
             return
         for b in self.buckets:
-            if self.overlap:
-                if b.work is not None:
-                    b.work.wait()
-                elif b.pending != len(b.segs):
-                    # partial bucket (e.g. frozen params) - reduce now
-                    dist.all_reduce(b.flat_grad, group=self.group)
+            if b.work is not None:
+                b.work.wait()
             else:
+                # The hook never launched this bucket's all-reduce this
+                # step: overlap off, frozen params leaving pending > 0,
+                # or grad accumulation where the final micro-step touched
+                # none of its params (grads from earlier micro-steps
+                # still need syncing). Reducing every un-reduced bucket
+                # keeps the collective count identical on all ranks
+                # (conditional compute can make 'did params get grads'
+                # rank-dependent).
                 dist.all_reduce(b.flat_grad, group=self.group)
             b.work = None
 
