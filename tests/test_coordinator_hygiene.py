@@ -1,0 +1,194 @@
+"""Round-2 coordinator/controller hardening tests (r1 VERDICT weak
+#3/#4/#5): quota lifecycle (forget on admission), resource-map quotas,
+WRR stability under queue churn, and a daemon soak that holds memory
+bounds over hundreds of short jobs."""
+from __future__ import annotations
+
+import time
+
+from torch_on_k8s_amd.controlplane.api import (SchedulingPolicy, TaskPhase,
+                                               TaskSpec, TaskType, TorchJob,
+                                               set_defaults)
+from torch_on_k8s_amd.controlplane.controller import (ControllerConfig,
+                                                      JobController)
+from torch_on_k8s_amd.controlplane.coordinator import (
+    Coordinator, Queue, QueueUnit, QuotaPlugin, WeightedRoundRobinSelector)
+from torch_on_k8s_amd.controlplane.node import NodeState
+from torch_on_k8s_amd.controlplane.runtime import FakeRuntime
+
+
+def mk_job(name, workers=1, master=True, queue="t", **kw):
+    tasks = {}
+    if master:
+        tasks[TaskType.MASTER] = TaskSpec(replicas=1)
+    if workers:
+        tasks[TaskType.WORKER] = TaskSpec(replicas=workers)
+    return TorchJob(name=name, tasks=tasks,
+                    scheduling=SchedulingPolicy(queue=queue), **kw)
+
+
+def mk_stack(num_gpus=8, quotas=None):
+    node = NodeState(num_gpus=num_gpus)
+    rt = FakeRuntime()
+    ctl = JobController(node, rt, ControllerConfig())
+    coord = Coordinator(dequeue_fn=ctl.reconcile,
+                        tenant_usage_fn=ctl.tenant_resource_usage,
+                        quotas=quotas, default_quota=num_gpus)
+    ctl.coordinator = coord
+    return ctl, coord, node, rt
+
+
+# ---------------------------------------------------------------------------
+# quota lifecycle: forget() on admission (no 60s double-count)
+# ---------------------------------------------------------------------------
+def test_sequential_same_tenant_admission_inside_ttl():
+    """Job A (4 GPUs) admitted, then job B (4 GPUs) of the SAME tenant
+    must be admittable immediately: A's assumed quota is forgotten once
+    its tasks exist (r1: it double-counted for 60s and stalled B)."""
+    ctl, coord, node, rt = mk_stack(num_gpus=8, quotas={"t": 8})
+    a = ctl.create_job(mk_job("a", workers=3))   # 4 tasks x 1 GPU
+    assert coord.schedule_once() is not None     # dequeues A
+    ctl.reconcile(a)  # DAG: workers start once master is Running
+    # A fully admitted -> assumed quota must be gone
+    assert a.uid not in coord.quota._assumed
+    b = ctl.create_job(mk_job("b", workers=3))
+    out = coord.schedule_once()
+    assert out is not None and out.name == "b"
+    ctl.reconcile(b)
+    # both jobs' tasks exist
+    assert len(ctl.handles["a"]) == 4 and len(ctl.handles["b"]) == 4
+
+
+def test_quota_forget_on_job_delete():
+    ctl, coord, node, rt = mk_stack(num_gpus=2, quotas={"t": 2})
+    a = ctl.create_job(mk_job("a", workers=1))
+    coord.schedule_once()
+    ctl.reconcile(a)  # second pass: workers created after master Running
+    assert a.uid not in coord.quota._assumed
+    ctl.delete_job("a")
+    b = ctl.create_job(mk_job("b", workers=1))
+    assert coord.schedule_once() is not None
+    ctl.reconcile(b)
+    assert len(ctl.handles["b"]) == 2
+
+
+# ---------------------------------------------------------------------------
+# resource-map quotas (reference covers full ResourceQuota lists)
+# ---------------------------------------------------------------------------
+def test_resource_map_quota_filters_cpu_and_mem():
+    quota = QuotaPlugin(quotas={"t": {"gpu": 8, "cpu": 16, "memory_mb": 1000}})
+    job = set_defaults(TorchJob(
+        name="j", tasks={TaskType.WORKER: TaskSpec(
+            replicas=2, gpus_per_task=1, cpus_per_task=4,
+            mem_mb_per_task=300)},
+        scheduling=SchedulingPolicy(queue="t")))
+    qu = QueueUnit(job)
+    # fits: 2 gpu, 8 cpu, 600 mb
+    assert quota.filter(qu, {"gpu": 0, "cpu": 0, "memory_mb": 0})
+    # cpu exhausted
+    assert not quota.filter(qu, {"gpu": 0, "cpu": 9, "memory_mb": 0})
+    # memory exhausted
+    assert not quota.filter(qu, {"gpu": 0, "cpu": 0, "memory_mb": 500})
+    # legacy int in_use means gpu only
+    assert quota.filter(qu, 6)
+    assert not quota.filter(qu, 7)
+
+
+def test_int_quota_means_gpu_only():
+    quota = QuotaPlugin(quotas={"t": 4})
+    job = set_defaults(TorchJob(
+        name="j", tasks={TaskType.WORKER: TaskSpec(
+            replicas=4, gpus_per_task=1, cpus_per_task=100)},
+        scheduling=SchedulingPolicy(queue="t")))
+    assert quota.filter(QueueUnit(job), 0)      # cpu not quota-tracked
+    assert not quota.filter(QueueUnit(job), 1)  # gpu would exceed
+
+
+def test_assumed_quota_ttl_expiry_still_works():
+    quota = QuotaPlugin(quotas={"t": 4})
+    quota.ASSUME_TTL = 0.05
+    job = set_defaults(mk_job("a", workers=3))
+    quota.pre_dequeue(QueueUnit(job))
+    b = QueueUnit(set_defaults(mk_job("b", workers=3)))
+    assert not quota.filter(b, 0)
+    time.sleep(0.08)
+    assert quota.filter(b, 0)
+
+
+# ---------------------------------------------------------------------------
+# WRR under queue churn (keyed by tenant, not index)
+# ---------------------------------------------------------------------------
+def _mk_queue(tenant, jobs, workers):
+    q = Queue(tenant)
+    for i in range(jobs):
+        q.add(QueueUnit(set_defaults(mk_job(
+            f"{tenant}{i}", workers=workers, queue=tenant))))
+    return q
+
+
+def test_wrr_cursor_survives_queue_removal():
+    """Removing a queue between calls must not shift the cursor onto a
+    different tenant's slot (r1: index-based state mis-pointed)."""
+    qa = _mk_queue("a", 1, 3)   # weight 4
+    qb = _mk_queue("b", 1, 3)   # weight 4
+    qc = _mk_queue("c", 1, 3)   # weight 4
+    sel = WeightedRoundRobinSelector()
+    first = sel.next([qa, qb, qc]).tenant
+    assert first == "a"
+    # tenant a disappears; the cursor was at a -> next must be b (the
+    # next name in order), not a skipped/doubled pick
+    second = sel.next([qb, qc]).tenant
+    assert second == "b"
+    third = sel.next([qb, qc]).tenant
+    assert third == "c"
+    # a returns; cursor at c -> wraps to a
+    fourth = sel.next([qa, qb, qc]).tenant
+    assert fourth == "a"
+
+
+def test_wrr_fairness_with_churn():
+    """Fairness holds (roughly proportional) while queues come and go."""
+    sel = WeightedRoundRobinSelector()
+    picks = {"a": 0, "b": 0}
+    for round_i in range(600):
+        qa = _mk_queue("a", 1, 5)   # weight 6
+        qb = _mk_queue("b", 1, 2)   # weight 3
+        queues = [qa, qb]
+        if round_i % 3 == 0:        # transient tenant churns in and out
+            queues.append(_mk_queue(f"x{round_i}", 1, 0))
+        t = sel.next(queues).tenant
+        if t in picks:
+            picks[t] += 1
+    total = picks["a"] + picks["b"]
+    # a:b should be ~2:1
+    assert 0.55 < picks["a"] / total < 0.78, picks
+
+
+# ---------------------------------------------------------------------------
+# daemon soak: hundreds of short jobs leave no residue
+# ---------------------------------------------------------------------------
+def test_soak_no_unbounded_growth():
+    ctl, coord, node, rt = mk_stack(num_gpus=8)
+    for i in range(300):
+        name = f"s{i}"
+        job = ctl.create_job(mk_job(name, workers=1, queue=f"ten{i % 7}"))
+        coord.schedule_once()
+        ctl.reconcile(job)
+        # finish both tasks
+        for key in list(ctl.handles.get(name, {})):
+            rt.set_phase(key, TaskPhase.SUCCEEDED, exit_code=0)
+        ctl.reconcile(job)
+        assert job.status.phase is not None
+        ctl.delete_job(name)
+    # all per-job state released
+    assert not ctl.jobs and not ctl.handles
+    assert not ctl._ports and not ctl._first_task_ts
+    assert not ctl._events_by_job
+    assert not ctl._event_last
+    assert len(ctl.events) <= ctl.events.maxlen
+    # coordinator: no tenant-name or assumed-quota leak
+    assert not coord.queues
+    assert not coord.quota._assumed
+    assert not coord._index
+    # every GPU slot returned
+    assert len(node.free_slots) == 8

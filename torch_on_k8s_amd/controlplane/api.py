@@ -102,6 +102,10 @@ class TaskSpec:
     replicas: int = 1
     restart_policy: RestartPolicy | None = None
     gpus_per_task: int = 1
+    # full resource requests (reference quota filtering covers every
+    # resource in the namespace ResourceQuotas, plugins/quota.go:97-131)
+    cpus_per_task: float = 0.0
+    mem_mb_per_task: int = 0
     dag_conditions: list[DAGCondition] = field(default_factory=list)
     spot: SpotTaskSpec | None = None
     # entrypoint override: argv list; None -> framework training entrypoint
@@ -220,6 +224,18 @@ class TorchJob:
     def total_gpus(self, include_aimaster: bool = True) -> int:
         return sum(s.replicas * s.gpus_per_task for t, s in self.tasks.items()
                    if include_aimaster or t != TaskType.AIMASTER)
+
+    def total_resources(self) -> dict:
+        """Aggregate resource request (the quota plugin's unit; reference
+        computes this over pod template resources, resources.go:28-109)."""
+        return {
+            "gpu": sum(s.replicas * s.gpus_per_task
+                       for s in self.tasks.values()),
+            "cpu": sum(s.replicas * s.cpus_per_task
+                       for s in self.tasks.values()),
+            "memory_mb": sum(s.replicas * s.mem_mb_per_task
+                             for s in self.tasks.values()),
+        }
 
 
 # Annotation keys (reference constants.go:62-78, elastic_scale.go:49-56)

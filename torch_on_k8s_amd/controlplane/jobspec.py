@@ -34,6 +34,8 @@ def _task_spec(d: dict) -> TaskSpec:
     spec = TaskSpec(
         replicas=int(d.get("replicas", 1)),
         gpus_per_task=int(d.get("gpusPerTask", 1)),
+        cpus_per_task=float(d.get("cpusPerTask", 0)),
+        mem_mb_per_task=int(d.get("memMbPerTask", 0)),
         command=d.get("command"),
         env={str(k): str(v) for k, v in (d.get("env") or {}).items()},
     )

@@ -65,7 +65,7 @@ class Manager:
         if self.gates.enabled(feat.JOB_COORDINATOR):
             self.coordinator = Coordinator(
                 dequeue_fn=self.controller.reconcile,
-                tenant_usage_fn=self.controller.tenant_gpu_usage,
+                tenant_usage_fn=self.controller.tenant_resource_usage,
                 quotas=quotas, default_quota=num_gpus)
             self.controller.coordinator = self.coordinator
         else:
@@ -130,15 +130,24 @@ class Manager:
                                   f"generation {existing.generation}")
 
     def publish_status(self):
+        live = set()
         for name, job in list(self.controller.jobs.items()):
+            live.add(f"{name}.json")
             path = os.path.join(self.status_dir, f"{name}.json")
             tmp = path + ".tmp"
-            evs = [e for e in self.controller.events if e.job == name]
+            evs = self.controller.events_for(name)
             with open(tmp, "w") as f:
                 json.dump(job_status_dict(job, evs), f, indent=2)
             os.replace(tmp, path)
+        # drop status files of deleted jobs (daemon hygiene)
+        try:
+            for f in os.listdir(self.status_dir):
+                if f.endswith(".json") and f not in live:
+                    os.unlink(os.path.join(self.status_dir, f))
+        except OSError:
+            pass
         if self.coordinator is not None:
-            for tenant, q in self.coordinator.queues.items():
+            for tenant, q in list(self.coordinator.queues.items()):
                 self.metrics.set_queue_depth(tenant, len(q))
 
     def autoscale_pass(self):
@@ -172,14 +181,18 @@ def main():
     ap.add_argument("--metrics-addr", type=int, default=8443)
     ap.add_argument("--feature-gates", default="")
     ap.add_argument("--quota", action="append", default=[],
-                    help="tenant=gpus, repeatable")
+                    help='tenant=gpus or tenant={"gpu":8,"cpu":64,'
+                         '"memory_mb":512000}; repeatable')
     ap.add_argument("--sync-period", type=float, default=0.5)
     args = ap.parse_args()
 
     quotas = {}
     for q in args.quota:
         k, _, v = q.partition("=")
-        quotas[k] = int(v)
+        try:
+            quotas[k] = int(v)
+        except ValueError:
+            quotas[k] = json.loads(v)  # resource-map quota
     gates = feat.FeatureGates.from_flag(args.feature_gates)
     mgr = Manager(args.workdir, num_gpus=args.num_gpus,
                   quotas=quotas or None, gates=gates,
