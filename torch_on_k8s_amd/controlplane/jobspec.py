@@ -1,7 +1,8 @@
 """YAML/dict TorchJob specs — the kubectl-facing surface.
 
-The reference's CRD YAML (config/crd/bases/, samples in config/samples/)
-maps to this schema:
+TWO dialects are accepted (auto-detected per document):
+
+1. The native node dialect:
 
     apiVersion: train.distributed.io/v1alpha1   # accepted, informational
     kind: TorchJob
@@ -19,6 +20,17 @@ maps to this schema:
         master: {replicas: 1, gpusPerTask: 1, restartPolicy: OnExitCode,
                  env: {TOK_TRAIN_STEPS: "100"}}
         worker: {replicas: 7, gpusPerTask: 1}
+
+2. The reference CRD dialect (group train.distributed.io/v1alpha1,
+   apis/train/v1alpha1/torchjob_types.go:88-206): `spec.torchTaskSpecs`
+   maps TaskType -> {numTasks, restartPolicy, spotTaskSpec, template};
+   the pod template's `torch` container carries command/env/resources.
+   GPU resources may be named `amd.com/gpu` (this framework's node) or
+   `nvidia.com/gpu` (reference manifests; accepted and mapped 1:1 so a
+   job written against the reference parses unchanged). RunPolicy fields
+   are inlined on spec — including the reference's `clenPodPolicy`
+   spelling (torchjob_types.go:142) — plus `minMembers`,
+   `enableTorchElastic`/`torchElasticPolicy` and `modelVersion`.
 """
 from __future__ import annotations
 
@@ -28,6 +40,23 @@ from torch_on_k8s_amd.controlplane.api import (CleanPodPolicy, DAGCondition,
                                                SpotTaskSpec, TaskPhase,
                                                TaskSpec, TaskType, TorchJob,
                                                set_defaults)
+
+GPU_RESOURCES = ("amd.com/gpu", "nvidia.com/gpu")
+
+
+def parse_quantity(q) -> float:
+    """k8s resource quantity ('4', '500m', '8Gi', '2G') -> float in base
+    units (memory suffixes return bytes)."""
+    if isinstance(q, (int, float)):
+        return float(q)
+    s = str(q).strip()
+    suffixes = {"Ki": 1 << 10, "Mi": 1 << 20, "Gi": 1 << 30, "Ti": 1 << 40,
+                "K": 10**3, "M": 10**6, "G": 10**9, "T": 10**12,
+                "k": 10**3, "m": 1e-3}
+    for suf in sorted(suffixes, key=len, reverse=True):
+        if s.endswith(suf):
+            return float(s[:-len(suf)]) * suffixes[suf]
+    return float(s)
 
 
 def _task_spec(d: dict) -> TaskSpec:
@@ -53,9 +82,172 @@ def _task_spec(d: dict) -> TaskSpec:
     return spec
 
 
+_CRD_RESTART = {"ExitCode": RestartPolicy.ON_EXIT_CODE,
+                "OnFailure": RestartPolicy.ON_FAILURE,
+                "Always": RestartPolicy.ALWAYS,
+                "Never": RestartPolicy.NEVER}
+
+
+def _task_spec_from_crd(d: dict) -> TaskSpec:
+    """TaskSpec from the reference shape: numTasks + pod template whose
+    `torch` (or first) container carries command/env/resources
+    (torchjob_types.go:88-104, constants.go:96-103)."""
+    spec = TaskSpec(replicas=int(d.get("numTasks", 1)))
+    rp = d.get("restartPolicy")
+    if rp:
+        spec.restart_policy = _CRD_RESTART.get(rp) or RestartPolicy(rp)
+    st = d.get("spotTaskSpec")
+    if st:
+        spec.spot = SpotTaskSpec(
+            num_spot_replicas=int(st.get("numSpotTasks", 0)),
+            labels=dict(st.get("labels") or {}))
+    tmpl = (d.get("template") or {}).get("spec") or {}
+    containers = tmpl.get("containers") or []
+    cont = None
+    for c in containers:
+        if c.get("name") == "torch":  # reference default container name
+            cont = c
+            break
+    if cont is None and containers:
+        cont = containers[0]
+    if cont:
+        if cont.get("command"):
+            spec.command = list(cont["command"]) + list(cont.get("args") or [])
+        for e in cont.get("env") or []:
+            if "name" in e:
+                spec.env[str(e["name"])] = str(e.get("value", ""))
+        res = cont.get("resources") or {}
+        req = dict(res.get("requests") or {})
+        req.update(res.get("limits") or {})  # limits win (GPU convention)
+        for gpu_key in GPU_RESOURCES:
+            if gpu_key in req:
+                spec.gpus_per_task = int(parse_quantity(req[gpu_key]))
+                break
+        else:
+            spec.gpus_per_task = 0
+        if "cpu" in req:
+            spec.cpus_per_task = parse_quantity(req["cpu"])
+        if "memory" in req:
+            spec.mem_mb_per_task = int(parse_quantity(req["memory"]) /
+                                       (1 << 20))
+    return spec
+
+
+def _job_from_crd_dict(doc: dict) -> TorchJob:
+    """Reference-CRD-dialect parser (torchjob_types.go:178-206 shape)."""
+    meta = doc.get("metadata") or {}
+    spec = doc.get("spec") or {}
+    tasks = {TaskType(t.lower()): _task_spec_from_crd(s or {})
+             for t, s in (spec.get("torchTaskSpecs") or {}).items()}
+    job = TorchJob(
+        name=meta.get("name", "torchjob"),
+        namespace=meta.get("namespace", "default"),
+        annotations={str(k): str(v)
+                     for k, v in (meta.get("annotations") or {}).items()},
+        labels=dict(meta.get("labels") or {}),
+        tasks=tasks,
+    )
+    # RunPolicy is INLINED on spec (torchjob_types.go:182), incl. the
+    # reference's `clenPodPolicy` field spelling
+    clean = spec.get("clenPodPolicy") or spec.get("cleanPodPolicy")
+    job.run_policy = RunPolicy(
+        clean_task_policy=CleanPodPolicy(clean) if clean
+        else CleanPodPolicy.RUNNING,
+        ttl_seconds_after_finished=spec.get("TTLSecondsAfterFinished"),
+        active_deadline_seconds=spec.get("activeDurations"),
+        backoff_limit=int(spec.get("backoffLimit", 3)))
+    sp = spec.get("schedulingPolicy") or {}
+    job.scheduling = SchedulingPolicy(
+        min_available=sp.get("minAvailable"),
+        queue=sp.get("queue", ""),
+        priority=sp.get("priority"))
+    if spec.get("minMembers"):
+        job.min_members = {TaskType(t.lower()): int(v)
+                           for t, v in spec["minMembers"].items()}
+    mv = spec.get("modelVersion")
+    if mv:
+        job.model_name = mv.get("modelName") or mv.get("name")
+    if spec.get("enableTorchElastic"):
+        ep = spec.get("torchElasticPolicy") or {}
+        job.elastic = ElasticPolicy(
+            min_replicas=int(ep.get("numMinReplicas", 1)),
+            max_replicas=int(ep.get("numMaxReplicas", 1)),
+            nproc_per_node=int(ep.get("numWorkersPerNodePolicy", 1)))
+    return set_defaults(job)
+
+
+def job_to_crd_dict(job: TorchJob) -> dict:
+    """Emit the reference CRD shape (amd.com/gpu resources) — the
+    round-trip counterpart of _job_from_crd_dict."""
+    task_specs = {}
+    for t, s in job.tasks.items():
+        res: dict = {}
+        if s.gpus_per_task:
+            res["amd.com/gpu"] = s.gpus_per_task
+        if s.cpus_per_task:
+            res["cpu"] = s.cpus_per_task
+        if s.mem_mb_per_task:
+            res["memory"] = f"{s.mem_mb_per_task}Mi"
+        cont = {"name": "torch",
+                "env": [{"name": k, "value": v} for k, v in s.env.items()]}
+        if s.command:
+            cont["command"] = list(s.command)
+        if res:
+            cont["resources"] = {"limits": res}
+        d = {"numTasks": s.replicas,
+             "template": {"spec": {"containers": [cont]}}}
+        if s.restart_policy is not None:
+            inv = {v: k for k, v in _CRD_RESTART.items()}
+            d["restartPolicy"] = inv.get(s.restart_policy,
+                                         s.restart_policy.value)
+        if s.spot is not None:
+            d["spotTaskSpec"] = {"numSpotTasks": s.spot.num_spot_replicas,
+                                 "labels": dict(s.spot.labels)}
+        task_specs[t.value.capitalize() if t != TaskType.AIMASTER
+                   else "AIMaster"] = d
+    spec: dict = {"torchTaskSpecs": task_specs,
+                  "backoffLimit": job.run_policy.backoff_limit,
+                  "clenPodPolicy": job.run_policy.clean_task_policy.value}
+    if job.run_policy.ttl_seconds_after_finished is not None:
+        spec["TTLSecondsAfterFinished"] = \
+            job.run_policy.ttl_seconds_after_finished
+    if job.run_policy.active_deadline_seconds is not None:
+        spec["activeDurations"] = job.run_policy.active_deadline_seconds
+    if job.scheduling.min_available or job.scheduling.queue or \
+            job.scheduling.priority:
+        spec["schedulingPolicy"] = {
+            k: v for k, v in [("minAvailable", job.scheduling.min_available),
+                              ("queue", job.scheduling.queue or None),
+                              ("priority", job.scheduling.priority)]
+            if v is not None}
+    if job.min_members:
+        spec["minMembers"] = {
+            (t.value.capitalize() if t != TaskType.AIMASTER else "AIMaster"):
+            v for t, v in job.min_members.items()}
+    if job.model_name:
+        spec["modelVersion"] = {"modelName": job.model_name}
+    if job.elastic is not None:
+        spec["enableTorchElastic"] = True
+        spec["torchElasticPolicy"] = {
+            "numMinReplicas": job.elastic.min_replicas,
+            "numMaxReplicas": job.elastic.max_replicas,
+            "numWorkersPerNodePolicy": job.elastic.nproc_per_node,
+        }
+    return {
+        "apiVersion": "train.distributed.io/v1alpha1",
+        "kind": "TorchJob",
+        "metadata": {"name": job.name, "namespace": job.namespace,
+                     "annotations": dict(job.annotations),
+                     "labels": dict(job.labels)},
+        "spec": spec,
+    }
+
+
 def job_from_dict(doc: dict) -> TorchJob:
     meta = doc.get("metadata") or {}
     spec = doc.get("spec") or {}
+    if "torchTaskSpecs" in spec:  # reference CRD dialect
+        return _job_from_crd_dict(doc)
     tasks = {TaskType(t.lower()): _task_spec(s or {})
              for t, s in (spec.get("tasks") or {}).items()}
     job = TorchJob(
