@@ -55,6 +55,11 @@ def init_distributed(backend: str | None = None,
     if world > 1 and not dist.is_initialized():
         if backend is None:
             backend = "nccl" if use_gpu else "gloo"
+        if backend == "nccl":
+            # xGMI-aware RCCL tuning (flagged; parallel/rccl.py) must be
+            # in the env before communicator init
+            from torch_on_k8s_amd.parallel.rccl import apply_rccl_env
+            apply_rccl_env(world)
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "23456")
         dist.init_process_group(
