@@ -47,8 +47,8 @@ def test_activation_checkpointing_matches():
     l1.backward()
     l2.backward()
     assert torch.allclose(l1, l2, atol=1e-6)
-    g1 = m1.layers[0].attn.q_proj.weight.grad
-    g2 = m2.layers[0].attn.q_proj.weight.grad
+    g1 = m1.layers[0].attn.qkv_proj.weight.grad
+    g2 = m2.layers[0].attn.qkv_proj.weight.grad
     assert torch.allclose(g1, g2, atol=1e-5)
 
 

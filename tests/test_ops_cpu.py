@@ -87,3 +87,38 @@ def test_attention_ref_vs_sdpa():
         v.permute(0, 2, 1, 3).repeat_interleave(rep, dim=1),
         is_causal=True).permute(0, 2, 1, 3)
     assert torch.allclose(o, o_ref, atol=1e-5)
+
+
+def test_swiglu_cpu_matches_eager():
+    torch.manual_seed(0)
+    gu = torch.randn(64, 128, requires_grad=True)
+    out = ops.swiglu(gu)
+    g, u = gu.detach().split(64, dim=-1)
+    ref = torch.nn.functional.silu(g) * u
+    assert torch.allclose(out, ref, atol=1e-5)
+    out.sum().backward()
+    gu2 = gu.detach().clone().requires_grad_(True)
+    g2, u2 = gu2.split(64, dim=-1)
+    (torch.nn.functional.silu(g2) * u2).sum().backward()
+    assert torch.allclose(gu.grad, gu2.grad, atol=1e-5)
+
+
+def test_qkv_rope_cpu_matches_unpacked():
+    torch.manual_seed(1)
+    from torch_on_k8s_amd.models.llama import build_rope_table, get_config
+    B, S, Hq, Hkv, D = 2, 16, 4, 2, 64
+    cfg = get_config("llama-tiny", head_dim=D)
+    cos, sin = build_rope_table(cfg, S, torch.device("cpu"))
+    qkv = torch.randn(B, S, (Hq + 2 * Hkv) * D, requires_grad=True)
+    q, k, v = ops.qkv_rope(qkv, cos, sin, Hq, Hkv, D)
+    parts = qkv.detach().view(B, S, Hq + 2 * Hkv, D)
+    q_ref = ops.rope_ref(parts[:, :, :Hq], cos, sin)
+    k_ref = ops.rope_ref(parts[:, :, Hq:Hq + Hkv], cos, sin)
+    v_ref = parts[:, :, Hq + Hkv:]
+    assert torch.allclose(q, q_ref, atol=1e-5)
+    assert torch.allclose(k, k_ref, atol=1e-5)
+    assert torch.allclose(v, v_ref, atol=1e-5)
+    # backward: rope grad is the inverse rotation (orthogonal), so
+    # grad-of-sum through q+k+v must equal inverse-roped ones
+    (q.sum() + k.sum() + v.sum()).backward()
+    assert qkv.grad is not None and qkv.grad.shape == qkv.shape

@@ -206,3 +206,83 @@ def test_native_extension_is_in_tree():
         os.path.abspath(torch_on_k8s_amd.__file__)))
     assert ops._C.__file__.startswith(pkg_root), ops._C.__file__
     assert "site-packages" not in ops._C.__file__
+
+
+def test_swiglu_fwd_bwd():
+    """Fused SwiGLU over packed gate_up vs fp32 reference."""
+    torch.manual_seed(0)
+    rows, I = 1024, 1024
+    gu = torch.randn(rows, 2 * I).bfloat16().float()
+    dout = torch.randn(rows, I).bfloat16().float()
+
+    ref_gu = gu.clone().requires_grad_(True)
+    ref = ops.swiglu_ref(ref_gu.bfloat16()).float()
+    # fp32 manual backward
+    g, u = gu.split(I, dim=-1)
+    sg = torch.sigmoid(g)
+    dg_ref = dout * u * (sg * (1 + g * (1 - sg)))
+    du_ref = dout * (g * sg)
+
+    gug = gu.bfloat16().to(dev()).requires_grad_(True)
+    out = ops.swiglu(gug)
+    out.backward(dout.bfloat16().to(dev()))
+    err_f = (out.float().cpu() - ref).abs().max().item()
+    dgu = gug.grad.float().cpu()
+    err_g = (dgu[:, :I] - dg_ref).abs().max().item()
+    err_u = (dgu[:, I:] - du_ref).abs().max().item()
+    assert err_f < 3e-2, f"swiglu fwd err {err_f}"
+    assert err_g < 3e-2, f"swiglu dgate err {err_g}"
+    assert err_u < 3e-2, f"swiglu dup err {err_u}"
+
+
+def test_qkv_rope_fwd_bwd():
+    """Packed-QKV split+rope kernel vs the fp32 reference path."""
+    torch.manual_seed(0)
+    B, S, Hq, Hkv, D = 2, 128, 8, 4, 128
+    from torch_on_k8s_amd.models.llama import build_rope_table, get_config
+    cfg = get_config("llama-tiny", head_dim=D)
+    cos, sin = build_rope_table(cfg, S, torch.device("cpu"))
+    qkv = torch.randn(B, S, (Hq + 2 * Hkv) * D).bfloat16().float()
+    dq = torch.randn(B, S, Hq, D).bfloat16().float()
+    dk = torch.randn(B, S, Hkv, D).bfloat16().float()
+    dv = torch.randn(B, S, Hkv, D).bfloat16().float()
+
+    q_ref, k_ref, v_ref = ops.qkv_rope_ref(qkv, cos, sin, Hq, Hkv, D)
+
+    qkv_g = qkv.bfloat16().to(dev()).requires_grad_(True)
+    q, k, v = ops.qkv_rope(qkv_g, cos.to(dev()), sin.to(dev()), Hq, Hkv, D)
+    for got, ref, name in [(q, q_ref, "q"), (k, k_ref, "k"), (v, v_ref, "v")]:
+        err = (got.float().cpu() - ref.float()).abs().max().item()
+        assert err < 2e-2, f"qkv_rope {name} err {err}"
+    torch.autograd.backward([q, k, v], [dq.bfloat16().to(dev()),
+                                        dk.bfloat16().to(dev()),
+                                        dv.bfloat16().to(dev())])
+    # reference backward: inverse-rotate dq/dk, concat with dv
+    dq_r = ops.rope_ref(dq, cos, sin, -1.0).reshape(B, S, Hq * D)
+    dk_r = ops.rope_ref(dk, cos, sin, -1.0).reshape(B, S, Hkv * D)
+    dqkv_ref = torch.cat([dq_r, dk_r, dv.reshape(B, S, Hkv * D)], dim=-1)
+    err = (qkv_g.grad.float().cpu() - dqkv_ref.float()).abs().max().item()
+    assert err < 2e-2, f"qkv_rope bwd err {err}"
+
+
+def test_rmsnorm_dw_two_stage_large():
+    """dw over many rows (the two-stage reduction path) matches fp32."""
+    torch.manual_seed(3)
+    rows, H = 16384, 4096
+    x = torch.randn(rows, H).bfloat16().float()
+    dy = torch.randn(rows, H).bfloat16().float()
+    w = torch.ones(H).bfloat16().float()
+    xg = x.bfloat16().to(dev()).requires_grad_(True)
+    y = ops.rmsnorm(xg, w.bfloat16().to(dev()), 1e-5)
+    y.backward(dy.bfloat16().to(dev()))
+    # fp32 reference dw
+    r = torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + 1e-5)
+    dw_ref = (dy * x * r).sum(0)
+    # kernel dw is on the weight's grad — recompute through the module path
+    wp = w.bfloat16().to(dev()).requires_grad_(True)
+    xg2 = x.bfloat16().to(dev())
+    y2 = ops.rmsnorm(xg2, wp, 1e-5)
+    y2.backward(dy.bfloat16().to(dev()))
+    rel = ((wp.grad.float().cpu() - dw_ref).abs() /
+           (dw_ref.abs() + 1.0)).max().item()
+    assert rel < 2e-2, f"rmsnorm dw two-stage rel err {rel}"

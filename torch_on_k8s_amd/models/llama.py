@@ -93,24 +93,31 @@ def build_rope_table(cfg: LlamaConfig, seq_len: int, device) -> tuple[torch.Tens
 
 
 class Attention(nn.Module):
+    """q/k/v projections PACKED into one GEMM (fewer, larger hipBLASLt
+    calls — ROADMAP §1a); the packed output is split + roped by the
+    single-pass qkv_rope kernel. Head order in qkv_proj's output:
+    [q heads | k heads | v heads]."""
+
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
         self.cfg = cfg
         H, D = cfg.num_heads, cfg.head_dim
         Hkv = cfg.num_kv_heads
-        self.q_proj = nn.Linear(cfg.hidden_size, H * D, bias=False)
-        self.k_proj = nn.Linear(cfg.hidden_size, Hkv * D, bias=False)
-        self.v_proj = nn.Linear(cfg.hidden_size, Hkv * D, bias=False)
+        self.qkv_proj = nn.Linear(cfg.hidden_size, (H + 2 * Hkv) * D,
+                                  bias=False)
         self.o_proj = nn.Linear(H * D, cfg.hidden_size, bias=False)
+
+    def project_qkv(self, x, cos, sin):
+        """Packed projection -> roped contiguous q, k, v."""
+        cfg = self.cfg
+        qkv = self.qkv_proj(x)
+        return ops.qkv_rope(qkv, cos, sin, cfg.num_heads, cfg.num_kv_heads,
+                            cfg.head_dim)
 
     def forward(self, x, cos, sin):
         B, S, _ = x.shape
         cfg = self.cfg
-        q = self.q_proj(x).view(B, S, cfg.num_heads, cfg.head_dim)
-        k = self.k_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
-        v = self.v_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
-        q = ops.apply_rope(q, cos, sin)
-        k = ops.apply_rope(k, cos, sin)
+        q, k, v = self.project_qkv(x, cos, sin)
         if cfg.attn_impl == "sdpa" and x.is_cuda:
             rep = cfg.num_heads // cfg.num_kv_heads
             qt = q.transpose(1, 2)
@@ -124,14 +131,18 @@ class Attention(nn.Module):
 
 
 class MLP(nn.Module):
+    """gate/up packed into one GEMM; fused SwiGLU kernel reads the
+    packed output directly (one pass fwd, one pass bwd, no silu
+    intermediate in HBM)."""
+
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
-        self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
-        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
+        self.gate_up_proj = nn.Linear(cfg.hidden_size,
+                                      2 * cfg.intermediate_size, bias=False)
         self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
 
     def forward(self, x):
-        return self.down_proj(F.silu(self.gate_proj(x)) * self.up_proj(x))
+        return self.down_proj(ops.swiglu(self.gate_up_proj(x)))
 
 
 class Block(nn.Module):
@@ -232,13 +243,9 @@ class LlamaModel(nn.Module):
             the caches and attends over cache[:T]."""
             B_, s, _ = x.shape
             h = blk.input_norm(x)
-            q = blk.attn.q_proj(h).view(B_, s, Hq, D)
-            k = blk.attn.k_proj(h).view(B_, s, Hkv, D)
-            v = blk.attn.v_proj(h).view(B_, s, Hkv, D)
-            q = ops.apply_rope(q, cos[pos0:pos0 + s].contiguous(),
-                               sin[pos0:pos0 + s].contiguous())
-            k = ops.apply_rope(k, cos[pos0:pos0 + s].contiguous(),
-                               sin[pos0:pos0 + s].contiguous())
+            q, k, v = blk.attn.project_qkv(
+                h, cos[pos0:pos0 + s].contiguous(),
+                sin[pos0:pos0 + s].contiguous())
             kc[i][:, pos0:pos0 + s] = k
             vc[i][:, pos0:pos0 + s] = v
             if s > 1:  # prefill
@@ -294,11 +301,7 @@ class LlamaModel(nn.Module):
             sn = torch.index_select(sin, 0, pos_long).contiguous()
             for i, blk in enumerate(self.layers):
                 h = blk.input_norm(x)
-                q = blk.attn.q_proj(h).view(B, 1, Hq, D)
-                k = blk.attn.k_proj(h).view(B, 1, Hkv, D)
-                v = blk.attn.v_proj(h).view(B, 1, Hkv, D)
-                q = ops.apply_rope(q, cs, sn)
-                k = ops.apply_rope(k, cs, sn)
+                q, k, v = blk.attn.project_qkv(h, cs, sn)
                 kc[i].index_copy_(1, pos_long, k)
                 vc[i].index_copy_(1, pos_long, v)
                 o = ops.attention_decode(q[:, 0], kc[i], vc[i], 0, T_dev=t32)

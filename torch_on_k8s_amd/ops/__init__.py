@@ -118,6 +118,93 @@ def apply_rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.T
 
 
 # --------------------------------------------------------------------------
+# Packed-QKV split + RoPE (one pass over the fused qkv projection output)
+# --------------------------------------------------------------------------
+def qkv_rope_ref(qkv: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+                 Hq: int, Hkv: int, D: int, sign: float = 1.0):
+    """fp32 reference. qkv: [B, S, (Hq+2Hkv)*D]; returns roped q, k and
+    copied v with contiguous [B,S,H,D] layouts."""
+    B, S, _ = qkv.shape
+    parts = qkv.view(B, S, Hq + 2 * Hkv, D)
+    q = parts[:, :, :Hq].contiguous()
+    k = parts[:, :, Hq:Hq + Hkv].contiguous()
+    v = parts[:, :, Hq + Hkv:].contiguous()
+    return (rope_ref(q, cos, sin, sign), rope_ref(k, cos, sin, sign), v)
+
+
+class _QKVRopeFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, qkv, cos, sin, Hq, Hkv, D):
+        ctx.save_for_backward(cos, sin)
+        ctx.dims = (Hq, Hkv, D)
+        if qkv.is_cuda:
+            q, k, v = _require_ext("qkv_rope").qkv_rope_fwd(
+                qkv.contiguous(), cos, sin, Hq, Hkv, D)
+            return q, k, v
+        return qkv_rope_ref(qkv, cos, sin, Hq, Hkv, D)
+
+    @staticmethod
+    def backward(ctx, dq, dk, dv):
+        cos, sin = ctx.saved_tensors
+        Hq, Hkv, D = ctx.dims
+        if dq.is_cuda:
+            dqkv = _C.qkv_rope_bwd(dq.contiguous(), dk.contiguous(),
+                                   dv.contiguous(), cos, sin)
+        else:
+            B, S = dq.shape[:2]
+            dqf = rope_ref(dq, cos, sin, -1.0).view(B, S, Hq * D)
+            dkf = rope_ref(dk, cos, sin, -1.0).view(B, S, Hkv * D)
+            dqkv = torch.cat([dqf, dkf, dv.reshape(B, S, Hkv * D)], dim=-1)
+        return dqkv, None, None, None, None, None
+
+
+def qkv_rope(qkv: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+             Hq: int, Hkv: int, D: int):
+    """Split the packed qkv projection [B,S,(Hq+2Hkv)*D] into contiguous
+    q/k/v with RoPE applied to q and k — one kernel instead of three
+    splits + two rope launches (GEMM-packing lever, ROADMAP §1a)."""
+    return _QKVRopeFn.apply(qkv, cos, sin, Hq, Hkv, D)
+
+
+# --------------------------------------------------------------------------
+# Fused SwiGLU over the packed gate_up projection
+# --------------------------------------------------------------------------
+def swiglu_ref(gu: torch.Tensor) -> torch.Tensor:
+    """fp32 reference: silu(gate) * up over packed [..., 2I]."""
+    I = gu.shape[-1] // 2
+    g, u = gu.float().split(I, dim=-1)
+    return (torch.nn.functional.silu(g) * u).to(gu.dtype)
+
+
+class _SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gu):
+        ctx.save_for_backward(gu)
+        if gu.is_cuda:
+            return _require_ext("swiglu").swiglu_fwd(gu.contiguous())
+        return swiglu_ref(gu)
+
+    @staticmethod
+    def backward(ctx, dout):
+        (gu,) = ctx.saved_tensors
+        if gu.is_cuda:
+            return _C.swiglu_bwd(dout.contiguous(), gu)
+        I = gu.shape[-1] // 2
+        g, u = gu.float().split(I, dim=-1)
+        sg = torch.sigmoid(g)
+        silu = g * sg
+        dsilu = sg * (1 + g * (1 - sg))
+        df = dout.float()
+        return torch.cat([df * u * dsilu, df * silu], dim=-1).to(gu.dtype)
+
+
+def swiglu(gu: torch.Tensor) -> torch.Tensor:
+    """silu(gate)*up over the PACKED gate_up output [..., 2I]; fwd+bwd
+    each one fused pass (no materialized silu intermediate)."""
+    return _SwiGLUFn.apply(gu)
+
+
+# --------------------------------------------------------------------------
 # Fused AdamW on a flat bucket
 # --------------------------------------------------------------------------
 def fused_adamw_(p: torch.Tensor, g: torch.Tensor, m: torch.Tensor,
@@ -300,5 +387,6 @@ def cross_entropy(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
 __all__ = [
     "rmsnorm", "rmsnorm_ref", "apply_rope", "rope_ref", "fused_adamw_",
     "attention", "attention_ref", "cross_entropy", "layernorm",
+    "qkv_rope", "qkv_rope_ref", "swiglu", "swiglu_ref",
     "hip_ext_available",
 ]

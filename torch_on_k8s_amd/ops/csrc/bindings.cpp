@@ -2,17 +2,35 @@
 //
 // The kernels themselves live in pure handwritten HIP files (*.hip); this
 // translation unit only does tensor checking, workspace allocation and
-// stream plumbing.
+// stream plumbing. Stream access uses PyTorch-ROCm's native HIP stream
+// API (at::hip::getCurrentHIPStreamMasqueradingAsCUDA — the real symbol
+// exported by a ROCm build of ATen) so this TU contains no CUDA-named
+// calls for hipify to rewrite.
 #include <torch/extension.h>
-#include <ATen/cuda/CUDAContext.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <hip/hip_runtime.h>
 
 extern "C" {
 hipError_t tok_rmsnorm_fwd(const void* x, const void* w, void* y, float* invr,
                            long nrows, int H, float eps, hipStream_t stream);
+int tok_rmsnorm_dw_rsplit(long nrows, int H);
 hipError_t tok_rmsnorm_bwd(const void* x, const void* w, const void* dy,
                            const float* invr, void* dx, float* dw_f32,
-                           long nrows, int H, hipStream_t stream);
+                           float* dw_ws, long nrows, int H,
+                           hipStream_t stream);
+hipError_t tok_swiglu_fwd(const void* gu, void* out, long rows, int I,
+                          hipStream_t stream);
+hipError_t tok_swiglu_bwd(const void* dout, const void* gu, void* dgu,
+                          long rows, int I, hipStream_t stream);
+hipError_t tok_qkv_rope_fwd(const void* qkv, void* q, void* k, void* v,
+                            const float* cos_tab, const float* sin_tab,
+                            long tokens, int Hq, int Hkv, int D,
+                            long table_rows, hipStream_t stream);
+hipError_t tok_qkv_rope_bwd(const void* dq, const void* dk, const void* dv,
+                            void* dqkv, const float* cos_tab,
+                            const float* sin_tab, long tokens, int Hq,
+                            int Hkv, int D, long table_rows,
+                            hipStream_t stream);
 hipError_t tok_rope(const void* x, void* y, const float* cos_tab,
                     const float* sin_tab, long rows_total, int heads, int D,
                     long table_rows, float sign, hipStream_t stream);
@@ -68,7 +86,7 @@ namespace {
   } while (0)
 
 hipStream_t current_stream() {
-  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 }
 
 std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
@@ -94,12 +112,85 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor dy,
   const long H = x.size(-1);
   const long nrows = x.numel() / H;
   auto dx = at::empty_like(x);
-  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto dw = at::empty({H}, x.options().dtype(at::kFloat));
+  const int rsplit = tok_rmsnorm_dw_rsplit(nrows, (int)H);
+  auto ws = at::empty({(long)rsplit * H}, x.options().dtype(at::kFloat));
   TOK_HIP_OK(tok_rmsnorm_bwd(x.data_ptr(), w.data_ptr(), dy.data_ptr(),
                              invr.data_ptr<float>(), dx.data_ptr(),
-                             dw.data_ptr<float>(), nrows, (int)H,
-                             current_stream()));
+                             dw.data_ptr<float>(), ws.data_ptr<float>(),
+                             nrows, (int)H, current_stream()));
   return {dx, dw};
+}
+
+// gu: [rows, 2I] packed gate|up -> out [rows, I] = silu(gate)*up
+at::Tensor swiglu_fwd(at::Tensor gu) {
+  CHECK_BF16_CUDA(gu);
+  const long I2 = gu.size(-1);
+  TORCH_CHECK(I2 % 16 == 0, "packed gate_up dim must be a multiple of 16");
+  const long I = I2 / 2;
+  const long rows = gu.numel() / I2;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto out = at::empty(sizes, gu.options());
+  TOK_HIP_OK(tok_swiglu_fwd(gu.data_ptr(), out.data_ptr(), rows, (int)I,
+                            current_stream()));
+  return out;
+}
+
+at::Tensor swiglu_bwd(at::Tensor dout, at::Tensor gu) {
+  CHECK_BF16_CUDA(dout);
+  CHECK_BF16_CUDA(gu);
+  const long I2 = gu.size(-1);
+  const long I = I2 / 2;
+  const long rows = gu.numel() / I2;
+  TORCH_CHECK(dout.numel() == rows * I, "dout shape mismatch");
+  auto dgu = at::empty_like(gu);
+  TOK_HIP_OK(tok_swiglu_bwd(dout.data_ptr(), gu.data_ptr(), dgu.data_ptr(),
+                            rows, (int)I, current_stream()));
+  return dgu;
+}
+
+// qkv: [B, S, (Hq+2Hkv)*D] -> roped q [B,S,Hq,D], k [B,S,Hkv,D], v [B,S,Hkv,D]
+std::vector<at::Tensor> qkv_rope_fwd(at::Tensor qkv, at::Tensor cos_tab,
+                                     at::Tensor sin_tab, long Hq, long Hkv,
+                                     long D) {
+  CHECK_BF16_CUDA(qkv);
+  TORCH_CHECK(cos_tab.scalar_type() == at::kFloat && cos_tab.is_contiguous());
+  TORCH_CHECK(sin_tab.scalar_type() == at::kFloat && sin_tab.is_contiguous());
+  TORCH_CHECK(D % 16 == 0, "head dim must be a multiple of 16");
+  const long B = qkv.size(0), S = qkv.size(1);
+  TORCH_CHECK(qkv.size(2) == (Hq + 2 * Hkv) * D, "packed qkv dim mismatch");
+  const long tokens = B * S;
+  const long table_rows = cos_tab.numel() / (D / 2);
+  TORCH_CHECK(S % table_rows == 0 || table_rows % S == 0 || table_rows == S,
+              "rope table rows must match sequence");
+  auto q = at::empty({B, S, Hq, D}, qkv.options());
+  auto k = at::empty({B, S, Hkv, D}, qkv.options());
+  auto v = at::empty({B, S, Hkv, D}, qkv.options());
+  TOK_HIP_OK(tok_qkv_rope_fwd(qkv.data_ptr(), q.data_ptr(), k.data_ptr(),
+                              v.data_ptr(), cos_tab.data_ptr<float>(),
+                              sin_tab.data_ptr<float>(), tokens, (int)Hq,
+                              (int)Hkv, (int)D, table_rows,
+                              current_stream()));
+  return {q, k, v};
+}
+
+at::Tensor qkv_rope_bwd(at::Tensor dq, at::Tensor dk, at::Tensor dv,
+                        at::Tensor cos_tab, at::Tensor sin_tab) {
+  CHECK_BF16_CUDA(dq);
+  CHECK_BF16_CUDA(dk);
+  CHECK_BF16_CUDA(dv);
+  const long B = dq.size(0), S = dq.size(1), Hq = dq.size(2), D = dq.size(3);
+  const long Hkv = dk.size(2);
+  const long tokens = B * S;
+  const long table_rows = cos_tab.numel() / (D / 2);
+  auto dqkv = at::empty({B, S, (Hq + 2 * Hkv) * D}, dq.options());
+  TOK_HIP_OK(tok_qkv_rope_bwd(dq.data_ptr(), dk.data_ptr(), dv.data_ptr(),
+                              dqkv.data_ptr(), cos_tab.data_ptr<float>(),
+                              sin_tab.data_ptr<float>(), tokens, (int)Hq,
+                              (int)Hkv, (int)D, table_rows,
+                              current_stream()));
+  return dqkv;
 }
 
 at::Tensor rope(at::Tensor x, at::Tensor cos_tab, at::Tensor sin_tab,
@@ -313,6 +404,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ce_bwd", &ce_bwd, "Fused cross-entropy backward (bf16, gfx950)");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16, gfx950)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (bf16, gfx950)");
+  mod.def("swiglu_fwd", &swiglu_fwd,
+          "Fused SwiGLU over packed gate_up (bf16, gfx950)");
+  mod.def("swiglu_bwd", &swiglu_bwd,
+          "Fused SwiGLU backward (bf16, gfx950)");
+  mod.def("qkv_rope_fwd", &qkv_rope_fwd,
+          "Packed-QKV split + RoPE (bf16, gfx950)");
+  mod.def("qkv_rope_bwd", &qkv_rope_bwd,
+          "Packed-QKV gather + inverse RoPE (bf16, gfx950)");
   mod.def("rope", &rope, "Rotary embedding rotate-half (bf16, gfx950)");
   mod.def("adamw_", &adamw_, "Fused AdamW on a flat bucket (gfx950)",
           py::arg("p"), py::arg("g"), py::arg("m"), py::arg("v"),
