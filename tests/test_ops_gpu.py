@@ -226,13 +226,18 @@ def test_swiglu_fwd_bwd():
     gug = gu.bfloat16().to(dev()).requires_grad_(True)
     out = ops.swiglu(gug)
     out.backward(dout.bfloat16().to(dev()))
+    # bf16 output rounding scales with magnitude (|dg| can reach ~10 for
+    # products of three ~N(0,1) terms): bound RELATIVE to the max ref
+    def tol(ref_t):
+        return max(3e-2, 2.5 / 256 * ref_t.abs().max().item())
+
     err_f = (out.float().cpu() - ref).abs().max().item()
     dgu = gug.grad.float().cpu()
     err_g = (dgu[:, :I] - dg_ref).abs().max().item()
     err_u = (dgu[:, I:] - du_ref).abs().max().item()
-    assert err_f < 3e-2, f"swiglu fwd err {err_f}"
-    assert err_g < 3e-2, f"swiglu dgate err {err_g}"
-    assert err_u < 3e-2, f"swiglu dup err {err_u}"
+    assert err_f < tol(ref), f"swiglu fwd err {err_f}"
+    assert err_g < tol(dg_ref), f"swiglu dgate err {err_g}"
+    assert err_u < tol(du_ref), f"swiglu dup err {err_u}"
 
 
 def test_qkv_rope_fwd_bwd():
@@ -286,3 +291,56 @@ def test_rmsnorm_dw_two_stage_large():
     rel = ((wp.grad.float().cpu() - dw_ref).abs() /
            (dw_ref.abs() + 1.0)).max().item()
     assert rel < 2e-2, f"rmsnorm dw two-stage rel err {rel}"
+
+
+def test_rmsnorm_residual_fused():
+    """Fused residual+RMSNorm (fwd: y, xr; bwd: dx includes dxr addend)
+    vs the unfused fp32 reference."""
+    torch.manual_seed(5)
+    rows, H = 512, 1024
+    x = torch.randn(rows, H).bfloat16().float()
+    res = torch.randn(rows, H).bfloat16().float()
+    w = torch.randn(H).bfloat16().float()
+    dy = torch.randn(rows, H).bfloat16().float()
+    dxr_in = torch.randn(rows, H).bfloat16().float()
+
+    xg = x.bfloat16().to(dev()).requires_grad_(True)
+    rg = res.bfloat16().to(dev()).requires_grad_(True)
+    wg = w.bfloat16().to(dev()).requires_grad_(True)
+    y, xr = ops.rmsnorm_residual(xg, rg, wg, 1e-5)
+    torch.autograd.backward([y, xr], [dy.bfloat16().to(dev()),
+                                      dxr_in.bfloat16().to(dev())])
+
+    # fp32 reference (with the bf16 residual-sum rounding the kernel does)
+    xr_ref = (x + res)
+    xr_q = xr_ref.to(torch.bfloat16).float()
+    r = torch.rsqrt(xr_q.pow(2).mean(-1, keepdim=True) + 1e-5)
+    y_ref = xr_q * r * w
+    c = (dy * w * xr_q).sum(-1, keepdim=True)
+    dx_ref = r * (w * dy - xr_q * (r * r / H) * c) + dxr_in
+    dw_ref = (dy * xr_q * r).sum(0)
+
+    assert (y.float().cpu() - y_ref).abs().max().item() < 3e-2
+    assert (xr.float().cpu() - xr_q).abs().max().item() < 2e-2
+    assert (xg.grad.float().cpu() - dx_ref).abs().max().item() < 4e-2
+    assert (rg.grad.float().cpu() - dx_ref).abs().max().item() < 4e-2
+    rel = ((wg.grad.float().cpu() - dw_ref).abs() /
+           (dw_ref.abs() + 1.0)).max().item()
+    assert rel < 3e-2, f"dw rel err {rel}"
+
+
+def test_llama_tiny_fused_block_matches_cpu():
+    """End-to-end: tiny model loss/grads on GPU (all fused kernels) match
+    the CPU fp32 path within bf16 tolerance."""
+    from torch_on_k8s_amd.models.llama import LlamaModel, get_config
+    torch.manual_seed(0)
+    cfg = get_config("llama-tiny")
+    m_cpu = LlamaModel(cfg)
+    m_gpu = LlamaModel(cfg).to(dev()).bfloat16()
+    m_gpu.load_state_dict({k: v.bfloat16()
+                           for k, v in m_cpu.state_dict().items()})
+    ids = torch.randint(0, cfg.vocab_size, (2, 64))
+    l_cpu = m_cpu(ids, ids)
+    l_gpu = m_gpu(ids.to(dev()), ids.to(dev()))
+    assert abs(l_cpu.item() - l_gpu.float().item()) < 0.05, \
+        (l_cpu.item(), l_gpu.float().item())

@@ -146,6 +146,12 @@ class MLP(nn.Module):
 
 
 class Block(nn.Module):
+    """Residual stream threaded as (branch, residual) pairs so every
+    residual-add fuses into the NEXT RMSNorm's read (ops.rmsnorm_residual:
+    one kernel does add + norm fwd, and norm-dx + residual-grad add bwd).
+    forward(x, res) returns (mlp_out, x_after_attn_add): the caller (next
+    block or final norm) performs the pending add inside its fused norm."""
+
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
         self.input_norm = RMSNorm(cfg.hidden_size, cfg.rms_eps)
@@ -153,10 +159,13 @@ class Block(nn.Module):
         self.post_attn_norm = RMSNorm(cfg.hidden_size, cfg.rms_eps)
         self.mlp = MLP(cfg)
 
-    def forward(self, x, cos, sin):
-        x = x + self.attn(self.input_norm(x), cos, sin)
-        x = x + self.mlp(self.post_attn_norm(x))
-        return x
+    def forward(self, x, res, cos, sin):
+        h, xr = ops.rmsnorm_residual(x, res, self.input_norm.weight,
+                                     self.input_norm.eps)
+        a = self.attn(h, cos, sin)
+        h2, xr2 = ops.rmsnorm_residual(a, xr, self.post_attn_norm.weight,
+                                       self.post_attn_norm.eps)
+        return self.mlp(h2), xr2
 
 
 class LlamaModel(nn.Module):
@@ -202,13 +211,15 @@ class LlamaModel(nn.Module):
         B, S = input_ids.shape
         cos, sin = self._rope(S, input_ids.device)
         x = self.embed_tokens(input_ids)
+        res = None  # pending residual, consumed by each fused norm
         for blk in self.layers:
             if self.activation_checkpointing and self.training:
-                x = torch.utils.checkpoint.checkpoint(
-                    blk, x, cos, sin, use_reentrant=False)
+                x, res = torch.utils.checkpoint.checkpoint(
+                    blk, x, res, cos, sin, use_reentrant=False)
             else:
-                x = blk(x, cos, sin)
-        return self.norm(x)
+                x, res = blk(x, res, cos, sin)
+        y, _ = ops.rmsnorm_residual(x, res, self.norm.weight, self.norm.eps)
+        return y
 
     def forward(self, input_ids: torch.Tensor,
                 labels: torch.Tensor | None = None):

@@ -79,6 +79,55 @@ def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor
     return _RMSNormFn.apply(x.contiguous(), w, eps)
 
 
+class _RMSNormResFn(torch.autograd.Function):
+    """Fused residual-add + RMSNorm: (x, res, w) -> (y, xr) with
+    xr = x + res (bf16) and y = rmsnorm(xr) * w. Backward fuses the
+    downstream residual grad (dxr) into the norm's dx pass, and dres
+    equals dx (the add routes gradients unchanged)."""
+
+    @staticmethod
+    def forward(ctx, x, res, w, eps):
+        if x.is_cuda:
+            y, xr, invr = _require_ext("rmsnorm_res").rmsnorm_res_fwd(
+                x, res, w, eps)
+        else:
+            xr = x + res if res is not None else x
+            xf = xr.float()
+            invr = torch.rsqrt(xf.pow(2).mean(-1) + eps).reshape(-1)
+            y = (xf * invr.view(*xr.shape[:-1], 1) * w.float()).to(x.dtype)
+        ctx.save_for_backward(xr, w, invr)
+        ctx.has_res = res is not None
+        return y, xr
+
+    @staticmethod
+    def backward(ctx, dy, dxr):
+        xr, w, invr = ctx.saved_tensors
+        dy = dy.contiguous()
+        if dxr is not None:
+            dxr = dxr.contiguous()
+        if xr.is_cuda:
+            dx, dw = _C.rmsnorm_res_bwd(xr, w, dy, dxr, invr)
+        else:
+            xf, wf, dyf = xr.float(), w.float(), dy.float()
+            H = xr.shape[-1]
+            r = invr.view(*xr.shape[:-1], 1)
+            c = (dyf * wf * xf).sum(-1, keepdim=True)
+            dx = (r * (wf * dyf - xf * (r * r / H) * c)).to(xr.dtype)
+            if dxr is not None:
+                dx = dx + dxr
+            dw = (dyf * xf * r).reshape(-1, H).sum(0)
+        dres = dx if ctx.has_res else None
+        return dx, dres, dw.to(w.dtype), None
+
+
+def rmsnorm_residual(x: torch.Tensor, res: torch.Tensor | None,
+                     w: torch.Tensor, eps: float = 1e-5):
+    """Returns (normed, x+res). One pass instead of add + norm; the
+    residual stream stays fused through backward too."""
+    return _RMSNormResFn.apply(
+        x.contiguous(), res.contiguous() if res is not None else None, w, eps)
+
+
 # --------------------------------------------------------------------------
 # RoPE (Llama rotate-half convention)
 # --------------------------------------------------------------------------

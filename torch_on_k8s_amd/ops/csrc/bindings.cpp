@@ -18,6 +18,13 @@ hipError_t tok_rmsnorm_bwd(const void* x, const void* w, const void* dy,
                            const float* invr, void* dx, float* dw_f32,
                            float* dw_ws, long nrows, int H,
                            hipStream_t stream);
+hipError_t tok_rmsnorm_res_fwd(const void* x, const void* res, const void* w,
+                               void* xr, void* y, float* invr, long nrows,
+                               int H, float eps, hipStream_t stream);
+hipError_t tok_rmsnorm_res_bwd(const void* xr, const void* w, const void* dy,
+                               const void* dxr_in, const float* invr,
+                               void* dx, float* dw_f32, float* dw_ws,
+                               long nrows, int H, hipStream_t stream);
 hipError_t tok_swiglu_fwd(const void* gu, void* out, long rows, int I,
                           hipStream_t stream);
 hipError_t tok_swiglu_bwd(const void* dout, const void* gu, void* dgu,
@@ -119,6 +126,55 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor dy,
                              invr.data_ptr<float>(), dx.data_ptr(),
                              dw.data_ptr<float>(), ws.data_ptr<float>(),
                              nrows, (int)H, current_stream()));
+  return {dx, dw};
+}
+
+// Fused residual + RMSNorm: (x, res?, w) -> (y, xr, invr)
+std::vector<at::Tensor> rmsnorm_res_fwd(at::Tensor x,
+                                        c10::optional<at::Tensor> res,
+                                        at::Tensor w, double eps) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(w);
+  const long H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "hidden dim must be a multiple of 8");
+  const long nrows = x.numel() / H;
+  const void* res_p = nullptr;
+  if (res.has_value()) {
+    CHECK_BF16_CUDA((*res));
+    TORCH_CHECK(res->sizes() == x.sizes(), "residual shape mismatch");
+    res_p = res->data_ptr();
+  }
+  auto xr = at::empty_like(x);
+  auto y = at::empty_like(x);
+  auto invr = at::empty({nrows}, x.options().dtype(at::kFloat));
+  TOK_HIP_OK(tok_rmsnorm_res_fwd(x.data_ptr(), res_p, w.data_ptr(),
+                                 xr.data_ptr(), y.data_ptr(),
+                                 invr.data_ptr<float>(), nrows, (int)H,
+                                 (float)eps, current_stream()));
+  return {y, xr, invr};
+}
+
+std::vector<at::Tensor> rmsnorm_res_bwd(at::Tensor xr, at::Tensor w,
+                                        at::Tensor dy,
+                                        c10::optional<at::Tensor> dxr,
+                                        at::Tensor invr) {
+  CHECK_BF16_CUDA(xr);
+  CHECK_BF16_CUDA(dy);
+  const long H = xr.size(-1);
+  const long nrows = xr.numel() / H;
+  const void* dxr_p = nullptr;
+  if (dxr.has_value()) {
+    CHECK_BF16_CUDA((*dxr));
+    dxr_p = dxr->data_ptr();
+  }
+  auto dx = at::empty_like(xr);
+  auto dw = at::empty({H}, xr.options().dtype(at::kFloat));
+  const int rsplit = tok_rmsnorm_dw_rsplit(nrows, (int)H);
+  auto ws = at::empty({(long)rsplit * H}, xr.options().dtype(at::kFloat));
+  TOK_HIP_OK(tok_rmsnorm_res_bwd(xr.data_ptr(), w.data_ptr(), dy.data_ptr(),
+                                 dxr_p, invr.data_ptr<float>(), dx.data_ptr(),
+                                 dw.data_ptr<float>(), ws.data_ptr<float>(),
+                                 nrows, (int)H, current_stream()));
   return {dx, dw};
 }
 
@@ -404,6 +460,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ce_bwd", &ce_bwd, "Fused cross-entropy backward (bf16, gfx950)");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16, gfx950)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (bf16, gfx950)");
+  mod.def("rmsnorm_res_fwd", &rmsnorm_res_fwd,
+          "Fused residual-add + RMSNorm forward (bf16, gfx950)",
+          py::arg("x"), py::arg("res"), py::arg("w"), py::arg("eps"));
+  mod.def("rmsnorm_res_bwd", &rmsnorm_res_bwd,
+          "Fused residual-add + RMSNorm backward (bf16, gfx950)",
+          py::arg("xr"), py::arg("w"), py::arg("dy"), py::arg("dxr"),
+          py::arg("invr"));
   mod.def("swiglu_fwd", &swiglu_fwd,
           "Fused SwiGLU over packed gate_up (bf16, gfx950)");
   mod.def("swiglu_bwd", &swiglu_bwd,

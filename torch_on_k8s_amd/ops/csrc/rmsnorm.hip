@@ -128,9 +128,132 @@ __global__ void colsum_kernel(const float* __restrict__ ws,
   }
 }
 
+// Fused residual-add + RMSNorm (forward): xr = x + res; y = norm(xr)*w.
+// One pass instead of a separate add kernel + norm read (the residual
+// stream is [B,S,4096] bf16 = 256 MB per pass at the flagship shape).
+// res == nullptr degenerates to plain RMSNorm that also emits xr = x.
+__global__ void rmsnorm_res_fwd_kernel(const bf16x8* __restrict__ x,
+                                       const bf16x8* __restrict__ res,
+                                       const bf16x8* __restrict__ w,
+                                       bf16x8* __restrict__ xr,
+                                       bf16x8* __restrict__ y,
+                                       float* __restrict__ invr,
+                                       long nrows, int hc, float eps) {
+  __shared__ float red[BLOCK / WAVE];
+  const int H = hc * 8;
+  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const bf16x8* xrow = x + row * hc;
+    const bf16x8* rrow = res ? res + row * hc : nullptr;
+    bf16x8* xrout = xr + row * hc;
+    float ss = 0.f;
+    for (int c = threadIdx.x; c < hc; c += BLOCK) {
+      bf16x8 v = xrow[c], o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bfbits2f(v.h[j]);
+        if (rrow) f += bfbits2f(rrow[c].h[j]);
+        // residual sum is kept in bf16 (same numerics as the unfused
+        // x + res add the model used before)
+        o.h[j] = f2bfbits(f);
+        float fq = bfbits2f(o.h[j]);
+        ss = fmaf(fq, fq, ss);
+      }
+      xrout[c] = o;
+    }
+    ss = block_reduce_sum(ss, red);
+    const float r = rsqrtf(ss / (float)H + eps);
+    if (threadIdx.x == 0) invr[row] = r;
+    bf16x8* yr = y + row * hc;
+    for (int c = threadIdx.x; c < hc; c += BLOCK) {
+      bf16x8 xv = xrout[c], wv = w[c], ov;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bfbits2f(xv.h[j]) * r * bfbits2f(wv.h[j]);
+        ov.h[j] = f2bfbits(f);
+      }
+      yr[c] = ov;
+    }
+    __syncthreads();
+  }
+}
+
+// Backward of the fused op: dx = rmsnorm_dx(xr, dy) + dxr (dxr = grad
+// flowing into the xr output from downstream residual uses; nullable).
+__global__ void rmsnorm_res_bwd_dx_kernel(const bf16x8* __restrict__ xr,
+                                          const bf16x8* __restrict__ w,
+                                          const bf16x8* __restrict__ dy,
+                                          const bf16x8* __restrict__ dxr,
+                                          const float* __restrict__ invr,
+                                          bf16x8* __restrict__ dx,
+                                          long nrows, int hc) {
+  __shared__ float red[BLOCK / WAVE];
+  const int H = hc * 8;
+  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const bf16x8* xrow = xr + row * hc;
+    const bf16x8* dyr = dy + row * hc;
+    const float r = invr[row];
+    float acc = 0.f;
+    for (int c = threadIdx.x; c < hc; c += BLOCK) {
+      bf16x8 xv = xrow[c], wv = w[c], dv = dyr[c];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        acc = fmaf(bfbits2f(dv.h[j]) * bfbits2f(wv.h[j]), bfbits2f(xv.h[j]), acc);
+    }
+    acc = block_reduce_sum(acc, red);
+    const float k = acc * r * r / (float)H;
+    bf16x8* dxrow = dx + row * hc;
+    const bf16x8* addrow = dxr ? dxr + row * hc : nullptr;
+    for (int c = threadIdx.x; c < hc; c += BLOCK) {
+      bf16x8 xv = xrow[c], wv = w[c], dv = dyr[c], ov;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = r * (bfbits2f(wv.h[j]) * bfbits2f(dv.h[j]) - bfbits2f(xv.h[j]) * k);
+        if (addrow) g += bfbits2f(addrow[c].h[j]);
+        ov.h[j] = f2bfbits(g);
+      }
+      dxrow[c] = ov;
+    }
+    __syncthreads();
+  }
+}
+
 }  // namespace
 
 extern "C" {
+
+int tok_rmsnorm_dw_rsplit(long nrows, int H);
+
+hipError_t tok_rmsnorm_res_fwd(const void* x, const void* res, const void* w,
+                               void* xr, void* y, float* invr, long nrows,
+                               int H, float eps, hipStream_t stream) {
+  const int hc = H / 8;
+  int grid = (int)(nrows < 8192 ? nrows : 8192);
+  if (grid < 1) grid = 1;
+  rmsnorm_res_fwd_kernel<<<grid, BLOCK, 0, stream>>>(
+      (const bf16x8*)x, (const bf16x8*)res, (const bf16x8*)w, (bf16x8*)xr,
+      (bf16x8*)y, invr, nrows, hc, eps);
+  return hipGetLastError();
+}
+
+hipError_t tok_rmsnorm_res_bwd(const void* xr, const void* w, const void* dy,
+                               const void* dxr_in, const float* invr,
+                               void* dx, float* dw_f32, float* dw_ws,
+                               long nrows, int H, hipStream_t stream) {
+  const int hc = H / 8;
+  int grid = (int)(nrows < 8192 ? nrows : 8192);
+  if (grid < 1) grid = 1;
+  rmsnorm_res_bwd_dx_kernel<<<grid, BLOCK, 0, stream>>>(
+      (const bf16x8*)xr, (const bf16x8*)w, (const bf16x8*)dy,
+      (const bf16x8*)dxr_in, invr, (bf16x8*)dx, nrows, hc);
+  int cblocks = (hc + BLOCK - 1) / BLOCK;
+  const int rsplit = tok_rmsnorm_dw_rsplit(nrows, H);
+  dim3 g(cblocks, rsplit);
+  rmsnorm_bwd_dw_kernel<<<g, BLOCK, 0, stream>>>(
+      (const bf16x8*)xr, (const bf16x8*)dy, invr, dw_ws, nrows, hc);
+  int g2 = (int)((H + BLOCK - 1) / BLOCK);
+  colsum_kernel<<<g2, BLOCK, 0, stream>>>(dw_ws, dw_f32, rsplit, (long)H);
+  return hipGetLastError();
+}
 
 hipError_t tok_rmsnorm_fwd(const void* x, const void* w, void* y, float* invr,
                            long nrows, int H, float eps, hipStream_t stream) {
