@@ -320,10 +320,17 @@ def test_rmsnorm_residual_fused():
     dx_ref = r * (w * dy - xr_q * (r * r / H) * c) + dxr_in
     dw_ref = (dy * xr_q * r).sum(0)
 
-    assert (y.float().cpu() - y_ref).abs().max().item() < 3e-2
-    assert (xr.float().cpu() - xr_q).abs().max().item() < 2e-2
-    assert (xg.grad.float().cpu() - dx_ref).abs().max().item() < 4e-2
-    assert (rg.grad.float().cpu() - dx_ref).abs().max().item() < 4e-2
+    def tol(ref_t):  # bf16 output rounding scales with magnitude
+        return max(3e-2, 2.5 / 256 * ref_t.abs().max().item())
+
+    err_y = (y.float().cpu() - y_ref).abs().max().item()
+    err_xr = (xr.float().cpu() - xr_q).abs().max().item()
+    err_dx = (xg.grad.float().cpu() - dx_ref).abs().max().item()
+    err_dr = (rg.grad.float().cpu() - dx_ref).abs().max().item()
+    assert err_y < tol(y_ref), f"y err {err_y}"
+    assert err_xr < 2e-2, f"xr err {err_xr}"
+    assert err_dx < tol(dx_ref), f"dx err {err_dx}"
+    assert err_dr < tol(dx_ref), f"dres err {err_dr}"
     rel = ((wg.grad.float().cpu() - dw_ref).abs() /
            (dw_ref.abs() + 1.0)).max().item()
     assert rel < 3e-2, f"dw rel err {rel}"

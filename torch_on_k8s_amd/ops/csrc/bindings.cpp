@@ -119,7 +119,7 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor dy,
   const long H = x.size(-1);
   const long nrows = x.numel() / H;
   auto dx = at::empty_like(x);
-  auto dw = at::empty({H}, x.options().dtype(at::kFloat));
+  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
   const int rsplit = tok_rmsnorm_dw_rsplit(nrows, (int)H);
   auto ws = at::empty({(long)rsplit * H}, x.options().dtype(at::kFloat));
   TOK_HIP_OK(tok_rmsnorm_bwd(x.data_ptr(), w.data_ptr(), dy.data_ptr(),
@@ -168,7 +168,7 @@ std::vector<at::Tensor> rmsnorm_res_bwd(at::Tensor xr, at::Tensor w,
     dxr_p = dxr->data_ptr();
   }
   auto dx = at::empty_like(xr);
-  auto dw = at::empty({H}, xr.options().dtype(at::kFloat));
+  auto dw = at::zeros({H}, xr.options().dtype(at::kFloat));
   const int rsplit = tok_rmsnorm_dw_rsplit(nrows, (int)H);
   auto ws = at::empty({(long)rsplit * H}, xr.options().dtype(at::kFloat));
   TOK_HIP_OK(tok_rmsnorm_res_bwd(xr.data_ptr(), w.data_ptr(), dy.data_ptr(),
