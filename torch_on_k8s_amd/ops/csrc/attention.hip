@@ -572,15 +572,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
   }
 }
 
-// DB=true: double-buffered (Q rm | QT | dO rm) tiles — 6*16K + 64K =
-// 160 KiB, the whole CU's LDS for the one resident block. The VGPR
-// budget for the in-flight stage objects comes from K no longer living
-// in registers: its fragments are re-loaded per use from the block's
-// L1/L2-resident 64 KB K slice (high reuse, loads issued ahead of the
-// MFMA chain). DB=false: synchronous staging (r1 structure) but with
-// the same K-per-use change (frees ~32 VGPRs -> no scratch spill).
-// Both instantiated; the launcher A/Bs via TOK_DK_DB.
-template <int D, bool CAUSAL, bool DB>
+template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
@@ -591,10 +583,13 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
   constexpr int DT = D / 32;
   constexpr int BNK = NW * 32;
   constexpr int KB = BN * D * 2;
-  constexpr int NBUF = DB ? 2 : 1;
-  __shared__ __attribute__((aligned(16))) char smem[NBUF * 3 * KB +
-                                                    BNK * D * 2];
-  char* v_lds = smem + NBUF * 3 * KB;
+  // Q rm | QT | dO rm | V block tile (staged once; keeping V in
+  // registers alongside K pushed the kernel to 256 VGPR + scratch spill)
+  __shared__ __attribute__((aligned(16))) char smem[3 * KB + BNK * D * 2];
+  char* q_lds = smem;
+  char* qt_lds = smem + KB;
+  char* do_lds = smem + 2 * KB;
+  char* v_lds = smem + 3 * KB;
 
   const int lane = threadIdx.x & 63;
   const int hi = lane >> 5;
@@ -611,8 +606,14 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
   bf16_t* dkp = dk + ((long)b * S * kv_tok) + (long)hkv * D;
 
   const int kvrow = n0w + (lane & 31);
-  const bool kv_ok = kvrow < S;
-  const bf16_t* kprow = kp + (long)kvrow * kv_tok + hi * 8;
+  bf16x8v k_reg[QC];
+#pragma unroll
+  for (int c = 0; c < QC; ++c) {
+    uint4 kr = {0, 0, 0, 0};
+    if (kvrow < S)
+      kr = *(const uint4*)(kp + (long)kvrow * kv_tok + c * 16 + hi * 8);
+    k_reg[c] = as_frag(kr);
+  }
   {  // stage the block's 256-row V tile (row-major, swizzled) once
     constexpr int VPR = D / 8;
 #pragma unroll 4
@@ -632,23 +633,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
     for (int r = 0; r < 16; ++r) dk_acc[t][r] = 0.f;
 
   const int m_start = CAUSAL ? (n0 / BN) * BN : 0;
-  const long dplane = (long)D * S_pad;
-
-  TileStage<D> q_st, do_st;
-  TileStageT<D> qt_st;
-  if constexpr (DB) {  // prologue: first tile of the first grouped head
-    const bf16_t* qp0 = q + ((long)b * S * q_tok) + (long)(hkv * rep) * D;
-    const bf16_t* dop0 = dout + ((long)b * S * q_tok) + (long)(hkv * rep) * D;
-    const bf16_t* qtp0 = q_t + ((long)b * Hq + hkv * rep) * dplane;
-    q_st.issue(qp0, m_start, S, q_tok);
-    do_st.issue(dop0, m_start, S, q_tok);
-    qt_st.issue(qtp0, m_start, S_pad);
-    q_st.write_rm(smem);
-    qt_st.write(smem + KB);
-    do_st.write_rm(smem + 2 * KB);
-    __syncthreads();
-  }
-  int cur = 0;
 
   for (int g = 0; g < rep; ++g) {
     const int hq = hkv * rep + g;
@@ -657,32 +641,20 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
     const float* lsep = lse + ((long)b * Hq + hq) * S;
     const float* dsp = dsum + ((long)b * Hq + hq) * S;
 
-    const bf16_t* qtp = q_t + ((long)b * Hq + hq) * dplane;
+    const bf16_t* qtp = q_t + ((long)b * Hq + hq) * (long)D * S_pad;
     for (int m0 = m_start; m0 < S; m0 += BN) {
-      char* q_lds = smem + cur * (3 * KB);
-      char* qt_lds = q_lds + KB;
-      char* do_lds = q_lds + 2 * KB;
-      const int m1 = m0 + BN;
-      const bool more = m1 < S || g + 1 < rep;
-      if constexpr (DB) {
-        if (m1 < S) {
-          q_st.issue(qp, m1, S, q_tok);
-          do_st.issue(dop, m1, S, q_tok);
-          qt_st.issue(qtp, m1, S_pad);
-        } else if (g + 1 < rep) {
-          q_st.issue(qp + D, m_start, S, q_tok);
-          do_st.issue(dop + D, m_start, S, q_tok);
-          qt_st.issue(qtp + dplane, m_start, S_pad);
-        }
-      } else {  // synchronous staging into the single buffer
+      {  // synchronous staging: the prefetch ring costs ~16 VGPRs and
+         // tips this kernel into scratch spill, which is worse
+        TileStage<D> q_st, do_st;
+        TileStageT<D> qt_st;
         q_st.issue(qp, m0, S, q_tok);
         do_st.issue(dop, m0, S, q_tok);
         qt_st.issue(qtp, m0, S_pad);
         q_st.write_rm(q_lds);
         qt_st.write(qt_lds);
         do_st.write_rm(do_lds);
-        __syncthreads();
       }
+      __syncthreads();
 
 #pragma unroll
       for (int qs = 0; qs < 2; ++qs) {
@@ -694,12 +666,10 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
         }
 #pragma unroll
         for (int c = 0; c < QC; ++c) {
-          uint4 kr = {0, 0, 0, 0};
-          if (kv_ok) kr = *(const uint4*)(kprow + c * 16);
           bf16x8v qa = read_bfrag<D * 2>(
               q_lds, qs * 32 + (lane & 31), c * 16 + hi * 8);
           sv = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              qa, as_frag(kr), sv, 0, 0, 0);
+              qa, k_reg[c], sv, 0, 0, 0);
           bf16x8v doa = read_bfrag<D * 2>(
               do_lds, qs * 32 + (lane & 31), c * 16 + hi * 8);
           bf16x8v vb = read_bfrag<D * 2>(
@@ -737,18 +707,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
               da1, b1, dk_acc[t], 0, 0, 0);
         }
       }
-      if constexpr (DB) {
-        if (more) {
-          char* nb = smem + (cur ^ 1) * (3 * KB);
-          q_st.write_rm(nb);
-          qt_st.write(nb + KB);
-          do_st.write_rm(nb + 2 * KB);
-        }
-        __syncthreads();
-        cur ^= 1;
-      } else {
-        __syncthreads();
-      }
+      __syncthreads();
     }
   }
 
@@ -975,28 +934,16 @@ hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
 
   dim3 gkv((S + NW * 32 - 1) / (NW * 32), Hkv, B);
   dim3 gq((S + NW * 32 - 1) / (NW * 32), Hq, B);
-  // dk staging variant: double-buffered by default; TOK_DK_DB=0 selects
-  // the synchronous-staging instantiation (A/B hook)
-  static const bool dk_db = [] {
-    const char* e = getenv("TOK_DK_DB");
-    return !(e && e[0] == '0');
-  }();
 #define LAUNCH_BWD(DD, CC)                                                    \
   do {                                                                        \
     attn_bwd_dv_kernel<DD, CC><<<gkv, NTHREADS, 0, stream>>>(                 \
         (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)dot_t,             \
         (const bf16_t*)dout, lse, dsum_ws, (bf16_t*)dv, B, S, Hq, Hkv,       \
         S_pad, scale);                                                        \
-    if (dk_db)                                                                \
-      attn_bwd_dk_kernel<DD, CC, true><<<gkv, NTHREADS, 0, stream>>>(         \
-          (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,               \
-          (const bf16_t*)dout, (const bf16_t*)q_t, lse, dsum_ws,              \
-          (bf16_t*)dk, B, S, Hq, Hkv, S_pad, scale);                          \
-    else                                                                      \
-      attn_bwd_dk_kernel<DD, CC, false><<<gkv, NTHREADS, 0, stream>>>(        \
-          (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,               \
-          (const bf16_t*)dout, (const bf16_t*)q_t, lse, dsum_ws,              \
-          (bf16_t*)dk, B, S, Hq, Hkv, S_pad, scale);                          \
+    attn_bwd_dk_kernel<DD, CC><<<gkv, NTHREADS, 0, stream>>>(                 \
+        (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,                 \
+        (const bf16_t*)dout, (const bf16_t*)q_t, lse, dsum_ws, (bf16_t*)dk,  \
+        B, S, Hq, Hkv, S_pad, scale);                                         \
     attn_bwd_dq_kernel<DD, CC><<<gq, NTHREADS, 0, stream>>>(                  \
         (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,                 \
         (const bf16_t*)dout, (const bf16_t*)k_t, lse, dsum_ws, (bf16_t*)dq,  \
