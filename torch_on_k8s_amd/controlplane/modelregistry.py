@@ -7,18 +7,23 @@ a Kaniko pod bakes `FROM busybox; COPY build/ /torch-on-k8s-model` and
 pushes an image, and Model.LatestVersion advances.
 
 Node-native redesign: there is no registry daemon on the box, so an
-"image" is an OCI-layout-shaped local artifact: the checkpoint directory
-is packed into a content-addressed tar.gz layer with a manifest.json
-recording the model path (TORCH_ON_K8S_MODEL_PATH parity,
-model/v1alpha1/constants.go:24-27), and Model.latest_version advances.
-Storage providers mirror the reference's LocalStorage/NFS split
-(pkg/storage/): both are directory roots with provenance metadata.
+"image" is a REAL OCI image layout on disk (opencontainers image-spec:
+oci-layout + index.json + blobs/sha256/{layer,config,manifest}) that
+`podman load` / `skopeo copy oci:<dir>` accept as-is — pushing to a
+registry is one skopeo invocation away, which is the Kaniko-push parity
+point (modelversion_controller.go:286-406). The config blob carries
+TORCH_ON_K8S_MODEL_PATH (model/v1alpha1/constants.go:24-27); build
+lifecycle is Building -> Succeeded/Failed and Model.latest_version only
+advances on success. Storage providers mirror the reference's
+LocalStorage/NFS split (pkg/storage/).
 """
 from __future__ import annotations
 
+import gzip
 import hashlib
 import json
 import os
+import shutil
 import tarfile
 import time
 from dataclasses import dataclass, field
@@ -88,41 +93,112 @@ class ModelRegistry:
         return self.build_version(job.model_name, version, src,
                                   source_job=job.name)
 
+    @staticmethod
+    def _sha256_file(path: str) -> str:
+        h = hashlib.sha256()
+        with open(path, "rb") as f:
+            for chunk in iter(lambda: f.read(1 << 20), b""):
+                h.update(chunk)
+        return h.hexdigest()
+
     def build_version(self, model_name: str, version: str, src_dir: str,
                       source_job: str | None = None) -> ModelVersion:
-        """The Kaniko-pod analog: pack src_dir into a layer tar.gz +
-        manifest; content digest = sha256 of the layer."""
+        """The Kaniko-pod analog: bake src_dir into an OCI image layout
+        (`FROM scratch; COPY src /torch-on-k8s-model; ENV
+        TORCH_ON_K8S_MODEL_PATH=...`), Building -> Succeeded/Failed."""
         model = self.ensure_model(model_name)
-        out = os.path.join(self.storage.artifact_dir(), model_name)
-        os.makedirs(out, exist_ok=True)
-        layer = os.path.join(out, f"{version}.tar.gz")
-        mv = ModelVersion(model=model_name, version=version, image_ref=layer,
-                          source_job=source_job, build_phase="Building")
+        image_dir = os.path.join(self.storage.artifact_dir(), model_name,
+                                 version)
+        mv = ModelVersion(model=model_name, version=version,
+                          image_ref=image_dir, source_job=source_job,
+                          build_phase="Building")
         model.versions[version] = mv
         try:
-            with tarfile.open(layer, "w:gz") as tf:
+            if not os.path.isdir(src_dir):
+                raise OSError(f"source dir missing: {src_dir}")
+            blobs = os.path.join(image_dir, "blobs", "sha256")
+            os.makedirs(blobs, exist_ok=True)
+
+            # layer: tar of src_dir at MODEL_IMAGE_PATH; OCI config needs
+            # the UNCOMPRESSED digest (diff_id), the manifest the
+            # compressed one
+            tmp_tar = os.path.join(image_dir, ".layer.tar")
+            with tarfile.open(tmp_tar, "w") as tf:
                 tf.add(src_dir, arcname=MODEL_IMAGE_PATH.lstrip("/"))
-            h = hashlib.sha256()
-            with open(layer, "rb") as f:
-                for chunk in iter(lambda: f.read(1 << 20), b""):
-                    h.update(chunk)
-            mv.digest = "sha256:" + h.hexdigest()
-            manifest = {
+            diff_id = self._sha256_file(tmp_tar)
+            tmp_gz = tmp_tar + ".gz"
+            with open(tmp_tar, "rb") as fin, \
+                    gzip.GzipFile(filename="", fileobj=open(tmp_gz, "wb"),
+                                  mtime=0) as fout:
+                shutil.copyfileobj(fin, fout)
+            os.unlink(tmp_tar)
+            layer_digest = self._sha256_file(tmp_gz)
+            layer_size = os.path.getsize(tmp_gz)
+            os.replace(tmp_gz, os.path.join(blobs, layer_digest))
+
+            def put_blob(obj) -> tuple[str, int]:
+                data = json.dumps(obj).encode()
+                d = hashlib.sha256(data).hexdigest()
+                with open(os.path.join(blobs, d), "wb") as f:
+                    f.write(data)
+                return d, len(data)
+
+            config_digest, config_size = put_blob({
+                "created": time.strftime(
+                    "%Y-%m-%dT%H:%M:%SZ", time.gmtime(mv.ts)),
+                "architecture": "amd64",
+                "os": "linux",
+                "config": {
+                    "Env": [f"{MODEL_PATH_ENV}={MODEL_IMAGE_PATH}"],
+                    "Labels": {
+                        "io.torch-on-k8s-amd.model": model_name,
+                        "io.torch-on-k8s-amd.version": version,
+                        "io.torch-on-k8s-amd.source-job": source_job or "",
+                    },
+                },
+                "rootfs": {"type": "layers",
+                           "diff_ids": [f"sha256:{diff_id}"]},
+                "history": [{"created_by":
+                             f"COPY {os.path.basename(src_dir)} "
+                             f"{MODEL_IMAGE_PATH}"}],
+            })
+            manifest_digest, manifest_size = put_blob({
                 "schemaVersion": 2,
-                "model": model_name,
-                "version": version,
-                "sourceJob": source_job,
-                "layers": [{"path": os.path.basename(layer),
-                            "digest": mv.digest}],
-                "modelPath": MODEL_IMAGE_PATH,
-                "created": mv.ts,
-            }
-            with open(os.path.join(out, f"{version}.manifest.json"), "w") as f:
-                json.dump(manifest, f, indent=2)
+                "mediaType": "application/vnd.oci.image.manifest.v1+json",
+                "config": {
+                    "mediaType": "application/vnd.oci.image.config.v1+json",
+                    "digest": f"sha256:{config_digest}",
+                    "size": config_size,
+                },
+                "layers": [{
+                    "mediaType":
+                        "application/vnd.oci.image.layer.v1.tar+gzip",
+                    "digest": f"sha256:{layer_digest}",
+                    "size": layer_size,
+                }],
+            })
+            with open(os.path.join(image_dir, "oci-layout"), "w") as f:
+                json.dump({"imageLayoutVersion": "1.0.0"}, f)
+            with open(os.path.join(image_dir, "index.json"), "w") as f:
+                json.dump({
+                    "schemaVersion": 2,
+                    "manifests": [{
+                        "mediaType":
+                            "application/vnd.oci.image.manifest.v1+json",
+                        "digest": f"sha256:{manifest_digest}",
+                        "size": manifest_size,
+                        "annotations": {
+                            "org.opencontainers.image.ref.name":
+                                f"{model_name}:{version}",
+                        },
+                    }],
+                }, f)
+            mv.digest = "sha256:" + manifest_digest
             mv.build_phase = "Succeeded"
             model.latest_version = version   # Model.LatestVersion advance
-        except OSError:
+        except (OSError, tarfile.TarError):
             mv.build_phase = "Failed"
+            shutil.rmtree(image_dir, ignore_errors=True)
         return mv
 
     def get_version(self, model_name: str, version: str) -> ModelVersion | None:
@@ -130,11 +206,20 @@ class ModelRegistry:
         return m.versions.get(version) if m else None
 
     def extract(self, model_name: str, version: str, dest: str) -> str:
-        """'docker run' analog: unpack the artifact for serving/resume."""
+        """'podman run' analog: walk index -> manifest -> layers and
+        unpack the rootfs for serving/resume."""
         mv = self.get_version(model_name, version)
         if mv is None or mv.build_phase != "Succeeded":
             raise KeyError(f"no built version {model_name}:{version}")
+        blobs = os.path.join(mv.image_ref, "blobs", "sha256")
+        with open(os.path.join(mv.image_ref, "index.json")) as f:
+            index = json.load(f)
+        mdigest = index["manifests"][0]["digest"].split(":", 1)[1]
+        with open(os.path.join(blobs, mdigest)) as f:
+            manifest = json.load(f)
         os.makedirs(dest, exist_ok=True)
-        with tarfile.open(mv.image_ref, "r:gz") as tf:
-            tf.extractall(dest)
+        for layer in manifest["layers"]:
+            ldigest = layer["digest"].split(":", 1)[1]
+            with tarfile.open(os.path.join(blobs, ldigest), "r:gz") as tf:
+                tf.extractall(dest)
         return os.path.join(dest, MODEL_IMAGE_PATH.lstrip("/"))
