@@ -105,3 +105,72 @@ def _run_scenario(ctl, rt, job, tmp_path, steps):
     # the restarted gang resumed from the checkpoint, not step 0
     agent = json.load(open(tmp_path / "work" / "elastic-e2e" / "agent.json"))
     assert agent["ckpt-completed-version"]["version"] == job.generation
+
+
+@pytest.mark.timeout(420)
+def test_fast_rejoin_keeps_processes_resident(tmp_path):
+    """Scale event with fast-rejoin: the surviving master/worker keep
+    their PIDs (model/optimizer state resident; only the process group
+    re-inits) and training resumes at the new world without a
+    checkpoint reload. Measures the scale-event downtime (r1 VERDICT
+    next-#9)."""
+    node = NodeState(num_gpus=0)
+    rt = LocalProcessRuntime(str(tmp_path / "work"))
+    ctl = JobController(node, rt,
+                        ControllerConfig(enable_gang_scheduling=False),
+                        elastic=ElasticScaler())
+    steps = 200  # long enough to still be running after the scale
+    job = TorchJob(
+        name="rejoin-e2e",
+        tasks={
+            TaskType.MASTER: TaskSpec(replicas=1, gpus_per_task=0,
+                                      env=task_env(steps)),
+            TaskType.WORKER: TaskSpec(replicas=1, gpus_per_task=0,
+                                      env=task_env(steps)),
+        },
+        elastic=ElasticPolicy(min_replicas=1, max_replicas=2),
+    )
+    ctl.create_job(job)
+    mpath = tmp_path / "work" / "rejoin-e2e" / "metrics.json"
+    try:
+        t0 = time.time()
+        while time.time() - t0 < 120:
+            ctl.reconcile(job)
+            if mpath.exists() and json.load(open(mpath))["step"] >= 3:
+                break
+            time.sleep(0.2)
+        assert mpath.exists(), "training never started"
+        hs = ctl.handles["rejoin-e2e"]
+        master_pid = hs[("rejoin-e2e", TaskType.MASTER, 0)].proc.pid
+        worker_pid = hs[("rejoin-e2e", TaskType.WORKER, 0)].proc.pid
+
+        t_scale = time.time()
+        ElasticScaler.scale(job, 2)
+        # wait until training runs at world 3
+        downtime = None
+        t0 = time.time()
+        while time.time() - t0 < 240:
+            ctl.reconcile(job)
+            if mpath.exists():
+                rec = json.load(open(mpath))
+                if rec.get("world_size") == 3:
+                    downtime = time.time() - t_scale
+                    break
+            time.sleep(0.2)
+        assert downtime is not None, \
+            (job.status.phase, [(e.reason, e.message) for e in ctl.events])
+        hs = ctl.handles["rejoin-e2e"]
+        # survivors kept their processes
+        assert hs[("rejoin-e2e", TaskType.MASTER, 0)].proc.pid == master_pid
+        assert hs[("rejoin-e2e", TaskType.WORKER, 0)].proc.pid == worker_pid
+        # the new worker is a different process
+        assert ("rejoin-e2e", TaskType.WORKER, 1) in hs
+        # survivors did NOT reload from checkpoint (no "resumed" in log)
+        mlog = (tmp_path / "work" / "rejoin-e2e" /
+                "rejoin-e2e-master-0.log").read_text()
+        assert "fast-rejoin: world=3" in mlog
+        assert "resumed at step" not in mlog
+        print(f"scale-event downtime (trigger -> first step at new world): "
+              f"{downtime:.1f}s")
+    finally:
+        ctl.delete_job(job.name)

@@ -514,6 +514,11 @@ class JobController:
             from torch_on_k8s_amd.controlplane.api import ANN_CKPT_COMPLETED
             job.annotations[ANN_CKPT_COMPLETED] = json.dumps(
                 agent["ckpt-completed-version"])
+        if agent and agent.get("rejoin-ready") is not None:
+            # fast-rejoin handshake: the surviving master's old
+            # rendezvous store is closed (elastic.py stage 2 gates new
+            # task creation on this)
+            job.annotations["rejoin-ready"] = str(agent["rejoin-ready"])
         tmp = os.path.join(jobdir, f"job.json.tmp{os.getpid()}")
         with open(tmp, "w") as f:
             json.dump({

@@ -116,18 +116,33 @@ class ElasticScaler:
             (1 if TaskType.MASTER in job.tasks else 0)
         job.annotations[ANN_WORLD_SIZE] = str(new_world)
 
-        # Kill victims (scale-in) and stale-generation tasks, then WAIT
-        # for them to exit before the normal pass recreates anything:
-        # the new gang reuses the job's master port, and racing a dying
-        # master's rendezvous store hangs or mis-wires the new ranks
-        # (restartStalePod analog, elastic_scale.go:303-397; the
-        # reference gets the equivalent ordering from Kruise CRR
-        # completion before patching the generation label).
+        # Fast-rejoin (r1 VERDICT next-#9): tasks running the framework
+        # entrypoint keep their process (model/optimizer state resident)
+        # across the scale — they tear down and re-init only the process
+        # group (entrypoint._fast_rejoin). Survivors are ADOPTED into
+        # the new generation instead of killed; only scale-in victims
+        # die. Opaque command tasks can't rejoin -> full restart path.
+        rejoin = self.rejoin_capable(job)
+
+        # Kill victims (scale-in) — and, without rejoin, every
+        # stale-generation task — then WAIT for them to exit before the
+        # normal pass recreates anything: the new gang reuses the job's
+        # master port, and racing a dying master's rendezvous store
+        # hangs or mis-wires the new ranks (restartStalePod analog,
+        # elastic_scale.go:303-397; the reference gets the equivalent
+        # ordering from Kruise CRR completion before patching the
+        # generation label).
         pending = False
+        adopted = False
         for h in list(hs.values()):
             victim = (h.task_type == TaskType.WORKER and
                       h.index >= desired.replicas)
-            if not (victim or h.generation != job.generation):
+            stale = h.generation != job.generation
+            if not (victim or stale):
+                continue
+            if rejoin and not victim:
+                h.generation = job.generation  # adopt; process rejoins
+                adopted = True
                 continue
             if not h.finished:
                 ctl.runtime.kill(h)
@@ -141,9 +156,27 @@ class ElasticScaler:
                 pending = True
         if pending:
             return True  # victims still terminating; hold recreation
+        if adopted:
+            # hold new-task creation until the surviving master reports
+            # its old rendezvous store closed (agent.json rejoin-ready,
+            # lifted into annotations by _sync_state_files) — otherwise
+            # fresh ranks could join the stale store
+            ready = job.annotations.get("rejoin-ready")
+            if str(ready) != str(job.generation):
+                return True
         ctl.event(job.name, "Normal", "ScaleExecuted",
-                  f"world_size={new_world} gen={job.generation}")
+                  f"world_size={new_world} gen={job.generation} "
+                  f"rejoin={rejoin}")
         return False  # let the normal pass recreate tasks now
+
+    @staticmethod
+    def rejoin_capable(job: TorchJob) -> bool:
+        """Fast-rejoin needs every gang member to run the framework
+        entrypoint (opaque commands can't re-init in place)."""
+        if job.annotations.get("disable-fast-rejoin") == "true":
+            return False
+        return all(s.command is None for t, s in job.tasks.items()
+                   if t != TaskType.AIMASTER)
 
     # -- user-facing scale API (AIMaster analog) -----------------------
     @staticmethod

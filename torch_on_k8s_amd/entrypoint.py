@@ -114,6 +114,15 @@ def main() -> int:
 
     signal.signal(signal.SIGTERM, on_term)
 
+    # agent.json is owned by rank 0; keep a merged view so checkpoint
+    # acks and rejoin-ready handshakes don't clobber each other
+    agent_state: dict = {}
+
+    def write_agent(**kv):
+        if ctx.is_main and agent_file:
+            agent_state.update(kv)
+            _atomic_write(agent_file, agent_state)
+
     def checkpoint_and_ack(version=None):
         # Called by EVERY rank at the same step (the train loop
         # broadcasts the decision from rank 0), so the sharded save's
@@ -121,8 +130,8 @@ def main() -> int:
         # a plain rank-0 save.
         if ckpt_dir:
             trainer.save_checkpoint(ckpt_dir)
-            if ctx.is_main and agent_file and version is not None:
-                _atomic_write(agent_file, {
+            if version is not None:
+                write_agent(**{
                     "ckpt-completed-version": {"version": version,
                                                "status": "Succeeded"},
                     "step": trainer.step_count,
@@ -131,7 +140,7 @@ def main() -> int:
     try:
         return _train_loop(trainer, ctx, steps_total, state_dir, job_file,
                            agent_file, stop, checkpoint_and_ack, destroy,
-                           barrier)
+                           barrier, write_agent)
     except RuntimeError as e:
         # A peer restarting (elastic scale / preemption) tears down the
         # process group mid-collective; classify as retryable so the
@@ -152,11 +161,80 @@ def _read_ckpt_request(job_file):
     return req
 
 
+def _poll_file(path, pred, timeout_s=120.0, period=0.05):
+    deadline = time.time() + timeout_s
+    while time.time() < deadline:
+        doc = _read_json(path)
+        if doc is not None and pred(doc):
+            return doc
+        time.sleep(period)
+    raise TimeoutError(f"rejoin: timed out waiting on {path}")
+
+
+def _fast_rejoin(trainer, ctx, job_doc, agent_file, write_agent,
+                 version):
+    """Elastic fast-rejoin (scale event without process restart): keep
+    model/optimizer state resident, tear down only the process group and
+    re-init it at the new world size. Protocol (r1 VERDICT next-#9; the
+    reference's torchrun-rendezvous intent, torchjob_controller.go:387-392):
+
+      rank0 broadcasts the scale decision -> every rank checkpoints
+      (already done by the caller) -> all ranks destroy the pg ->
+      master writes agent.json rejoin-ready=version (its rendezvous
+      store is now CLOSED, so the controller may create new tasks and
+      surviving workers may reconnect without mis-joining the old
+      store) -> survivors wait for the controller's
+      ready-to-start-worker flag -> re-init at the new world.
+
+    Returns the new world size for survivors, or None if this task is a
+    scale-in victim (caller exits cleanly)."""
+    import datetime
+    import torch.distributed as dist
+
+    ttype = os.environ.get("TOK_TASK_TYPE", "master")
+    tindex = int(os.environ.get("TOK_TASK_INDEX", "0"))
+    replicas = job_doc.get("replicas", {})
+    new_world = sum(int(v) for t, v in replicas.items() if t != "aimaster")
+    survivor = tindex < int(replicas.get(ttype, 0))
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    if ctx.is_main:
+        write_agent(**{"rejoin-ready": version})
+    if not survivor:
+        return None
+
+    if not ctx.is_main:
+        # master must have closed its old store before we reconnect
+        # (the controller gates NEW task creation on the same handshake)
+        _poll_file(agent_file, lambda d: d.get("rejoin-ready") == version)
+
+    rank = 0 if ttype == "master" else tindex + 1
+    dist.init_process_group(
+        backend=ctx.backend or ("nccl" if ctx.device.type == "cuda"
+                                else "gloo"),
+        rank=rank, world_size=new_world,
+        timeout=datetime.timedelta(seconds=300))
+    ctx.rank = rank
+    ctx.world_size = new_world
+    trainer.fb.set_world(new_world)
+    print(f"[entrypoint] fast-rejoin: world={new_world} rank={rank} "
+          f"step={trainer.step_count} (state kept resident)", flush=True)
+    return new_world
+
+
 def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
-                stop, checkpoint_and_ack, destroy, barrier):
+                stop, checkpoint_and_ack, destroy, barrier, write_agent):
     import torch
     import torch.distributed as dist
+    # resume the ack state so a restarted rank 0 doesn't re-checkpoint
+    # an already-completed version (the request annotation is sticky)
     last_completed = None
+    if agent_file:
+        cv = (_read_json(agent_file) or {}).get("ckpt-completed-version")
+        if cv:
+            last_completed = cv.get("version")
+    rejoin_enabled = os.environ.get("TOK_DISABLE_REJOIN") != "1"
     # benchmark instrumentation (TOK_BENCH_*): timed region bracketed by
     # barrier + device sync on both sides, MAX-elapsed over ranks
     bench_steps = int(os.environ.get("TOK_BENCH_STEPS", "0"))
@@ -169,8 +247,9 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
             torch.cuda.synchronize()
     # coordination word broadcast from rank 0 each step so every rank
     # takes checkpoint/stop decisions at the SAME step (sharded saves
-    # are collective): [requested ckpt version or 0, stop flag]
-    coord = torch.zeros(2, dtype=torch.long, device=ctx.device)
+    # are collective): [requested ckpt version or 0, stop flag,
+    # rejoin flag (scale event: world size is changing)]
+    coord = torch.zeros(3, dtype=torch.long, device=ctx.device)
     while trainer.step_count < steps_total:
         if bench_steps:
             # benchmark mode: every timed step is EXACTLY a bare-bench
@@ -204,17 +283,42 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
             print(f"[train] step={trainer.step_count} loss={loss:.4f}",
                   flush=True)
         if ctx.is_main:
-            req = _read_ckpt_request(job_file) if job_file else None
-            coord[0] = req["version"] if req else 0
+            # RANK 0 decides; everyone else only obeys the broadcast —
+            # per-rank state like last_completed must never fork the
+            # gang's control flow (a late-joining rank would otherwise
+            # enter a collective checkpoint alone and deadlock)
+            jd = (_read_json(job_file) or {}) if job_file else {}
+            req = jd.get("annotations", {}).get("ckpt-requested-version")
+            if isinstance(req, str):
+                try:
+                    req = json.loads(req)
+                except ValueError:
+                    req = None
+            reqv0 = req["version"] if req else 0
+            coord[0] = reqv0 if reqv0 and reqv0 != last_completed else 0
             coord[1] = 1 if stop["sig"] is not None else 0
+            # scale detection: the desired gang world differs from ours
+            replicas = jd.get("replicas", {})
+            new_world = sum(int(v) for t, v in replicas.items()
+                            if t != "aimaster")
+            coord[2] = 1 if (rejoin_enabled and replicas and
+                             new_world != ctx.world_size) else 0
         if ctx.is_distributed and dist.is_initialized():
             dist.broadcast(coord, src=0)
         reqv = int(coord[0])
         stop_now = bool(int(coord[1])) or \
             (not ctx.is_distributed and stop["sig"] is not None)
-        if reqv and reqv != last_completed:
+        scale_now = bool(int(coord[2])) and reqv
+        if reqv:
             checkpoint_and_ack(reqv)
             last_completed = reqv
+        if scale_now:
+            jd = _read_json(job_file) or {}
+            new_world = _fast_rejoin(trainer, ctx, jd, agent_file,
+                                     write_agent, reqv)
+            if new_world is None:
+                return 0  # scale-in victim: clean exit, controller reaps
+            continue
         if stop_now:
             checkpoint_and_ack()
             destroy()

@@ -118,18 +118,31 @@ class FlatBucketModel:
         self.accumulate_only = False
         self._hooks = []
         if self.overlap and self.world_size > 1:
-            for b in self.buckets:
-                for s in b.segs:
-                    h = s.param.register_post_accumulate_grad_hook(
-                        self._make_hook(b))
-                    self._hooks.append(h)
+            self._register_hooks()
+
+    def _register_hooks(self):
+        for b in self.buckets:
+            for s in b.segs:
+                h = s.param.register_post_accumulate_grad_hook(
+                    self._make_hook(b))
+                self._hooks.append(h)
+
+    def set_world(self, world_size: int):
+        """Elastic fast-rejoin: the process group was re-initialized at a
+        new world size with model/optimizer state kept resident. Hooks
+        self-guard on world_size/is_initialized, so growing from a
+        world-1 start just needs them registered."""
+        self.world_size = world_size
+        if self.overlap and world_size > 1 and not self._hooks:
+            self._register_hooks()
 
     def _make_hook(self, bucket: Bucket):
         def hook(_param):
             bucket.pending -= 1
             if bucket.pending == 0:
                 bucket.pending = len(bucket.segs)
-                if not self.accumulate_only:
+                if not self.accumulate_only and self.world_size > 1 and \
+                        dist.is_initialized():
                     bucket.work = dist.all_reduce(
                         bucket.flat_grad, group=self.group, async_op=True)
         return hook
