@@ -36,10 +36,11 @@ MODEL_IMAGE_PATH = "/torch-on-k8s-model"  # path baked into the artifact
 class ModelVersion:
     model: str
     version: str
-    image_ref: str          # local artifact path (registry analog)
+    image_ref: str          # OCI image layout dir (registry analog)
     digest: str = ""
     build_phase: str = "Created"   # Created|Building|Succeeded|Failed
     source_job: str | None = None
+    storage: dict = field(default_factory=dict)  # provider provenance
     ts: float = field(default_factory=time.time)
 
 
@@ -52,8 +53,9 @@ class Model:
 
 
 class StorageProvider:
-    """Reference pkg/storage interface analog: yields the model output
-    dir a job's tasks mount, and where artifacts land."""
+    """Reference pkg/storage interface analog (storage/interface.go:26-35):
+    yields the model output dir a job's tasks mount, where artifacts
+    land, and the provenance the reference records in PV specs."""
 
     def __init__(self, root: str, kind: str = "local"):
         self.root = root
@@ -69,6 +71,72 @@ class StorageProvider:
         d = os.path.join(self.root, "artifacts")
         os.makedirs(d, exist_ok=True)
         return d
+
+    def provenance(self) -> dict:
+        """What the reference bakes into the PV spec; consumers (task
+        env injection, ModelVersion records) read this to know where
+        the bytes physically live."""
+        return {"kind": self.kind, "root": self.root}
+
+    def task_env(self) -> dict:
+        """Env vars injected into tasks that mount this storage."""
+        return {"TOK_STORAGE_KIND": self.kind, "TOK_STORAGE_ROOT": self.root}
+
+
+class LocalStorageProvider(StorageProvider):
+    """hostPath PV pinned by node affinity
+    (localstorage/local_storage.go:36-109): on the single-node design
+    the affinity collapses to recording WHICH node owns the bytes so a
+    multi-node deployment can schedule readers onto it."""
+
+    def __init__(self, root: str, node_name: str | None = None):
+        super().__init__(root, kind="local")
+        import socket
+        self.node_name = node_name or socket.gethostname()
+
+    def provenance(self) -> dict:
+        return {"kind": self.kind, "root": self.root,
+                "nodeAffinity": {"kubernetes.io/hostname": self.node_name}}
+
+    def task_env(self) -> dict:
+        env = super().task_env()
+        env["TOK_STORAGE_NODE"] = self.node_name
+        return env
+
+
+class NFSStorageProvider(StorageProvider):
+    """NFS-backed PV (nfs/nfs.go:36-89): carries server + export path
+    provenance; any node may mount it (no affinity)."""
+
+    def __init__(self, root: str, server: str, path: str):
+        super().__init__(root, kind="nfs")
+        self.server = server
+        self.path = path
+
+    def provenance(self) -> dict:
+        return {"kind": self.kind, "root": self.root,
+                "server": self.server, "path": self.path}
+
+    def task_env(self) -> dict:
+        env = super().task_env()
+        env["TOK_STORAGE_NFS_SERVER"] = self.server
+        env["TOK_STORAGE_NFS_PATH"] = self.path
+        return env
+
+
+def storage_from_spec(root: str, spec: dict | None) -> StorageProvider:
+    """Registry parity (storage/registry/registry.go:36-44): pick the
+    provider by which field set is present in the storage spec."""
+    spec = spec or {}
+    if "nfs" in spec:
+        nfs = spec["nfs"]
+        return NFSStorageProvider(root, server=nfs.get("server", ""),
+                                  path=nfs.get("path", "/"))
+    if "localStorage" in spec:
+        ls = spec["localStorage"]
+        return LocalStorageProvider(ls.get("path", root),
+                                    node_name=ls.get("nodeName"))
+    return LocalStorageProvider(root)
 
 
 class ModelRegistry:
@@ -111,6 +179,7 @@ class ModelRegistry:
                                  version)
         mv = ModelVersion(model=model_name, version=version,
                           image_ref=image_dir, source_job=source_job,
+                          storage=self.storage.provenance(),
                           build_phase="Building")
         model.versions[version] = mv
         try:

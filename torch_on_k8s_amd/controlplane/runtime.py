@@ -135,9 +135,11 @@ class LocalProcessRuntime(Runtime):
     """One OS process per task ('kubelet' analog). GPU isolation via
     HIP_VISIBLE_DEVICES over the node's GPU slots."""
 
-    def __init__(self, workdir: str, python: str | None = None):
+    def __init__(self, workdir: str, python: str | None = None,
+                 storage=None):
         self.workdir = workdir
         self.python = python or sys.executable
+        self.storage = storage  # StorageProvider: provenance env for tasks
         os.makedirs(workdir, exist_ok=True)
 
     def start_task(self, job, t, index, gpu_slots, extra_env):
@@ -145,6 +147,8 @@ class LocalProcessRuntime(Runtime):
                        generation=job.generation)
         spec = job.tasks[t]
         env = dict(os.environ)
+        if self.storage is not None:
+            env.update(self.storage.task_env())
         env.update(cluster_env(job, t, index))
         env.update(spec.env)
         env.update(extra_env)

@@ -31,7 +31,7 @@ from torch_on_k8s_amd.controlplane.jobspec import (job_from_yaml,
 from torch_on_k8s_amd.controlplane.metrics import (JobMetrics,
                                                    start_metrics_server)
 from torch_on_k8s_amd.controlplane.modelregistry import (ModelRegistry,
-                                                         StorageProvider)
+                                                         storage_from_spec)
 from torch_on_k8s_amd.controlplane.node import NodeState
 from torch_on_k8s_amd.controlplane.runtime import LocalProcessRuntime
 from torch_on_k8s_amd.utils.logging import get_logger
@@ -42,7 +42,7 @@ log = get_logger("manager")
 class Manager:
     def __init__(self, workdir: str, num_gpus: int = 8, quotas=None,
                  gates: feat.FeatureGates | None = None,
-                 sync_period: float = 0.5):
+                 sync_period: float = 0.5, storage_spec: dict | None = None):
         self.workdir = workdir
         self.spool = os.path.join(workdir, "spool")
         self.status_dir = os.path.join(workdir, "status")
@@ -52,8 +52,10 @@ class Manager:
         self.sync_period = sync_period
 
         node = NodeState(num_gpus=num_gpus)
-        runtime = LocalProcessRuntime(os.path.join(workdir, "jobs"))
-        storage = StorageProvider(os.path.join(workdir, "models"))
+        storage = storage_from_spec(os.path.join(workdir, "models"),
+                                    storage_spec)
+        runtime = LocalProcessRuntime(os.path.join(workdir, "jobs"),
+                                      storage=storage)
         self.registry = ModelRegistry(storage)
         self.metrics = JobMetrics()
         cfg = ControllerConfig(
@@ -184,6 +186,10 @@ def main():
                     help='tenant=gpus or tenant={"gpu":8,"cpu":64,'
                          '"memory_mb":512000}; repeatable')
     ap.add_argument("--sync-period", type=float, default=0.5)
+    ap.add_argument("--storage", default="",
+                    help='storage spec JSON, e.g. {"nfs": {"server": '
+                         '"10.0.0.2", "path": "/exports/models"}} or '
+                         '{"localStorage": {"path": "/data/models"}}')
     args = ap.parse_args()
 
     quotas = {}
@@ -196,7 +202,9 @@ def main():
     gates = feat.FeatureGates.from_flag(args.feature_gates)
     mgr = Manager(args.workdir, num_gpus=args.num_gpus,
                   quotas=quotas or None, gates=gates,
-                  sync_period=args.sync_period)
+                  sync_period=args.sync_period,
+                  storage_spec=json.loads(args.storage) if args.storage
+                  else None)
     start_metrics_server(args.metrics_addr)
     log.info("workdir=%s gpus=%d gates=%s", args.workdir, args.num_gpus,
              gates.as_dict())
