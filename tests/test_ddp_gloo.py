@@ -354,6 +354,11 @@ def test_rccl_env_tuning_flagged(monkeypatch):
     from torch_on_k8s_amd.parallel import rccl
     for k in list(rccl.XGMI_TUNING) + ["TOK_RCCL_TUNE"]:
         monkeypatch.delenv(k, raising=False)
+    # OPT-IN: default applies only the required IPC setting
+    applied = rccl.apply_rccl_env(8)
+    assert "NCCL_MIN_NCHANNELS" not in os.environ
+    assert applied["HSA_ENABLE_IPC_MODE_LEGACY"] == "0"
+    monkeypatch.setenv("TOK_RCCL_TUNE", "1")
     applied = rccl.apply_rccl_env(8)
     assert os.environ["NCCL_MIN_NCHANNELS"] == "28"
     assert "NCCL_BUFFSIZE" in applied
@@ -361,9 +366,3 @@ def test_rccl_env_tuning_flagged(monkeypatch):
     monkeypatch.setenv("NCCL_MIN_NCHANNELS", "64")
     rccl.apply_rccl_env(8)
     assert os.environ["NCCL_MIN_NCHANNELS"] == "64"
-    # flag off: no tuning beyond required IPC setting
-    for k in list(rccl.XGMI_TUNING):
-        monkeypatch.delenv(k, raising=False)
-    monkeypatch.setenv("TOK_RCCL_TUNE", "0")
-    rccl.apply_rccl_env(8)
-    assert "NCCL_MIN_NCHANNELS" not in os.environ

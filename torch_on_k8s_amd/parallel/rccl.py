@@ -10,12 +10,13 @@ therefore about channel count and buffering, not algorithm overrides —
 RCCL's topology detection already prefers the right algorithm for the
 fully-connected node.
 
-Applied only when TOK_RCCL_TUNE != "0" (flag per r1 VERDICT next-#2),
-and only via setdefault so an operator's explicit env always wins.
-Values are starting points chosen from the link math above; the bench
-harness (bench.py --gpus N) is the way to validate them per workload —
-an 8-GPU node was not available to this builder (driver runs the
-scaling bench), so they are deliberately conservative.
+Applied only when TOK_RCCL_TUNE == "1" (OPT-IN; flag per r1 VERDICT
+next-#2), and only via setdefault so an operator's explicit env always
+wins. Opt-in because these values are unvalidated on real multi-GPU
+hardware (an 8-GPU node was not available to this builder): forcing
+more channels can also COST compute throughput during overlap — each
+RCCL channel occupies CUs that the backward pass wants. Validate with
+`bench.py --gpus N` A/B before enabling in production.
 """
 from __future__ import annotations
 
@@ -48,7 +49,7 @@ def apply_rccl_env(world_size: int = 0) -> dict:
     for k, v in REQUIRED.items():
         os.environ.setdefault(k, v)
         applied[k] = os.environ[k]
-    if os.environ.get("TOK_RCCL_TUNE", "1") == "0":
+    if os.environ.get("TOK_RCCL_TUNE", "0") != "1":
         return applied
     if world_size and world_size < 2:
         return applied  # single rank: no collectives to tune
