@@ -527,15 +527,23 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
         const int qg_lane = m0 + qs * 32 + (lane & 31);
         const float lse_lane = (qg_lane < S) ? lsep[qg_lane] : NEG_INF;
         f32x16 pt;
+        // interior subtile: all q rows >= every kv row of the wave and
+        // in-bounds -> the per-element mask chain is dead VALU work
+        // (PMC r2: bwd is ~11 VALU/MFMA). One loop, uniform flag: the
+        // exp stays common so the compiler shares registers.
+        const bool need_mask = (CAUSAL && m0 + qs * 32 < n0w + 32) ||
+                               (m0 + qs * 32 + 32 > S) || (n0w + 32 > S);
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int rowidx = (r & 3) + 8 * (r >> 2) + 4 * hi;
-          const int qg = m0 + qs * 32 + rowidx;
           const float lse_r = __shfl(lse_lane, rowidx, 64);
-          float pv = 0.f;
-          if (qg < S && kvrow < S && (!CAUSAL || qg >= kvrow) &&
-              lse_r != NEG_INF)
-            pv = __expf(sv[r] * scale - lse_r);
+          float pv = __expf(sv[r] * scale - lse_r);
+          if (need_mask) {
+            const int qg = m0 + qs * 32 + rowidx;
+            if (!(qg < S && kvrow < S && (!CAUSAL || qg >= kvrow) &&
+                  lse_r != NEG_INF))
+              pv = 0.f;
+          }
           pt[r] = pv;
         }
         bf16x8v pa0, pa1;
@@ -681,16 +689,21 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
         const float lse_lane = (qg_lane < S) ? lsep[qg_lane] : NEG_INF;
         const float ds_lane = (qg_lane < S) ? dsp[qg_lane] : 0.f;
         f32x16 dst;
+        // interior fast path, one loop + uniform flag (see dv kernel)
+        const bool need_mask = (CAUSAL && m0 + qs * 32 < n0w + 32) ||
+                               (m0 + qs * 32 + 32 > S) || (n0w + 32 > S);
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int rowidx = (r & 3) + 8 * (r >> 2) + 4 * hi;
-          const int qg = m0 + qs * 32 + rowidx;
           const float lse_r = __shfl(lse_lane, rowidx, 64);
           const float ds_r = __shfl(ds_lane, rowidx, 64);
-          float pv = 0.f;
-          if (qg < S && kvrow < S && (!CAUSAL || qg >= kvrow) &&
-              lse_r != NEG_INF)
-            pv = __expf(sv[r] * scale - lse_r);
+          float pv = __expf(sv[r] * scale - lse_r);
+          if (need_mask) {
+            const int qg = m0 + qs * 32 + rowidx;
+            if (!(qg < S && kvrow < S && (!CAUSAL || qg >= kvrow) &&
+                  lse_r != NEG_INF))
+              pv = 0.f;
+          }
           dst[r] = pv * (dpv[r] - ds_r);
         }
         bf16x8v da0, da1;
@@ -826,13 +839,21 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
         }
 
         f32x16 dst;
+        // interior subtile: every (q, kv) pair of this wave is strictly
+        // causal-valid and in-bounds -> the per-element mask chain is
+        // dead VALU work (PMC r2: bwd is ~11 VALU/MFMA). One loop with
+        // a uniform flag keeps the register footprint shared.
+        const bool need_mask = (CAUSAL && n0 + ks * 32 + 32 > m0w) ||
+                               (n0 + ks * 32 + 32 > S) || (m0w + 32 > S);
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const int kvg = n0 + ks * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          float pv = 0.f;
-          if (qrow < S && kvg < S && (!CAUSAL || kvg <= qrow) &&
-              lse_lane != NEG_INF)
-            pv = __expf(sv[r] * scale - lse_lane);
+          float pv = __expf(sv[r] * scale - lse_lane);
+          if (need_mask) {
+            const int kvg = n0 + ks * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+            if (!(qrow < S && kvg < S && (!CAUSAL || kvg <= qrow) &&
+                  lse_lane != NEG_INF))
+              pv = 0.f;
+          }
           dst[r] = pv * (dpv[r] - ds_lane);
         }
 
