@@ -270,6 +270,37 @@ class ModelRegistry:
             shutil.rmtree(image_dir, ignore_errors=True)
         return mv
 
+    def reindex(self) -> int:
+        """Rebuild the in-memory Model/ModelVersion view from the OCI
+        layouts on disk (a fresh process — e.g. a serving daemon —
+        pointed at the manager's artifact store). Returns the number of
+        versions found."""
+        found = 0
+        root = self.storage.artifact_dir()
+        for model_name in sorted(os.listdir(root)):
+            mdir = os.path.join(root, model_name)
+            if not os.path.isdir(mdir):
+                continue
+            for version in sorted(os.listdir(mdir)):
+                vdir = os.path.join(mdir, version)
+                idx = os.path.join(vdir, "index.json")
+                if not os.path.isfile(idx):
+                    continue
+                try:
+                    with open(idx) as f:
+                        digest = json.load(f)["manifests"][0]["digest"]
+                except (OSError, ValueError, KeyError, IndexError):
+                    continue
+                model = self.ensure_model(model_name)
+                if version not in model.versions:
+                    model.versions[version] = ModelVersion(
+                        model=model_name, version=version, image_ref=vdir,
+                        digest=digest, build_phase="Succeeded",
+                        storage=self.storage.provenance())
+                    model.latest_version = version  # sorted: last wins
+                found += 1
+        return found
+
     def get_version(self, model_name: str, version: str) -> ModelVersion | None:
         m = self.models.get(model_name)
         return m.versions.get(version) if m else None
