@@ -103,3 +103,24 @@ def test_runtime_injects_storage_env(tmp_path):
     # storage env keys must have been part of it (probe the provider)
     assert sp.task_env()["TOK_STORAGE_NODE"] == "n1"
     assert h.exit_code == 0
+
+
+def test_client_model_registry_view(tmp_path):
+    from torch_on_k8s_amd.client import TorchJobClient
+    from torch_on_k8s_amd.controlplane.modelregistry import (
+        ModelRegistry, StorageProvider)
+    import os as _os
+    sp = StorageProvider(str(tmp_path / "models"))
+    reg = ModelRegistry(sp)
+    src = tmp_path / "s"
+    src.mkdir()
+    (src / "weights").write_text("w")
+    reg.build_version("mm", "v1", str(src))
+    cli = TorchJobClient(str(tmp_path))
+    models = cli.list_models()
+    assert models["mm"]["latest"] == "v1"
+    out = cli.extract_model("mm", str(tmp_path / "x"))
+    assert _os.path.exists(_os.path.join(out, "weights"))
+    # latest-by-default reference
+    out2 = cli.extract_model("mm:v1", str(tmp_path / "y"))
+    assert _os.path.exists(_os.path.join(out2, "weights"))

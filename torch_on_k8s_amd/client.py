@@ -66,6 +66,34 @@ class TorchJobClient:
         raise TimeoutError(f"job {name} did not reach {phases}; "
                            f"last status: {self.get(name)}")
 
+    # -- model registry view (Model/ModelVersion read API analog) ------
+    def _registry(self):
+        from torch_on_k8s_amd.controlplane.modelregistry import (
+            ModelRegistry, StorageProvider)
+        reg = ModelRegistry(StorageProvider(
+            os.path.join(self.workdir, "models")))
+        reg.reindex()
+        return reg
+
+    def list_models(self) -> dict:
+        """{model: {latest, versions: [..]}} from the OCI artifact store."""
+        reg = self._registry()
+        return {m.name: {"latest": m.latest_version,
+                         "versions": sorted(m.versions)}
+                for m in reg.models.values()}
+
+    def extract_model(self, ref: str, dest: str) -> str:
+        """Unpack model:version (OCI layout) into dest; returns the
+        model rootfs path (podman-run analog)."""
+        model, _, version = ref.partition(":")
+        reg = self._registry()
+        if not version:
+            m = reg.models.get(model)
+            version = m.latest_version if m else None
+        if not version:
+            raise KeyError(f"no built version for {model}")
+        return reg.extract(model, version, dest)
+
 
 def main():
     """kubectl-style CLI over the manager workdir:
@@ -83,6 +111,10 @@ def main():
         pc = sub.add_parser(c)
         pc.add_argument("name")
     sub.add_parser("list")
+    sub.add_parser("models")
+    p_ex = sub.add_parser("extract")
+    p_ex.add_argument("ref", help="model[:version] (default: latest)")
+    p_ex.add_argument("dest")
     args = ap.parse_args()
 
     cli = TorchJobClient(args.workdir)
@@ -103,6 +135,10 @@ def main():
         for n in cli.list():
             st = cli.get(n) or {}
             print(f"{n:30s} {st.get('phase')}")
+    elif args.cmd == "models":
+        print(json.dumps(cli.list_models(), indent=2))
+    elif args.cmd == "extract":
+        print(cli.extract_model(args.ref, args.dest))
 
 
 if __name__ == "__main__":
