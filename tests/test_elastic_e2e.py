@@ -174,3 +174,65 @@ def test_fast_rejoin_keeps_processes_resident(tmp_path):
               f"{downtime:.1f}s")
     finally:
         ctl.delete_job(job.name)
+
+
+@pytest.mark.timeout(420)
+def test_fast_rejoin_scale_in(tmp_path):
+    """Scale-IN with fast-rejoin: 3 ranks -> 2; the victim worker exits
+    cleanly on its own, survivors keep their PIDs and continue at the
+    smaller world, and the job runs to completion."""
+    node = NodeState(num_gpus=0)
+    rt = LocalProcessRuntime(str(tmp_path / "work"))
+    ctl = JobController(node, rt,
+                        ControllerConfig(enable_gang_scheduling=False),
+                        elastic=ElasticScaler())
+    steps = 40
+    job = TorchJob(
+        name="shrink-e2e",
+        tasks={
+            TaskType.MASTER: TaskSpec(replicas=1, gpus_per_task=0,
+                                      env=task_env(steps)),
+            TaskType.WORKER: TaskSpec(replicas=2, gpus_per_task=0,
+                                      env=task_env(steps)),
+        },
+        elastic=ElasticPolicy(min_replicas=1, max_replicas=2),
+    )
+    ctl.create_job(job)
+    mpath = tmp_path / "work" / "shrink-e2e" / "metrics.json"
+    try:
+        t0 = time.time()
+        while time.time() - t0 < 120:
+            ctl.reconcile(job)
+            if mpath.exists() and json.load(open(mpath))["step"] >= 3:
+                break
+            time.sleep(0.2)
+        assert mpath.exists(), "training never started"
+        assert json.load(open(mpath))["world_size"] == 3
+        master_pid = ctl.handles["shrink-e2e"][
+            ("shrink-e2e", TaskType.MASTER, 0)].proc.pid
+
+        ElasticScaler.scale(job, 1)  # 2 workers -> 1 (world 3 -> 2)
+        t0 = time.time()
+        done = False
+        while time.time() - t0 < 240:
+            ctl.reconcile(job)
+            if job.status.phase in (JobConditionType.SUCCEEDED,
+                                    JobConditionType.FAILED):
+                done = True
+                break
+            time.sleep(0.2)
+        assert done and job.status.phase == JobConditionType.SUCCEEDED, \
+            (job.status.phase, [(e.reason, e.message) for e in ctl.events])
+        # master survived the shrink in place
+        mlog = (tmp_path / "work" / "shrink-e2e" /
+                "shrink-e2e-master-0.log").read_text()
+        assert "fast-rejoin: world=2" in mlog
+        hs_master = ctl.handles.get("shrink-e2e", {}).get(
+            ("shrink-e2e", TaskType.MASTER, 0))
+        if hs_master is not None:  # handle kept until cleanup
+            assert hs_master.proc.pid == master_pid
+        # the victim (worker-1) exited cleanly by itself (exit 0 logged
+        # as task success, not a failover)
+        assert job.status.restart_count == 0
+    finally:
+        ctl.delete_job(job.name)
