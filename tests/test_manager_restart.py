@@ -1,0 +1,118 @@
+"""Manager-restart durability (reference adopt/claim, pod.go:717-745 +
+the API-server-backed state the operator rebuilds from): a new
+controller over the same workdir must ADOPT the running gang — same
+PIDs, no duplicate processes, same rendezvous port, preserved
+generation — and drive the job to completion."""
+from __future__ import annotations
+
+import json
+import os
+import time
+
+import pytest
+
+from torch_on_k8s_amd.controlplane.api import (JobConditionType, TaskSpec,
+                                               TaskType, TorchJob)
+from torch_on_k8s_amd.controlplane.controller import (ControllerConfig,
+                                                      JobController)
+from torch_on_k8s_amd.controlplane.node import NodeState
+from torch_on_k8s_amd.controlplane.runtime import LocalProcessRuntime
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def task_env(steps):
+    return {
+        "TOK_BACKEND": "gloo",
+        "TOK_TRAIN_STEPS": str(steps),
+        "TOK_TRAINER_CONFIG": json.dumps(
+            {"model": "llama-tiny", "micro_batch": 1, "seq_len": 32}),
+        "PYTHONPATH": ROOT,
+    }
+
+
+def mk_job(steps):
+    return TorchJob(
+        name="restart-e2e",
+        tasks={
+            TaskType.MASTER: TaskSpec(replicas=1, gpus_per_task=0,
+                                      env=task_env(steps)),
+            TaskType.WORKER: TaskSpec(replicas=1, gpus_per_task=0,
+                                      env=task_env(steps)),
+        })
+
+
+@pytest.mark.timeout(420)
+def test_manager_restart_adopts_running_gang(tmp_path):
+    work = str(tmp_path / "work")
+    steps = 60
+    rt_a = LocalProcessRuntime(work)
+    ctl_a = JobController(NodeState(num_gpus=0), rt_a,
+                          ControllerConfig(enable_gang_scheduling=False))
+    job_a = ctl_a.create_job(mk_job(steps))
+    mpath = tmp_path / "work" / "restart-e2e" / "metrics.json"
+    t0 = time.time()
+    while time.time() - t0 < 120:
+        ctl_a.reconcile(job_a)
+        if mpath.exists() and json.load(open(mpath))["step"] >= 2:
+            break
+        time.sleep(0.2)
+    assert mpath.exists(), "training never started"
+    pids_a = {k: h.proc.pid for k, h in ctl_a.handles["restart-e2e"].items()}
+    port_a = ctl_a._master_port(job_a)
+    try:
+        # "manager crash": drop controller A without touching processes
+        del ctl_a
+
+        rt_b = LocalProcessRuntime(work)
+        ctl_b = JobController(NodeState(num_gpus=0), rt_b,
+                              ControllerConfig(enable_gang_scheduling=False))
+        job_b = ctl_b.create_job(mk_job(steps))
+        hs_b = ctl_b.handles["restart-e2e"]
+        # adopted, not recreated: same pids, no extra processes
+        assert {k: h.pid for k, h in hs_b.items()} == pids_a, \
+            (pids_a, {k: (h.pid, h.proc) for k, h in hs_b.items()})
+        assert all(h.proc is None for h in hs_b.values())
+        assert any(e.reason == "TaskAdopted" for e in ctl_b.events)
+        # rendezvous port restored from the persisted job view
+        assert ctl_b._master_port(job_b) == port_a
+
+        t0 = time.time()
+        while time.time() - t0 < 240:
+            ctl_b.reconcile(job_b)
+            if job_b.status.phase in (JobConditionType.SUCCEEDED,
+                                      JobConditionType.FAILED):
+                break
+            time.sleep(0.2)
+        assert job_b.status.phase == JobConditionType.SUCCEEDED, \
+            (job_b.status.phase,
+             [(e.reason, e.message) for e in ctl_b.events])
+        assert job_b.status.restart_count == 0
+    finally:
+        # belt-and-braces: no orphans left behind
+        ctl = locals().get("ctl_b") or locals().get("ctl_a")
+        if ctl is not None:
+            ctl.delete_job("restart-e2e")
+
+
+def test_adopted_exit_marker_grades_completion(tmp_path):
+    """A framework task that exits while the manager is down leaves an
+    exit marker; the next manager grades it instead of re-running a
+    finished task as failed."""
+    from torch_on_k8s_amd.controlplane.runtime import TaskHandle, task_name
+    rt = LocalProcessRuntime(str(tmp_path))
+    h = TaskHandle("j", TaskType.MASTER, 0)
+    h.pid = 2**22 + os.getpid()  # definitely not alive
+    from torch_on_k8s_amd.controlplane.api import TaskPhase
+    h.phase = TaskPhase.RUNNING
+    d = tmp_path / "j" / "tasks"
+    d.mkdir(parents=True)
+    (d / f"{task_name('j', TaskType.MASTER, 0)}.exit").write_text("0")
+    rt.poll(h)
+    assert h.phase == TaskPhase.SUCCEEDED and h.exit_code == 0
+    # without a marker: conservative failure with AdoptedExit reason
+    h2 = TaskHandle("j", TaskType.WORKER, 0)
+    h2.pid = 2**22 + os.getpid()
+    h2.phase = TaskPhase.RUNNING
+    rt.poll(h2)
+    assert h2.phase == TaskPhase.FAILED and h2.reason == "AdoptedExit"

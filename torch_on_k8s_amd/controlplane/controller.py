@@ -104,10 +104,13 @@ class JobController:
 
     def create_job(self, job: TorchJob) -> TorchJob:
         """OnOwnerCreate analog (eventhandler.go:38-64): default, mark
-        Created, enqueue to the coordinator (or reconcile directly)."""
+        Created, adopt any orphans a previous manager left running
+        (pod.go:717-745 adopt/claim parity), enqueue to the coordinator
+        (or reconcile directly)."""
         set_defaults(job)
         self.jobs[job.name] = job
         self.handles.setdefault(job.name, {})
+        self._adopt_orphans(job)
         job.status.set_condition(JobConditionType.CREATED, "JobCreated")
         if self.metrics:
             self.metrics.created()
@@ -125,6 +128,45 @@ class JobController:
         self.reconcile(job)
         if self.metrics:
             self.metrics.deleted()
+
+    def _adopt_orphans(self, job: TorchJob):
+        """Manager-restart durability: rebuild handles over task
+        processes a previous manager left running (their pid records
+        live in the job state dir), claim their GPU slots, and restore
+        the job's persisted generation/annotations/master port so the
+        gang is neither lost, duplicated, nor spuriously scaled."""
+        rt = self.runtime
+        if not hasattr(rt, "adoptable_tasks"):
+            return
+        recs = rt.adoptable_tasks(job.name)
+        if not recs:
+            return
+        # restore the controller's persisted job view (job.json): the
+        # running tasks were launched against THAT generation and port
+        import json
+        import os
+        try:
+            with open(os.path.join(rt.workdir, job.name, "job.json")) as f:
+                saved = json.load(f)
+        except (OSError, ValueError):
+            saved = {}
+        if saved.get("generation"):
+            job.generation = max(job.generation, int(saved["generation"]))
+        for k, v in (saved.get("annotations") or {}).items():
+            job.annotations.setdefault(k, v)
+        if saved.get("master_port"):
+            self._ports[job.name] = int(saved["master_port"])
+        hs = self.handles[job.name]
+        for rec in recs:
+            t = TaskType(rec["task_type"])
+            idx = int(rec["index"])
+            h = rt.adopt_task(job, t, idx, rec)
+            if h.gpu_slots:
+                self.node.allocate_specific(
+                    h.gpu_slots, (job.name, t.value, idx))
+            hs[h.key] = h
+            self.event(job.name, "Normal", "TaskAdopted",
+                       f"{t.value}-{idx} pid {h.pid} on GPUs {h.gpu_slots}")
 
     def _master_port(self, job: TorchJob) -> int:
         """Per-job master port: RANDOM start within the configured range
@@ -527,6 +569,9 @@ class JobController:
                 "annotations": job.annotations,
                 "replicas": {t.value: s.replicas
                              for t, s in job.tasks.items()},
+                # persisted so a restarted manager adopts at the SAME
+                # rendezvous port (running ranks hold it)
+                "master_port": self._ports.get(job.name),
             }, f)
         os.replace(tmp, os.path.join(jobdir, "job.json"))
 

@@ -32,6 +32,17 @@ class NodeState:
             self.alloc[s] = owner
         return got
 
+    def allocate_specific(self, slots, owner) -> tuple:
+        """Claim EXACT slots (adoption of an already-running task after
+        a manager restart — the process still holds those GPUs)."""
+        for s in slots:
+            if self.alloc.get(s) is not None and self.alloc[s] != owner:
+                raise RuntimeError(f"slot {s} already owned by "
+                                   f"{self.alloc[s]}")
+        for s in slots:
+            self.alloc[s] = owner
+        return tuple(slots)
+
     def release(self, slots) -> None:
         for s in slots:
             self.alloc[s] = None

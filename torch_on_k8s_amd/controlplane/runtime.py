@@ -35,6 +35,7 @@ class TaskHandle:
     restart_count: int = 0
     spot: bool = False   # preemptible overflow replica (SpotTaskSpec)
     proc: object = None  # subprocess.Popen for the local runtime
+    pid: int | None = None  # adopted orphan (manager restart): no Popen
 
     @property
     def key(self):
@@ -165,22 +166,139 @@ class LocalProcessRuntime(Runtime):
         h.proc = subprocess.Popen(argv, env=env, stdout=logf, stderr=logf,
                                   start_new_session=True)
         h.phase = TaskPhase.RUNNING
+        self._write_task_record(logdir, job, t, index, h)
+        return h
+
+    # -- adoption (reference adopt/claim of orphans, pod.go:717-745):
+    # a restarted manager rebuilds handles from per-task pid records so
+    # running gangs are neither lost nor duplicated -------------------
+    @staticmethod
+    def _task_record_path(logdir: str, name: str) -> str:
+        return os.path.join(logdir, "tasks", f"{name}.json")
+
+    def _write_task_record(self, logdir, job, t, index, h):
+        import json
+        d = os.path.join(logdir, "tasks")
+        os.makedirs(d, exist_ok=True)
+        rec = {
+            "pid": h.proc.pid,
+            "task_type": t.value,
+            "index": index,
+            "generation": h.generation,
+            "gpu_slots": list(h.gpu_slots),
+            "spot": h.spot,
+            "start_time": h.start_time,
+        }
+        path = self._task_record_path(logdir, task_name(job.name, t, index))
+        tmp = path + ".tmp"
+        with open(tmp, "w") as f:
+            json.dump(rec, f)
+        os.replace(tmp, path)
+        try:  # a fresh start invalidates any previous exit marker
+            os.unlink(path.replace(".json", ".exit"))
+        except OSError:
+            pass
+
+    def adoptable_tasks(self, job_name: str) -> list:
+        """Live task records for a job (pid still alive). Called by the
+        controller at job (re-)creation before it would start fresh
+        tasks."""
+        import json
+        d = os.path.join(self.workdir, job_name, "tasks")
+        out = []
+        try:
+            files = os.listdir(d)
+        except OSError:
+            return out
+        for f in files:
+            if not f.endswith(".json"):
+                continue
+            try:
+                with open(os.path.join(d, f)) as fh:
+                    rec = json.load(fh)
+            except (OSError, ValueError):
+                continue
+            if "pid" in rec and self._pid_running(int(rec["pid"])):
+                out.append(rec)
+        return out
+
+    def adopt_task(self, job, t, index, rec) -> TaskHandle:
+        """Build a handle over an already-running orphan process."""
+        h = TaskHandle(job.name, t, index,
+                       gpu_slots=tuple(rec.get("gpu_slots") or ()),
+                       generation=int(rec.get("generation", job.generation)))
+        h.spot = bool(rec.get("spot"))
+        h.start_time = rec.get("start_time", h.start_time)
+        h.pid = int(rec["pid"])
+        h.phase = TaskPhase.RUNNING
         return h
 
     def poll(self, h):
-        if h.proc is None or h.finished:
+        if h.finished:
             return h
-        rc = h.proc.poll()
-        if rc is not None:
-            h.exit_code = rc
-            h.phase = TaskPhase.SUCCEEDED if rc == 0 else TaskPhase.FAILED
+        if h.proc is not None:
+            rc = h.proc.poll()
+            if rc is not None:
+                h.exit_code = rc
+                h.phase = TaskPhase.SUCCEEDED if rc == 0 else TaskPhase.FAILED
+            return h
+        if h.pid is not None:  # adopted orphan: not our child
+            if self._pid_running(h.pid):
+                return h  # still alive
+
+            # gone: the framework entrypoint leaves an exit marker; an
+            # opaque command that died while orphaned can't be graded
+            rc = self._read_exit_marker(h)
+            if rc is None:
+                h.exit_code = None
+                h.phase = TaskPhase.FAILED
+                h.reason = "AdoptedExit"
+            else:
+                h.exit_code = rc
+                h.phase = TaskPhase.SUCCEEDED if rc == 0 else TaskPhase.FAILED
         return h
 
+    @staticmethod
+    def _pid_running(pid: int) -> bool:
+        """Alive and NOT a zombie (os.kill(pid, 0) succeeds on zombies;
+        an exited-but-unreaped task must grade as finished)."""
+        try:
+            os.kill(pid, 0)
+        except ProcessLookupError:
+            return False
+        except PermissionError:
+            return True
+        try:
+            with open(f"/proc/{pid}/stat") as f:
+                # state is the field after the parenthesised comm
+                return f.read().rpartition(")")[2].split()[0] != "Z"
+        except (OSError, IndexError):
+            return False
+
+    def _read_exit_marker(self, h):
+        path = self._task_record_path(
+            os.path.join(self.workdir, h.job_name),
+            task_name(h.job_name, h.task_type, h.index)).replace(
+                ".json", ".exit")
+        try:
+            with open(path) as f:
+                return int(f.read().strip())
+        except (OSError, ValueError):
+            return None
+
+    def _live_pid(self, h):
+        if h.proc is not None:
+            return h.proc.pid if h.proc.poll() is None else None
+        if h.pid is not None:
+            return h.pid if self._pid_running(h.pid) else None
+        return None
+
     def kill(self, h, grace=True):
-        if h.proc is None or h.proc.poll() is not None:
+        pid = self._live_pid(h)
+        if pid is None:
             return
         try:
-            pgid = os.getpgid(h.proc.pid)
+            pgid = os.getpgid(pid)
             os.killpg(pgid, signal.SIGTERM if grace else signal.SIGKILL)
         except ProcessLookupError:
             pass
@@ -192,4 +310,10 @@ class LocalProcessRuntime(Runtime):
             except subprocess.TimeoutExpired:
                 self.kill(h, grace=False)
                 h.proc.wait(timeout=10)
+        elif h.pid is not None:
+            deadline = time.time() + timeout
+            while self._live_pid(h) is not None:
+                if time.time() > deadline:
+                    self.kill(h, grace=False)
+                time.sleep(0.1)
         return self.poll(h)

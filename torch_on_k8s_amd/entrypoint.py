@@ -339,5 +339,27 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
     return 0
 
 
+def _write_exit_marker(rc: int):
+    """Exit code record for manager-restart adoption: an adopted orphan
+    is not the new manager's child, so its exit status can't be reaped —
+    it reads this marker instead (runtime._read_exit_marker)."""
+    sd = os.environ.get("TOK_STATE_DIR")
+    if not sd:
+        return
+    name = (f"{os.environ.get('TOK_JOB_NAME', 'job')}-"
+            f"{os.environ.get('TOK_TASK_TYPE', 'task')}-"
+            f"{os.environ.get('TOK_TASK_INDEX', '0')}")
+    try:
+        os.makedirs(os.path.join(sd, "tasks"), exist_ok=True)
+        tmp = os.path.join(sd, "tasks", f".{name}.exit.tmp{os.getpid()}")
+        with open(tmp, "w") as f:
+            f.write(str(rc))
+        os.replace(tmp, os.path.join(sd, "tasks", f"{name}.exit"))
+    except OSError:
+        pass
+
+
 if __name__ == "__main__":
-    sys.exit(main())
+    _rc = main()
+    _write_exit_marker(_rc)
+    sys.exit(_rc)
