@@ -141,11 +141,11 @@ def main() -> int:
         return _train_loop(trainer, ctx, steps_total, state_dir, job_file,
                            agent_file, stop, checkpoint_and_ack, destroy,
                            barrier, write_agent)
-    except RuntimeError as e:
+    except (RuntimeError, TimeoutError) as e:
         # A peer restarting (elastic scale / preemption) tears down the
-        # process group mid-collective; classify as retryable so the
-        # controller restarts us into the new rendezvous (exit-code
-        # contract, controlplane/failover.py).
+        # process group mid-collective, or a torn rejoin timed out;
+        # classify as retryable so the controller restarts us into the
+        # new rendezvous (exit-code contract, controlplane/failover.py).
         print(f"[entrypoint] collective aborted: {e}", flush=True)
         return 143
 
@@ -204,17 +204,22 @@ def _fast_rejoin(trainer, ctx, job_doc, agent_file, write_agent,
     if not survivor:
         return None
 
+    # A torn rejoin (peer crashed mid-scale) must FAIL FAST into the
+    # normal failover path (RuntimeError -> exit 143 -> controller
+    # restarts the gang fresh), not hang the gang on a dead rendezvous.
+    timeout_s = float(os.environ.get("TOK_REJOIN_TIMEOUT", "300"))
     if not ctx.is_main:
         # master must have closed its old store before we reconnect
         # (the controller gates NEW task creation on the same handshake)
-        _poll_file(agent_file, lambda d: d.get("rejoin-ready") == version)
+        _poll_file(agent_file, lambda d: d.get("rejoin-ready") == version,
+                   timeout_s=timeout_s)
 
     rank = 0 if ttype == "master" else tindex + 1
     dist.init_process_group(
         backend=ctx.backend or ("nccl" if ctx.device.type == "cuda"
                                 else "gloo"),
         rank=rank, world_size=new_world,
-        timeout=datetime.timedelta(seconds=300))
+        timeout=datetime.timedelta(seconds=timeout_s))
     ctx.rank = rank
     ctx.world_size = new_world
     trainer.fb.set_world(new_world)
