@@ -56,8 +56,13 @@ def parse_args(argv=None):
     ap.add_argument("--attn", type=str, default="hip", choices=["hip", "sdpa"])
     ap.add_argument("--no-overlap", action="store_true",
                     help="disable grad-sync/backward overlap (ablation)")
-    ap.add_argument("--hip-graph", action="store_true",
-                    help="capture the training step in a hipGraph")
+    ap.add_argument("--hip-graph", dest="hip_graph", action="store_true",
+                    default=None,
+                    help="capture the training step in a hipGraph "
+                         "(default: ON for single-GPU runs — measured "
+                         "+1.1%% at mbs8; OFF at N>1 where RCCL-in-graph "
+                         "is unvalidated)")
+    ap.add_argument("--no-hip-graph", dest="hip_graph", action="store_false")
     ap.add_argument("--via-manager", action="store_true",
                     help="run the step through a gang-scheduled TorchJob "
                          "(control-plane path; reports job-to-Running too)")
@@ -67,6 +72,12 @@ def parse_args(argv=None):
     ap.add_argument("--timeout", type=float, default=1800.0,
                     help="via-manager: max seconds to wait for the job")
     return ap.parse_args(argv)
+
+
+def _resolve_hip_graph(args, world: int) -> bool:
+    if args.hip_graph is not None:
+        return args.hip_graph
+    return world == 1 and args.device == "cuda"
 
 
 def _maybe_self_spawn(args):
@@ -148,7 +159,7 @@ def run_direct(args):
         bucket_mb=args.bucket_mb,
         activation_checkpointing=args.activation_checkpointing,
         overlap_grad_sync=not args.no_overlap,
-        hip_graph=args.hip_graph,
+        hip_graph=_resolve_hip_graph(args, n_gpus),
         dtype="bf16" if args.device == "cuda" else "fp32",
     )
     trainer = Trainer(cfg, ctx)
@@ -206,7 +217,7 @@ def run_via_manager(args):
         "bucket_mb": args.bucket_mb,
         "activation_checkpointing": args.activation_checkpointing,
         "overlap_grad_sync": not args.no_overlap,
-        "hip_graph": args.hip_graph,
+        "hip_graph": _resolve_hip_graph(args, n),
         "dtype": "bf16" if args.device == "cuda" else "fp32",
     }
     env = {
