@@ -261,114 +261,6 @@ __global__ void rmsnorm_res_bwd_dx_kernel_t(const bf16x8* __restrict__ xr,
   }
 }
 
-// -------- wave-per-row fused kernels (H <= 4096) -----------------------
-// One WAVE owns one row: H/64 bf16x8 vecs per lane stay in registers,
-// the reduce is wave-level shuffles only — ZERO __syncthreads. The
-// block-per-row versions above spent ~2 barriers per row and measured
-// ~4 TB/s; barrier-free is the natural shape for H=4096 (8 vecs/lane).
-template <int VPL>  // vecs per lane = H / (8*64)
-__global__ __launch_bounds__(BLOCK, 2) void rmsnorm_res_fwd_wave_kernel(const bf16x8* __restrict__ x,
-                                            const bf16x8* __restrict__ res,
-                                            const bf16x8* __restrict__ w,
-                                            bf16x8* __restrict__ xr,
-                                            bf16x8* __restrict__ y,
-                                            float* __restrict__ invr,
-                                            long nrows, float eps) {
-  constexpr int HC = VPL * WAVE;   // vecs per row
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int wid = threadIdx.x / WAVE;
-  const int wpb = blockDim.x / WAVE;
-  const float inv_h = 1.0f / (float)(HC * 8);
-  bf16x8 wc[VPL];
-#pragma unroll
-  for (int t = 0; t < VPL; ++t) wc[t] = w[lane + t * WAVE];
-  for (long row = (long)blockIdx.x * wpb + wid; row < nrows;
-       row += (long)gridDim.x * wpb) {
-    const bf16x8* xrow = x + row * HC;
-    const bf16x8* rrow = res ? res + row * HC : nullptr;
-    bf16x8* xrout = xr + row * HC;
-    bf16x8 cache[VPL];
-    float ss = 0.f;
-#pragma unroll
-    for (int t = 0; t < VPL; ++t) {
-      const int c = lane + t * WAVE;
-      bf16x8 v = xrow[c], o;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float f = bfbits2f(v.h[j]);
-        if (rrow) f += bfbits2f(rrow[c].h[j]);
-        o.h[j] = f2bfbits(f);
-        float fq = bfbits2f(o.h[j]);
-        ss = fmaf(fq, fq, ss);
-      }
-      cache[t] = o;
-      xrout[c] = o;
-    }
-    ss = wave_reduce_sum(ss);
-    const float r = rsqrtf(ss * inv_h + eps);
-    if (lane == 0) invr[row] = r;
-    bf16x8* yr = y + row * HC;
-#pragma unroll
-    for (int t = 0; t < VPL; ++t) {
-      bf16x8 ov;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        ov.h[j] = f2bfbits(bfbits2f(cache[t].h[j]) * r * bfbits2f(wc[t].h[j]));
-      yr[lane + t * WAVE] = ov;
-    }
-  }
-}
-
-template <int VPL>
-__global__ __launch_bounds__(BLOCK, 2) void rmsnorm_res_bwd_dx_wave_kernel(
-    const bf16x8* __restrict__ xr, const bf16x8* __restrict__ w,
-    const bf16x8* __restrict__ dy, const bf16x8* __restrict__ dxr,
-    const float* __restrict__ invr, bf16x8* __restrict__ dx, long nrows) {
-  constexpr int HC = VPL * WAVE;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int wid = threadIdx.x / WAVE;
-  const int wpb = blockDim.x / WAVE;
-  const float inv_h = 1.0f / (float)(HC * 8);
-  bf16x8 wc[VPL];
-#pragma unroll
-  for (int t = 0; t < VPL; ++t) wc[t] = w[lane + t * WAVE];
-  for (long row = (long)blockIdx.x * wpb + wid; row < nrows;
-       row += (long)gridDim.x * wpb) {
-    const bf16x8* xrow = xr + row * HC;
-    const bf16x8* dyr = dy + row * HC;
-    const float r = invr[row];
-    bf16x8 xc[VPL], dc[VPL];
-    float acc = 0.f;
-#pragma unroll
-    for (int t = 0; t < VPL; ++t) {
-      const int c = lane + t * WAVE;
-      xc[t] = xrow[c];
-      dc[t] = dyr[c];
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        acc = fmaf(bfbits2f(dc[t].h[j]) * bfbits2f(wc[t].h[j]),
-                   bfbits2f(xc[t].h[j]), acc);
-    }
-    acc = wave_reduce_sum(acc);
-    const float kf = acc * r * r * inv_h;
-    bf16x8* dxrow = dx + row * HC;
-    const bf16x8* addrow = dxr ? dxr + row * HC : nullptr;
-#pragma unroll
-    for (int t = 0; t < VPL; ++t) {
-      const int c = lane + t * WAVE;
-      bf16x8 ov;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float g = r * (bfbits2f(wc[t].h[j]) * bfbits2f(dc[t].h[j]) -
-                       bfbits2f(xc[t].h[j]) * kf);
-        if (addrow) g += bfbits2f(addrow[c].h[j]);
-        ov.h[j] = f2bfbits(g);
-      }
-      dxrow[c] = ov;
-    }
-  }
-}
-
 }  // namespace
 
 extern "C" {
@@ -390,29 +282,6 @@ hipError_t tok_rmsnorm_res_fwd(const void* x, const void* res, const void* w,
                                void* xr, void* y, float* invr, long nrows,
                                int H, float eps, hipStream_t stream) {
   const int hc = H / 8;
-  // wave-per-row fast path: zero barriers, row data in registers
-  if (hc % WAVE == 0 && hc / WAVE <= 8) {
-    const int wpb = BLOCK / WAVE;
-    long grid = (nrows + wpb - 1) / wpb;
-    if (grid > 32768) grid = 32768;
-    if (grid < 1) grid = 1;
-#define LAUNCH_WAVE_FWD(V)                                                    \
-    rmsnorm_res_fwd_wave_kernel<V><<<(int)grid, BLOCK, 0, stream>>>(          \
-        (const bf16x8*)x, (const bf16x8*)res, (const bf16x8*)w,               \
-        (bf16x8*)xr, (bf16x8*)y, invr, nrows, eps)
-    switch (hc / WAVE) {
-      case 1: LAUNCH_WAVE_FWD(1); break;
-      case 2: LAUNCH_WAVE_FWD(2); break;
-      case 4: LAUNCH_WAVE_FWD(4); break;
-      case 8: LAUNCH_WAVE_FWD(8); break;
-      case 3: LAUNCH_WAVE_FWD(3); break;
-      case 5: LAUNCH_WAVE_FWD(5); break;
-      case 6: LAUNCH_WAVE_FWD(6); break;
-      default: LAUNCH_WAVE_FWD(7); break;
-    }
-#undef LAUNCH_WAVE_FWD
-    return hipGetLastError();
-  }
   int grid = (int)(nrows < 8192 ? nrows : 8192);
   if (grid < 1) grid = 1;
   if (hc <= BLOCK)
@@ -439,34 +308,6 @@ hipError_t tok_rmsnorm_res_bwd(const void* xr, const void* w, const void* dy,
                                void* dx, float* dw_f32, float* dw_ws,
                                long nrows, int H, hipStream_t stream) {
   const int hc = H / 8;
-  if (hc % WAVE == 0 && hc / WAVE <= 8) {
-    const int wpb = BLOCK / WAVE;
-    long grid = (nrows + wpb - 1) / wpb;
-    if (grid > 32768) grid = 32768;
-    if (grid < 1) grid = 1;
-#define LAUNCH_WAVE_BWD(V)                                                    \
-    rmsnorm_res_bwd_dx_wave_kernel<V><<<(int)grid, BLOCK, 0, stream>>>(       \
-        (const bf16x8*)xr, (const bf16x8*)w, (const bf16x8*)dy,               \
-        (const bf16x8*)dxr_in, invr, (bf16x8*)dx, nrows)
-    switch (hc / WAVE) {
-      case 1: LAUNCH_WAVE_BWD(1); break;
-      case 2: LAUNCH_WAVE_BWD(2); break;
-      case 4: LAUNCH_WAVE_BWD(4); break;
-      case 8: LAUNCH_WAVE_BWD(8); break;
-      case 3: LAUNCH_WAVE_BWD(3); break;
-      case 5: LAUNCH_WAVE_BWD(5); break;
-      case 6: LAUNCH_WAVE_BWD(6); break;
-      default: LAUNCH_WAVE_BWD(7); break;
-    }
-#undef LAUNCH_WAVE_BWD
-    const int rsplit = tok_rmsnorm_dw_rsplit(nrows, H);
-    int cblocks = (hc + BLOCK - 1) / BLOCK;
-    dim3 g(cblocks, rsplit);
-    rmsnorm_bwd_dw_kernel<<<g, BLOCK, 0, stream>>>(
-        (const bf16x8*)xr, (const bf16x8*)dy, invr, dw_ws, nrows, hc);
-    launch_colsum(dw_ws, dw_f32, rsplit, (long)H, stream);
-    return hipGetLastError();
-  }
   int grid = (int)(nrows < 8192 ? nrows : 8192);
   if (grid < 1) grid = 1;
   if (hc <= BLOCK)
