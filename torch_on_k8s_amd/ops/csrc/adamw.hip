@@ -27,12 +27,10 @@ __global__ void adamw_kernel(bf16x8* __restrict__ p, const bf16x8* __restrict__ 
   }
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
-    // non-temporal: the optimizer state (2/3 of this kernel's ~176 GB
-    // of traffic) is touched exactly once per step
-    bf16x8 pv = nt_load16(p + i);
-    bf16x8 gv = nt_load16(g + i);
-    f32x4v m0 = nt_load16(m + i * 2), m1 = nt_load16(m + i * 2 + 1);
-    f32x4v v0 = nt_load16(v + i * 2), v1 = nt_load16(v + i * 2 + 1);
+    bf16x8 pv = p[i];
+    bf16x8 gv = g[i];
+    f32x4v m0 = m[i * 2], m1 = m[i * 2 + 1];
+    f32x4v v0 = v[i * 2], v1 = v[i * 2 + 1];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float* mj = (j < 4) ? &m0.v[j] : &m1.v[j - 4];
@@ -49,11 +47,11 @@ __global__ void adamw_kernel(bf16x8* __restrict__ p, const bf16x8* __restrict__ 
       pf -= lr * mhat / (sqrtf(vhat) + eps);
       pv.h[j] = f2bfbits(pf);
     }
-    nt_store16(p + i, pv);
-    nt_store16(m + i * 2, m0);
-    nt_store16(m + i * 2 + 1, m1);
-    nt_store16(v + i * 2, v0);
-    nt_store16(v + i * 2 + 1, v1);
+    p[i] = pv;
+    m[i * 2] = m0;
+    m[i * 2 + 1] = m1;
+    v[i * 2] = v0;
+    v[i * 2 + 1] = v1;
   }
 }
 

@@ -41,27 +41,6 @@ struct alignas(16) f32x4v {
   float v[4];
 };
 
-// Non-temporal 16-byte accesses for pure streaming kernels (working set
-// >> L2; every byte touched once): skip L2 fills / avoid evicting data
-// other kernels reuse. The builtin wants a scalar/ext-vector pointee.
-using u32x4nt = __attribute__((ext_vector_type(4))) unsigned int;
-
-template <typename T>
-DEVINL T nt_load16(const T* p) {
-  static_assert(sizeof(T) == 16, "16B only");
-  union { u32x4nt u; T v; } c;
-  c.u = __builtin_nontemporal_load((const u32x4nt*)p);
-  return c.v;
-}
-
-template <typename T>
-DEVINL void nt_store16(T* p, T v) {
-  static_assert(sizeof(T) == 16, "16B only");
-  union { u32x4nt u; T v; } c;
-  c.v = v;
-  __builtin_nontemporal_store(c.u, (u32x4nt*)p);
-}
-
 // ---- wave / block reductions ------------------------------------------
 DEVINL float wave_reduce_sum(float x) {
 #pragma unroll
