@@ -23,23 +23,6 @@ constexpr int BLOCK = 256;
 
 DEVINL float sigmoidf(float x) { return 1.0f / (1.0f + __expf(-x)); }
 
-// Pure streaming kernel (every byte touched exactly once, working set
-// ~2-5 GB >> 32 MB L2): non-temporal loads/stores keep the tensors from
-// evicting data other kernels reuse, and skip needless L2 fills.
-using u32x4 = __attribute__((ext_vector_type(4))) unsigned int;
-
-DEVINL bf16x8 nt_load(const bf16x8* p) {
-  union { u32x4 u; bf16x8 v; } c;
-  c.u = __builtin_nontemporal_load((const u32x4*)p);
-  return c.v;
-}
-
-DEVINL void nt_store(bf16x8* p, bf16x8 v) {
-  union { u32x4 u; bf16x8 v; } c;
-  c.v = v;
-  __builtin_nontemporal_store(c.u, (u32x4*)p);
-}
-
 // out[r, c] = silu(g) * u,  g = gu[r, c], u = gu[r, I + c]
 // 2D grid: blockIdx.x covers columns, blockIdx.y strides rows — no
 // 64-bit integer division in the hot loop (no HW divide on CDNA; the
@@ -51,14 +34,14 @@ __global__ void swiglu_fwd_kernel(const bf16x8* __restrict__ gu,
   if (c >= iv) return;
   for (long row = blockIdx.y; row < rows; row += gridDim.y) {
     const bf16x8* gur = gu + row * (2 * iv);
-    bf16x8 g = nt_load(gur + c), u = nt_load(gur + c + iv), o;
+    bf16x8 g = gur[c], u = gur[c + iv], o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const float gf = bfbits2f(g.h[j]);
       const float s = gf * sigmoidf(gf);
       o.h[j] = f2bfbits(s * bfbits2f(u.h[j]));
     }
-    nt_store(out + row * iv + c, o);
+    out[row * iv + c] = o;
   }
 }
 
@@ -72,8 +55,8 @@ __global__ void swiglu_bwd_kernel(const bf16x8* __restrict__ dout,
   if (c >= iv) return;
   for (long row = blockIdx.y; row < rows; row += gridDim.y) {
     const bf16x8* gur = gu + row * (2 * iv);
-    bf16x8 g = nt_load(gur + c), u = nt_load(gur + c + iv);
-    bf16x8 do8 = nt_load(dout + row * iv + c);
+    bf16x8 g = gur[c], u = gur[c + iv];
+    bf16x8 do8 = dout[row * iv + c];
     bf16x8 dg, du;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -86,8 +69,8 @@ __global__ void swiglu_bwd_kernel(const bf16x8* __restrict__ dout,
       du.h[j] = f2bfbits(d * silu);
     }
     bf16x8* dgur = dgu + row * (2 * iv);
-    nt_store(dgur + c, dg);
-    nt_store(dgur + c + iv, du);
+    dgur[c] = dg;
+    dgur[c + iv] = du;
   }
 }
 
