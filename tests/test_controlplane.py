@@ -480,3 +480,40 @@ def test_model_version_build_failure_lifecycle(tmp_path):
     assert bad.build_phase == "Failed"
     assert reg.models["m2"].latest_version == "v1"
     assert not _os.path.exists(bad.image_ref)
+
+
+def test_rejoin_gate_holds_new_task_creation():
+    """Fast-rejoin stage-2 gate (the reference's refreshStaleService
+    stale-selector trick, elastic_scale.go:402-424: new members must not
+    see the OLD rendezvous): after a scale-out checkpoint completes, new
+    worker creation is HELD until the surviving master reports its store
+    closed (rejoin-ready annotation)."""
+    import json as _json
+    ctl, node, rt = mk_ctl(num_gpus=0)
+    ctl.elastic = ElasticScaler()
+    job = mk_job("rg", workers=1,
+                 elastic=ElasticPolicy(min_replicas=1, max_replicas=2))
+    for s in job.tasks.values():
+        s.gpus_per_task = 0
+    ctl.create_job(job)
+    ctl.reconcile(job)
+    ctl.reconcile(job)
+    assert len(ctl.handles["rg"]) == 2  # master + worker running
+
+    ElasticScaler.scale(job, 2)  # generation bump
+    ctl.reconcile(job)           # stage 1: checkpoint requested
+    assert ANN_CKPT_REQUESTED in job.annotations
+    ElasticScaler.complete_checkpoint(job)  # data plane acks
+    ctl.reconcile(job)           # stage 2: adopt survivors...
+    # ...but the NEW worker (index 1) must NOT exist yet: the master has
+    # not reported rejoin-ready, so its old store may still be open
+    assert ("rg", TaskType.WORKER, 1) not in ctl.handles["rg"]
+    # survivors were adopted into the new generation, not killed
+    assert ctl.handles["rg"][("rg", TaskType.MASTER, 0)].generation == \
+        job.generation
+    assert not rt.killed
+
+    # master closes its store and handshakes
+    job.annotations["rejoin-ready"] = str(job.generation)
+    ctl.reconcile(job)
+    assert ("rg", TaskType.WORKER, 1) in ctl.handles["rg"]
