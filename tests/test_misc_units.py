@@ -124,3 +124,19 @@ def test_client_model_registry_view(tmp_path):
     # latest-by-default reference
     out2 = cli.extract_model("mm:v1", str(tmp_path / "y"))
     assert _os.path.exists(_os.path.join(out2, "weights"))
+
+
+def test_training_metrics_gauges():
+    from torch_on_k8s_amd.controlplane.metrics import JobMetrics, HAVE_PROM
+    m = JobMetrics()
+    m.set_training_metrics("j1", {"step": 7, "loss": 3.25,
+                                  "tokens_per_s": 17600.0})
+    if HAVE_PROM:
+        from prometheus_client import REGISTRY
+        assert REGISTRY.get_sample_value(
+            "torch_on_k8s_amd_job_tokens_per_second", {"job": "j1"}) == 17600.0
+        assert REGISTRY.get_sample_value(
+            "torch_on_k8s_amd_job_step", {"job": "j1"}) == 7
+        m.remove_job("j1")
+        assert REGISTRY.get_sample_value(
+            "torch_on_k8s_amd_job_tokens_per_second", {"job": "j1"}) is None

@@ -50,6 +50,18 @@ class JobMetrics:
                 "all_delay": Histogram(
                     "torch_on_k8s_amd_job_all_tasks_launch_delay_seconds",
                     "job create -> job running", **kw),
+                # data-plane gauges fed from the trainers' structured
+                # metrics files (the reference has no data-plane view at
+                # all — it scrapes worker stdout with a regex)
+                "job_tokens_per_s": Gauge(
+                    "torch_on_k8s_amd_job_tokens_per_second",
+                    "training throughput per job", ["job"], **kw),
+                "job_step": Gauge(
+                    "torch_on_k8s_amd_job_step",
+                    "latest optimizer step per job", ["job"], **kw),
+                "job_loss": Gauge(
+                    "torch_on_k8s_amd_job_loss",
+                    "latest training loss per job", ["job"], **kw),
             }
         self.m = JobMetrics._created
 
@@ -91,6 +103,27 @@ class JobMetrics:
     def set_queue_depth(self, queue: str, depth: int):
         if HAVE_PROM:
             self.m["queue_pending"].labels(queue=queue).set(depth)
+
+    def set_training_metrics(self, job: str, rec: dict):
+        """Publish a trainer metrics.json record (step/loss/tokens_per_s)."""
+        if not HAVE_PROM:
+            return
+        if rec.get("tokens_per_s") is not None:
+            self.m["job_tokens_per_s"].labels(job=job).set(
+                rec["tokens_per_s"])
+        if rec.get("step") is not None:
+            self.m["job_step"].labels(job=job).set(rec["step"])
+        if rec.get("loss") is not None:
+            self.m["job_loss"].labels(job=job).set(rec["loss"])
+
+    def remove_job(self, job: str):
+        if not HAVE_PROM:
+            return
+        for g in ("job_tokens_per_s", "job_step", "job_loss"):
+            try:
+                self.m[g].remove(job)
+            except KeyError:
+                pass
 
 
 def start_metrics_server(port: int = 8443):

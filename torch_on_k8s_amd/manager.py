@@ -141,6 +141,13 @@ class Manager:
             with open(tmp, "w") as f:
                 json.dump(job_status_dict(job, evs), f, indent=2)
             os.replace(tmp, path)
+            # data-plane gauges from the trainer's structured metrics
+            try:
+                with open(os.path.join(self.workdir, "jobs", name,
+                                       "metrics.json")) as f:
+                    self.metrics.set_training_metrics(name, json.load(f))
+            except (OSError, ValueError):
+                pass
         # drop status files of deleted jobs (daemon hygiene)
         try:
             for f in os.listdir(self.status_dir):
