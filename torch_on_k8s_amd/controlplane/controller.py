@@ -73,6 +73,7 @@ class JobController:
         self._events_by_job: dict[str, object] = {}
         self._event_last: dict = {}
         self._ports: dict[str, int] = {}
+        self._last_jobjson: dict[str, str] = {}
         import random as _random
         lo, hi = self.cfg.master_port_range
         self._next_port = _random.randrange(lo, hi)
@@ -193,6 +194,7 @@ class JobController:
             self._ports.pop(job.name, None)
             self._first_task_ts.pop(job.name, None)
             self._events_by_job.pop(job.name, None)
+            self._last_jobjson.pop(job.name, None)
             for k in [k for k in self._event_last if k[0] == job.name]:
                 del self._event_last[k]
             if self.coordinator is not None:
@@ -561,19 +563,23 @@ class JobController:
             # rendezvous store is closed (elastic.py stage 2 gates new
             # task creation on this)
             job.annotations["rejoin-ready"] = str(agent["rejoin-ready"])
+        doc = json.dumps({
+            "name": job.name,
+            "generation": job.generation,
+            "annotations": job.annotations,
+            "replicas": {t.value: s.replicas
+                         for t, s in job.tasks.items()},
+            # persisted so a restarted manager adopts at the SAME
+            # rendezvous port (running ranks hold it)
+            "master_port": self._ports.get(job.name),
+        }, sort_keys=True)
+        if self._last_jobjson.get(job.name) == doc:
+            return  # unchanged: skip the write (reconcile runs at ~2 Hz)
         tmp = os.path.join(jobdir, f"job.json.tmp{os.getpid()}")
         with open(tmp, "w") as f:
-            json.dump({
-                "name": job.name,
-                "generation": job.generation,
-                "annotations": job.annotations,
-                "replicas": {t.value: s.replicas
-                             for t, s in job.tasks.items()},
-                # persisted so a restarted manager adopts at the SAME
-                # rendezvous port (running ranks hold it)
-                "master_port": self._ports.get(job.name),
-            }, f)
+            f.write(doc)
         os.replace(tmp, os.path.join(jobdir, "job.json"))
+        self._last_jobjson[job.name] = doc
 
     def reconcile_all(self):
         for job in list(self.jobs.values()):

@@ -223,6 +223,14 @@ def _fast_rejoin(trainer, ctx, job_doc, agent_file, write_agent,
     ctx.rank = rank
     ctx.world_size = new_world
     trainer.fb.set_world(new_world)
+    if trainer.cfg.hip_graph and new_world > 1:
+        # a graph captured at world 1 contains NO collectives — replaying
+        # it after scale-out would silently skip the gradient sync
+        trainer.cfg.hip_graph = False
+        trainer._graph = None
+        trainer._graph_state = None
+        print("[entrypoint] fast-rejoin: hipGraph step capture disabled "
+              "(world > 1; RCCL-in-graph unvalidated)", flush=True)
     print(f"[entrypoint] fast-rejoin: world={new_world} rank={rank} "
           f"step={trainer.step_count} (state kept resident)", flush=True)
     return new_world

@@ -76,6 +76,8 @@ class Manager:
             read_trainer_metrics(lambda job: os.path.join(
                 workdir, "jobs", job.name, "metrics.json")))
         self._spooled: dict = {}  # filename -> (job name, mtime)
+        self._last_status: dict = {}  # job -> serialized status (skip
+                                      # unchanged writes; ~2 Hz loop)
 
     # ------------------------------------------------------------------
     def sync_spool(self):
@@ -136,11 +138,14 @@ class Manager:
         for name, job in list(self.controller.jobs.items()):
             live.add(f"{name}.json")
             path = os.path.join(self.status_dir, f"{name}.json")
-            tmp = path + ".tmp"
             evs = self.controller.events_for(name)
-            with open(tmp, "w") as f:
-                json.dump(job_status_dict(job, evs), f, indent=2)
-            os.replace(tmp, path)
+            doc = json.dumps(job_status_dict(job, evs), indent=2)
+            if self._last_status.get(name) != doc:
+                tmp = path + ".tmp"
+                with open(tmp, "w") as f:
+                    f.write(doc)
+                os.replace(tmp, path)
+                self._last_status[name] = doc
             # data-plane gauges from the trainer's structured metrics
             try:
                 with open(os.path.join(self.workdir, "jobs", name,
@@ -153,6 +158,8 @@ class Manager:
             for f in os.listdir(self.status_dir):
                 if f.endswith(".json") and f not in live:
                     os.unlink(os.path.join(self.status_dir, f))
+                    self._last_status.pop(f[:-5], None)
+                    self.metrics.remove_job(f[:-5])
         except OSError:
             pass
         if self.coordinator is not None:
