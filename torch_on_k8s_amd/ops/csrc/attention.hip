@@ -517,7 +517,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
         f32x16 sv;
 #pragma unroll
         for (int r = 0; r < 16; ++r) sv[r] = 0.f;
-        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int c = 0; c < QC; ++c) {
           bf16x8v qa = read_bfrag<D * 2>(
@@ -525,7 +524,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
           sv = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               qa, k_reg[c], sv, 0, 0, 0);
         }
-        __builtin_amdgcn_s_setprio(0);
         const int qg_lane = m0 + qs * 32 + (lane & 31);
         const float lse_lane = (qg_lane < S) ? lsep[qg_lane] : NEG_INF;
         f32x16 pt;
@@ -550,7 +548,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
         }
         bf16x8v pa0, pa1;
         repack_pa(pt, pa0, pa1);
-        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int t = 0; t < DT; ++t) {
           bf16x8v b0 = read_bfrag<128>(
@@ -562,7 +559,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
           dv_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               pa1, b1, dv_acc[t], 0, 0, 0);
         }
-        __builtin_amdgcn_s_setprio(0);
       }
       if (more) {
         char* nq = smem + (cur ^ 1) * (2 * KB);
@@ -676,7 +672,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
           sv[r] = 0.f;
           dpv[r] = 0.f;
         }
-        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int c = 0; c < QC; ++c) {
           bf16x8v qa = read_bfrag<D * 2>(
@@ -690,7 +685,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
           dpv = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               doa, vb, dpv, 0, 0, 0);
         }
-        __builtin_amdgcn_s_setprio(0);
         const int qg_lane = m0 + qs * 32 + (lane & 31);
         const float lse_lane = (qg_lane < S) ? lsep[qg_lane] : NEG_INF;
         const float ds_lane = (qg_lane < S) ? dsp[qg_lane] : 0.f;
@@ -714,7 +708,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
         }
         bf16x8v da0, da1;
         repack_pa(dst, da0, da1);
-        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int t = 0; t < DT; ++t) {
           bf16x8v b0 = read_bfrag<128>(
@@ -726,7 +719,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
           dk_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               da1, b1, dk_acc[t], 0, 0, 0);
         }
-        __builtin_amdgcn_s_setprio(0);
       }
       __syncthreads();
     }
@@ -834,7 +826,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
           sv[r] = 0.f;
           dpv[r] = 0.f;
         }
-        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int c = 0; c < QC; ++c) {
           bf16x8v ka = read_bfrag<D * 2>(
@@ -846,7 +837,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
           dpv = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               va, do_reg[c], dpv, 0, 0, 0);
         }
-        __builtin_amdgcn_s_setprio(0);
 
         f32x16 dst;
         // interior subtile: every (q, kv) pair of this wave is strictly
@@ -887,7 +877,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
           da0 = f0.f;
           da1 = f1.f;
         }
-        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int t = 0; t < DT; ++t) {
           bf16x8v b0 = read_bfrag<128>(
@@ -899,7 +888,6 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dq_kernel(
           dq_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               da1, b1, dq_acc[t], 0, 0, 0);
         }
-        __builtin_amdgcn_s_setprio(0);
       }
     }
     if (more) {
