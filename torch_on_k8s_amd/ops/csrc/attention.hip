@@ -27,8 +27,9 @@
 //    Dsum = rowsum(dO*O) preprocess. All kernels: 0 scratch spill.
 //
 // Measured (B=2 S=4096 Hq=32 Hkv=8 D=128 causal, random data): fwd
-// 358-371 TF (parity with torch's aotriton flash), bwd 194 TF incl. the
-// three operand transposes.
+// 358-371 TF (parity with torch's aotriton flash), bwd 204-206 TF incl.
+// the three operand transposes (r2: uniform-flag interior fast path —
+// the bwd kernels are VALU-overhead-bound, profiles/pmc_attention_r2.csv).
 #include "common.h"
 
 namespace {
