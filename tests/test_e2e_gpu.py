@@ -39,6 +39,9 @@ def test_manager_job_with_preemption_on_gpu(tmp_path):
                 "replicas": 1, "gpusPerTask": 1,
                 "env": {
                     "TOK_TRAIN_STEPS": "12",
+                    # tiny steps are ~ms on GPU: pace them so the
+                    # preemption below deterministically lands mid-run
+                    "TOK_STEP_DELAY": "0.3",
                     "TOK_TRAINER_CONFIG": json.dumps(
                         {"model": "llama-tiny", "micro_batch": 2,
                          "seq_len": 256}),

@@ -248,6 +248,7 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
         if cv:
             last_completed = cv.get("version")
     rejoin_enabled = os.environ.get("TOK_DISABLE_REJOIN") != "1"
+    step_delay = float(os.environ.get("TOK_STEP_DELAY", "0"))
     # benchmark instrumentation (TOK_BENCH_*): timed region bracketed by
     # barrier + device sync on both sides, MAX-elapsed over ranks
     bench_steps = int(os.environ.get("TOK_BENCH_STEPS", "0"))
@@ -292,6 +293,9 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
                     })
             continue
         loss = trainer.train_step()
+        if step_delay > 0:
+            time.sleep(step_delay)  # test hook: deterministic pacing so
+            # e2e tests can interleave control-plane actions mid-run
         if ctx.is_main:
             print(f"[train] step={trainer.step_count} loss={loss:.4f}",
                   flush=True)
