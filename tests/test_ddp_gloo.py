@@ -366,3 +366,50 @@ def test_rccl_env_tuning_flagged(monkeypatch):
     monkeypatch.setenv("NCCL_MIN_NCHANNELS", "64")
     rccl.apply_rccl_env(8)
     assert os.environ["NCCL_MIN_NCHANNELS"] == "64"
+
+
+PERIODIC_WORKER = r"""
+import os, sys
+sys.path.insert(0, os.environ["TOK_ROOT"])
+import torch
+from torch_on_k8s_amd.parallel.env import init_distributed, destroy
+from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
+
+ctx = init_distributed(backend="gloo")
+cfg = TrainerConfig(model="llama-tiny", micro_batch=1, seq_len=32)
+tr = Trainer(cfg, ctx)
+writers = []
+for _ in range(4):
+    tr.train_step()
+    if tr.step_count % 2 == 0:
+        writers.append(tr.snapshot_checkpoint_async(os.environ["TOK_CKPT"]))
+for w in writers:
+    w.join(timeout=120)
+    assert not w.is_alive()
+destroy()
+"""
+
+
+def test_periodic_async_sharded_snapshot(tmp_path):
+    """2-rank async sharded snapshots: every shard lands, rank 0
+    publishes atomically, and the result loads at world 1."""
+    ck = str(tmp_path / "ckpt")
+    procs = []
+    env0 = dict(os.environ, TOK_ROOT=ROOT, TOK_CKPT=ck,
+                MASTER_ADDR="127.0.0.1", MASTER_PORT="29723",
+                WORLD_SIZE="2")
+    for r in range(2):
+        env = dict(env0, RANK=str(r), LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen(
+            [sys.executable, "-c", PERIODIC_WORKER], env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
+    for p in procs:
+        out, err = p.communicate(timeout=300)
+        assert p.returncode == 0, f"worker failed:\n{out}\n{err}"
+    # loads into a single-rank trainer (bucket-indexed shards)
+    from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
+    from torch_on_k8s_amd.parallel.env import DistContext
+    tr = Trainer(TrainerConfig(model="llama-tiny", micro_batch=1,
+                               seq_len=32), DistContext())
+    tr.load_checkpoint(ck)
+    assert tr.step_count == 4

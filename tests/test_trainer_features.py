@@ -73,3 +73,43 @@ def test_sharded_checkpoint_roundtrip(tmp_path):
     l1 = tr.train_step()
     l2 = tr2.train_step()
     assert abs(l1 - l2) < 1e-6
+
+
+def test_async_snapshot_resumable(tmp_path):
+    """snapshot_checkpoint_async: training continues while the writer
+    serializes; the published checkpoint is complete, atomic, and
+    resumable at the snapshot step."""
+    import torch
+    from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
+    from torch_on_k8s_amd.parallel.env import DistContext
+    torch.manual_seed(0)
+    tr = Trainer(TrainerConfig(model="llama-tiny", micro_batch=1,
+                               seq_len=32), DistContext())
+    tr.train_step()
+    tr.train_step()
+    ck = str(tmp_path / "ckpt")
+    t = tr.snapshot_checkpoint_async(ck)
+    # training continues immediately (mutates weights AFTER the snapshot)
+    tr.train_step()
+    t.join(timeout=120)
+    assert not t.is_alive()
+    import os as _os
+    assert _os.path.exists(_os.path.join(ck, "meta.json"))
+    assert not any(".tmp-" in d for d in _os.listdir(str(tmp_path)))
+
+    torch.manual_seed(0)
+    tr2 = Trainer(TrainerConfig(model="llama-tiny", micro_batch=1,
+                                seq_len=32), DistContext())
+    tr2.load_checkpoint(ck)
+    assert tr2.step_count == 2  # the snapshot step, not the later one
+    # weights equal the state at snapshot time: one more step from the
+    # restore must match a fresh run's third step bit-for-bit
+    torch.manual_seed(0)
+    ref = Trainer(TrainerConfig(model="llama-tiny", micro_batch=1,
+                                seq_len=32), DistContext())
+    ref.train_step()
+    ref.train_step()
+    ref.train_step()
+    tr2.train_step()
+    for b2, br in zip(tr2.fb.buckets, ref.fb.buckets):
+        assert torch.equal(b2.flat_param, br.flat_param)

@@ -243,6 +243,92 @@ class Trainer:
             json.dump(rec, f)
         os.replace(tmp, self.cfg.metrics_path)
 
+    # ---- async periodic checkpoints ----------------------------------
+    # Two phases: a SHORT blocking snapshot (device -> host copies at
+    # PCIe rate) at a step boundary, then a background thread serializes
+    # to disk while training continues. Sharded: every rank snapshots at
+    # the SAME step (step % N == 0 needs no coordination — steps are
+    # lockstep); each writer drops its shard + a .done marker and rank 0's
+    # writer performs the atomic rename once all markers exist, so the
+    # background phase needs no collectives.
+    def snapshot_checkpoint_async(self, path: str) -> "threading.Thread":
+        import threading
+        world = self.fb.world_size
+        sharded = world > 1
+        tmp = f"{path}.tmp-{self.step_count}"
+        os.makedirs(tmp, exist_ok=True)
+
+        # phase 1 (blocking, short): host snapshot
+        msd = None
+        meta = None
+        if self.ctx.is_main:
+            msd = {k: v.detach().to("cpu", non_blocking=True).clone()
+                   for k, v in self.module.state_dict().items()}
+            meta = {
+                "step": self.step_count,
+                "model": self.cfg.model,
+                "model_config": self.model_cfg.to_dict(),
+                "trainer_config": self.cfg.to_dict(),
+                "optim_format": "sharded" if sharded else "single",
+                "optim_shards": world if sharded else 1,
+                "num_buckets": len(self.fb.buckets),
+            }
+        osd = self.opt.state_dict()
+        if sharded:
+            shard = {
+                "step": osd["step"],
+                "exp_avg": {i: t.to("cpu", non_blocking=True)
+                            for i, t in enumerate(osd["exp_avg"])
+                            if i % world == self.ctx.rank},
+                "exp_avg_sq": {i: t.to("cpu", non_blocking=True)
+                               for i, t in enumerate(osd["exp_avg_sq"])
+                               if i % world == self.ctx.rank},
+            }
+        else:
+            shard = {
+                "step": osd["step"],
+                "exp_avg": [t.to("cpu", non_blocking=True)
+                            for t in osd["exp_avg"]],
+                "exp_avg_sq": [t.to("cpu", non_blocking=True)
+                               for t in osd["exp_avg_sq"]],
+            }
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()  # host buffers now stable
+
+        rank, is_main = self.ctx.rank, self.ctx.is_main
+
+        def writer():
+            if is_main:
+                torch.save(msd, os.path.join(tmp, "model.pt"))
+                with open(os.path.join(tmp, "meta.json"), "w") as f:
+                    json.dump(meta, f, indent=2)
+            if sharded:
+                torch.save(shard,
+                           os.path.join(tmp, f"optim-shard-{rank}.pt"))
+            elif is_main:
+                torch.save(shard, os.path.join(tmp, "optim.pt"))
+            open(os.path.join(tmp, f".done-{rank}"), "w").close()
+            if is_main:
+                # wait for every shard, then atomically publish
+                want = {f".done-{r}" for r in range(world if sharded else 1)}
+                deadline = time.time() + 600
+                while time.time() < deadline:
+                    if want <= set(os.listdir(tmp)):
+                        break
+                    time.sleep(0.05)
+                else:
+                    return  # incomplete; leave tmp for inspection
+                for m in want:
+                    os.unlink(os.path.join(tmp, m))
+                if os.path.exists(path):
+                    import shutil
+                    shutil.rmtree(path)
+                os.replace(tmp, path)
+
+        t = threading.Thread(target=writer, daemon=True)
+        t.start()
+        return t
+
     # ---- checkpoint / resume (elastic protocol) ------
     def save_checkpoint(self, path: str, sharded: bool | None = None):
         """Atomic checkpoint.

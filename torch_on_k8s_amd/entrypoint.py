@@ -249,6 +249,12 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
             last_completed = cv.get("version")
     rejoin_enabled = os.environ.get("TOK_DISABLE_REJOIN") != "1"
     step_delay = float(os.environ.get("TOK_STEP_DELAY", "0"))
+    # periodic async checkpoints (production safety net on top of the
+    # elastic/SIGTERM event-driven ones): short host snapshot at the
+    # step boundary, disk serialization in the background
+    ckpt_every = int(os.environ.get("TOK_CKPT_EVERY", "0"))
+    ckpt_dir = os.path.join(state_dir, "ckpt") if state_dir else None
+    ckpt_writer = None
     # benchmark instrumentation (TOK_BENCH_*): timed region bracketed by
     # barrier + device sync on both sides, MAX-elapsed over ranks
     bench_steps = int(os.environ.get("TOK_BENCH_STEPS", "0"))
@@ -296,6 +302,12 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
         if step_delay > 0:
             time.sleep(step_delay)  # test hook: deterministic pacing so
             # e2e tests can interleave control-plane actions mid-run
+        if ckpt_every and ckpt_dir and \
+                trainer.step_count % ckpt_every == 0:
+            # deterministic across ranks (steps are lockstep); keep one
+            # snapshot in flight — skip the interval if still writing
+            if ckpt_writer is None or not ckpt_writer.is_alive():
+                ckpt_writer = trainer.snapshot_checkpoint_async(ckpt_dir)
         if ctx.is_main:
             print(f"[train] step={trainer.step_count} loss={loss:.4f}",
                   flush=True)
@@ -326,6 +338,11 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
         stop_now = bool(int(coord[1])) or \
             (not ctx.is_distributed and stop["sig"] is not None)
         scale_now = bool(int(coord[2])) and reqv
+        if (reqv or stop_now) and ckpt_writer is not None and \
+                ckpt_writer.is_alive():
+            # a synchronous (elastic/SIGTERM) save is about to touch the
+            # same checkpoint path: let the background writer finish
+            ckpt_writer.join(timeout=600)
         if reqv:
             checkpoint_and_ack(reqv)
             last_completed = reqv
