@@ -140,3 +140,14 @@ def test_training_metrics_gauges():
         m.remove_job("j1")
         assert REGISTRY.get_sample_value(
             "torch_on_k8s_amd_job_tokens_per_second", {"job": "j1"}) is None
+
+
+def test_client_logs(tmp_path):
+    from torch_on_k8s_amd.client import TorchJobClient
+    d = tmp_path / "jobs" / "j1"
+    d.mkdir(parents=True)
+    (d / "j1-master-0.log").write_text("line1\nline2\nline3\n")
+    cli = TorchJobClient(str(tmp_path))
+    assert "line2" in cli.logs("j1")
+    assert cli.logs("j1", tail=1).strip() == "line3"
+    assert cli.logs("nope") == ""

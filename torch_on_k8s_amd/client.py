@@ -66,6 +66,21 @@ class TorchJobClient:
         raise TimeoutError(f"job {name} did not reach {phases}; "
                            f"last status: {self.get(name)}")
 
+    def logs(self, name: str, task: str = "master", index: int = 0,
+             tail: int | None = None) -> str:
+        """Task process log (kubectl logs analog; the runtime writes
+        one file per task under the job state dir)."""
+        path = os.path.join(self.workdir, "jobs", name,
+                            f"{name}-{task}-{index}.log")
+        try:
+            with open(path) as f:
+                text = f.read()
+        except OSError:
+            return ""
+        if tail is not None:
+            text = "\n".join(text.splitlines()[-tail:])
+        return text
+
     # -- model registry view (Model/ModelVersion read API analog) ------
     def _registry(self):
         from torch_on_k8s_amd.controlplane.modelregistry import (
@@ -111,6 +126,11 @@ def main():
         pc = sub.add_parser(c)
         pc.add_argument("name")
     sub.add_parser("list")
+    p_logs = sub.add_parser("logs")
+    p_logs.add_argument("name")
+    p_logs.add_argument("--task", default="master")
+    p_logs.add_argument("--index", type=int, default=0)
+    p_logs.add_argument("--tail", type=int, default=None)
     sub.add_parser("models")
     p_ex = sub.add_parser("extract")
     p_ex.add_argument("ref", help="model[:version] (default: latest)")
@@ -135,6 +155,9 @@ def main():
         for n in cli.list():
             st = cli.get(n) or {}
             print(f"{n:30s} {st.get('phase')}")
+    elif args.cmd == "logs":
+        print(cli.logs(args.name, task=args.task, index=args.index,
+                       tail=args.tail))
     elif args.cmd == "models":
         print(json.dumps(cli.list_models(), indent=2))
     elif args.cmd == "extract":
