@@ -171,8 +171,8 @@ def _poll_file(path, pred, timeout_s=120.0, period=0.05):
     raise TimeoutError(f"rejoin: timed out waiting on {path}")
 
 
-def _fast_rejoin(trainer, ctx, job_doc, agent_file, write_agent,
-                 version):
+def _fast_rejoin(trainer, ctx, new_world, worker_replicas, agent_file,
+                 write_agent, version):
     """Elastic fast-rejoin (scale event without process restart): keep
     model/optimizer state resident, tear down only the process group and
     re-init it at the new world size. Protocol (r1 VERDICT next-#9; the
@@ -193,9 +193,9 @@ def _fast_rejoin(trainer, ctx, job_doc, agent_file, write_agent,
 
     ttype = os.environ.get("TOK_TASK_TYPE", "master")
     tindex = int(os.environ.get("TOK_TASK_INDEX", "0"))
-    replicas = job_doc.get("replicas", {})
-    new_world = sum(int(v) for t, v in replicas.items() if t != "aimaster")
-    survivor = tindex < int(replicas.get(ttype, 0))
+    # survivorship from the BROADCAST values (a per-rank job.json
+    # re-read could race a second spec update and fork the gang)
+    survivor = (tindex < worker_replicas) if ttype == "worker" else True
 
     if dist.is_initialized():
         dist.destroy_process_group()
@@ -214,7 +214,9 @@ def _fast_rejoin(trainer, ctx, job_doc, agent_file, write_agent,
         _poll_file(agent_file, lambda d: d.get("rejoin-ready") == version,
                    timeout_s=timeout_s)
 
-    rank = 0 if ttype == "master" else tindex + 1
+    # a survivor's rank never changes across scale events (its index and
+    # the master's presence are fixed), so the original env RANK holds
+    rank = int(os.environ.get("RANK", "0"))
     dist.init_process_group(
         backend=ctx.backend or ("nccl" if ctx.device.type == "cuda"
                                 else "gloo"),
@@ -268,8 +270,8 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
     # coordination word broadcast from rank 0 each step so every rank
     # takes checkpoint/stop decisions at the SAME step (sharded saves
     # are collective): [requested ckpt version or 0, stop flag,
-    # rejoin flag (scale event: world size is changing)]
-    coord = torch.zeros(3, dtype=torch.long, device=ctx.device)
+    # new world size (0 = no scale), new worker replicas]
+    coord = torch.zeros(4, dtype=torch.long, device=ctx.device)
     while trainer.step_count < steps_total:
         if bench_steps:
             # benchmark mode: every timed step is EXACTLY a bare-bench
@@ -330,8 +332,10 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
             replicas = jd.get("replicas", {})
             new_world = sum(int(v) for t, v in replicas.items()
                             if t != "aimaster")
-            coord[2] = 1 if (rejoin_enabled and replicas and
-                             new_world != ctx.world_size) else 0
+            scale = (rejoin_enabled and replicas and
+                     new_world != ctx.world_size)
+            coord[2] = new_world if scale else 0
+            coord[3] = int(replicas.get("worker", 0))
         if ctx.is_distributed and dist.is_initialized():
             dist.broadcast(coord, src=0)
         reqv = int(coord[0])
@@ -347,8 +351,8 @@ def _train_loop(trainer, ctx, steps_total, state_dir, job_file, agent_file,
             checkpoint_and_ack(reqv)
             last_completed = reqv
         if scale_now:
-            jd = _read_json(job_file) or {}
-            new_world = _fast_rejoin(trainer, ctx, jd, agent_file,
+            new_world = _fast_rejoin(trainer, ctx, int(coord[2]),
+                                     int(coord[3]), agent_file,
                                      write_agent, reqv)
             if new_world is None:
                 return 0  # scale-in victim: clean exit, controller reaps
