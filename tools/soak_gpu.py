@@ -29,6 +29,9 @@ def main():
     ap.add_argument("--micro-batch", type=int, default=8)
     ap.add_argument("--seq-len", type=int, default=4096)
     ap.add_argument("--hip-graph", action="store_true", default=True)
+    ap.add_argument("--ckpt-every", type=int, default=0,
+                    help="async periodic snapshot every N steps")
+    ap.add_argument("--ckpt-dir", default="/tmp/soak-ckpt")
     args = ap.parse_args()
 
     torch.manual_seed(7)
@@ -39,13 +42,23 @@ def main():
 
     losses = []
     mem_marks = {}
+    writer = None
+    snapshots = 0
     for i in range(args.steps):
         loss = float(tr.train_step())
         assert loss == loss and abs(loss) < 1e4, f"loss blew up: {loss}"
         losses.append(loss)
+        if args.ckpt_every and (i + 1) % args.ckpt_every == 0 and \
+                (writer is None or not writer.is_alive()):
+            writer = tr.snapshot_checkpoint_async(args.ckpt_dir)
+            snapshots += 1
         if i in (20, args.steps - 1):
             torch.cuda.synchronize()
             mem_marks[i] = torch.cuda.memory_allocated()
+    if writer is not None:
+        writer.join(timeout=600)
+        # the published checkpoint must resume
+        tr.load_checkpoint(args.ckpt_dir)
     first = sum(losses[:10]) / 10
     last = sum(losses[-10:]) / 10
     growth = mem_marks[args.steps - 1] - mem_marks[20]
@@ -57,6 +70,7 @@ def main():
         "mem_at_end_gb": round(mem_marks[args.steps - 1] / 2**30, 2),
         "mem_growth_mb": round(growth / 2**20, 2),
         "max_mem_gb": round(torch.cuda.max_memory_allocated() / 2**30, 2),
+        "async_snapshots": snapshots,
     }
     print(json.dumps(out))
     assert last < first, "loss did not decrease over the soak"
