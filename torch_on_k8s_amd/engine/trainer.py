@@ -389,16 +389,22 @@ class Trainer:
             barrier(self.ctx)  # no rank resumes before the rename lands
 
     def load_checkpoint(self, path: str):
-        sd = torch.load(os.path.join(path, "model.pt"), map_location=self.device,
-                        weights_only=True)
+        # map_location CPU: tensors stream host->device via copy_ into
+        # the EXISTING buffers, so resume never doubles the device
+        # footprint (loading ~75 GB of optimizer state with
+        # map_location=cuda OOMs a trainer that is already resident —
+        # found by the r2 GPU soak)
+        sd = torch.load(os.path.join(path, "model.pt"),
+                        map_location="cpu", weights_only=True)
         self.module.load_state_dict(sd)
+        del sd
         with open(os.path.join(path, "meta.json")) as f:
             meta = json.load(f)
         self.step_count = meta["step"]
         nshards = meta.get("optim_shards", 1)
         if meta.get("optim_format", "single") == "single":
             osd = torch.load(os.path.join(path, "optim.pt"),
-                             map_location=self.device, weights_only=True)
+                             map_location="cpu", weights_only=True)
             self.opt.load_state_dict(osd)
             return
         # sharded: every rank reads all shards (bucket-indexed, world-
@@ -408,7 +414,7 @@ class Trainer:
             self.opt.step_dev.fill_(self.step_count)
         for r in range(nshards):
             shard = torch.load(os.path.join(path, f"optim-shard-{r}.pt"),
-                               map_location=self.device, weights_only=True)
+                               map_location="cpu", weights_only=True)
             self.opt.step_count = shard["step"]
             if self.opt.step_dev is not None:
                 self.opt.step_dev.fill_(shard["step"])
