@@ -262,7 +262,10 @@ class Trainer:
         msd = None
         meta = None
         if self.ctx.is_main:
-            msd = {k: v.detach().to("cpu", non_blocking=True).clone()
+            # copy=True: on a CPU trainer .to("cpu") would ALIAS the
+            # live tensors and the background writer would serialize
+            # state the next step is mutating (torn snapshot)
+            msd = {k: v.detach().to("cpu", non_blocking=True, copy=True)
                    for k, v in self.module.state_dict().items()}
             meta = {
                 "step": self.step_count,
@@ -277,19 +280,20 @@ class Trainer:
         if sharded:
             shard = {
                 "step": osd["step"],
-                "exp_avg": {i: t.to("cpu", non_blocking=True)
+                "exp_avg": {i: t.to("cpu", non_blocking=True, copy=True)
                             for i, t in enumerate(osd["exp_avg"])
                             if i % world == self.ctx.rank},
-                "exp_avg_sq": {i: t.to("cpu", non_blocking=True)
+                "exp_avg_sq": {i: t.to("cpu", non_blocking=True,
+                                       copy=True)
                                for i, t in enumerate(osd["exp_avg_sq"])
                                if i % world == self.ctx.rank},
             }
         else:
             shard = {
                 "step": osd["step"],
-                "exp_avg": [t.to("cpu", non_blocking=True)
+                "exp_avg": [t.to("cpu", non_blocking=True, copy=True)
                             for t in osd["exp_avg"]],
-                "exp_avg_sq": [t.to("cpu", non_blocking=True)
+                "exp_avg_sq": [t.to("cpu", non_blocking=True, copy=True)
                                for t in osd["exp_avg_sq"]],
             }
         if self.device.type == "cuda":
