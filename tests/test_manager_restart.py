@@ -116,3 +116,45 @@ def test_adopted_exit_marker_grades_completion(tmp_path):
     h2.phase = TaskPhase.RUNNING
     rt.poll(h2)
     assert h2.phase == TaskPhase.FAILED and h2.reason == "AdoptedExit"
+
+
+def test_manager_graceful_shutdown_keeps_tasks(tmp_path):
+    """SIGTERM to the manager daemon: it exits cleanly WITHOUT killing
+    running gangs (the next manager adopts them)."""
+    import signal
+    import subprocess
+    import sys
+    mgr = subprocess.Popen(
+        [sys.executable, "-m", "torch_on_k8s_amd.manager",
+         "--workdir", str(tmp_path), "--num-gpus", "0",
+         "--metrics-addr", "0", "--sync-period", "0.05"],
+        env=dict(os.environ, PYTHONPATH=ROOT),
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+    try:
+        spec = ("kind: TorchJob\nmetadata: {name: gj}\nspec:\n  tasks:\n"
+                "    master: {replicas: 1, gpusPerTask: 0, env: {"
+                "TOK_BACKEND: gloo, TOK_TRAIN_STEPS: '500', "
+                "TOK_STEP_DELAY: '0.2', TOK_TRAINER_CONFIG: '"
+                '{"model": "llama-tiny", "micro_batch": 1, "seq_len": 32}'
+                "'}}\n")
+        os.makedirs(tmp_path / "spool", exist_ok=True)
+        with open(tmp_path / "spool" / "gj.yaml", "w") as f:
+            f.write(spec)
+        # wait for the task pid record
+        rec = tmp_path / "jobs" / "gj" / "tasks" / "gj-master-0.json"
+        t0 = time.time()
+        while time.time() - t0 < 60 and not rec.exists():
+            time.sleep(0.2)
+        assert rec.exists(), mgr.stdout
+        task_pid = json.load(open(rec))["pid"]
+        os.kill(mgr.pid, signal.SIGTERM)
+        assert mgr.wait(timeout=30) == 0
+        # task survived the manager
+        os.kill(task_pid, 0)
+    finally:
+        if mgr.poll() is None:
+            mgr.kill()
+        try:
+            os.kill(json.load(open(rec))["pid"], signal.SIGKILL)
+        except (OSError, ValueError, FileNotFoundError):
+            pass

@@ -111,3 +111,14 @@ def test_registry_reindex(tmp_path):
     assert reg2.reindex() == 2
     assert reg2.models["m"].latest_version == "v2"
     assert reg2.get_version("m", "v1").build_phase == "Succeeded"
+
+
+def test_serve_metrics_endpoint(trained_ckpt):
+    srv = InferenceServer.from_checkpoint(trained_ckpt, "cpu")
+    c = _client(build_app(srv))
+    c.post("/v1/generate", json={"prompt_ids": [[1, 2]],
+                                 "max_new_tokens": 3})
+    r = c.get("/metrics")
+    assert r.status_code == 200
+    assert "tok_serve_requests_total 1" in r.text
+    assert "tok_serve_tokens_out_total 3" in r.text

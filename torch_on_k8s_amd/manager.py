@@ -184,9 +184,24 @@ class Manager:
         self.publish_status()
 
     def run_forever(self):
-        while True:
+        """Main loop with graceful shutdown: on SIGTERM/SIGINT, publish
+        final statuses and exit 0 WITHOUT killing task processes — a
+        restarted manager adopts them (controller._adopt_orphans), so a
+        manager upgrade never interrupts running gangs."""
+        import signal as _signal
+        stop = {"flag": False}
+
+        def _stop(*_):
+            stop["flag"] = True
+
+        _signal.signal(_signal.SIGTERM, _stop)
+        _signal.signal(_signal.SIGINT, _stop)
+        while not stop["flag"]:
             self.step()
             time.sleep(self.sync_period)
+        self.publish_status()
+        log.info("manager shutting down; %d job(s) keep running for "
+                 "adoption by the next manager", len(self.controller.jobs))
 
 
 def main():

@@ -149,6 +149,18 @@ def build_app(srv: InferenceServer):
                 "trained_steps": srv.meta.get("step"),
                 "config": srv.meta.get("model_config")}
 
+    @app.get("/metrics")
+    def metrics():
+        """Prometheus exposition (serving-side observability)."""
+        from fastapi.responses import PlainTextResponse
+        lines = [
+            "# TYPE tok_serve_requests_total counter",
+            f"tok_serve_requests_total {srv.requests}",
+            "# TYPE tok_serve_tokens_out_total counter",
+            f"tok_serve_tokens_out_total {srv.tokens_out}",
+        ]
+        return PlainTextResponse("\n".join(lines) + "\n")
+
     @app.post("/v1/generate")
     def generate(req: GenerateRequest):
         vocab = srv.meta.get("model_config", {}).get("vocab_size")
