@@ -192,3 +192,28 @@ def test_soak_no_unbounded_growth():
     assert not coord._index
     # every GPU slot returned
     assert len(node.free_slots) == 8
+
+
+def test_adopted_job_bypasses_admission_queue():
+    """Manager-restart wedge regression: an adopted job was already
+    admitted in its previous life — re-queueing it would deadlock (the
+    quota filter counts its OWN live GPUs against the tenant)."""
+    node = NodeState(num_gpus=8)
+    rt = FakeRuntime()
+    ctl = JobController(node, rt, ControllerConfig())
+    coord = Coordinator(dequeue_fn=ctl.reconcile,
+                        tenant_usage_fn=ctl.tenant_resource_usage,
+                        quotas={"t": 8}, default_quota=8)
+    ctl.coordinator = coord
+    # a previous manager left an 8-GPU gang running (full tenant quota)
+    rt.adoptable["big"] = [
+        {"pid": 1, "task_type": "master", "index": 0, "generation": 1,
+         "gpu_slots": [0]},
+    ] + [{"pid": 2 + i, "task_type": "worker", "index": i, "generation": 1,
+          "gpu_slots": [1 + i]} for i in range(7)]
+    job = ctl.create_job(mk_job("big", workers=7))
+    # adopted straight into Running territory, never queued
+    assert not coord.is_queuing(job.uid)
+    assert len(ctl.handles["big"]) == 8
+    assert len(node.free_slots) == 0
+    assert any(e.reason == "TaskAdopted" for e in ctl.events)

@@ -111,13 +111,16 @@ class JobController:
         set_defaults(job)
         self.jobs[job.name] = job
         self.handles.setdefault(job.name, {})
-        self._adopt_orphans(job)
+        adopted = self._adopt_orphans(job)
         job.status.set_condition(JobConditionType.CREATED, "JobCreated")
         if self.metrics:
             self.metrics.created()
-        if self.coordinator is not None:
+        if self.coordinator is not None and not adopted:
             self.coordinator.enqueue_or_update(job)
         else:
+            # adopted jobs were admitted in their previous life: going
+            # through the queue again would wedge them — the quota
+            # filter counts their own live GPUs against the tenant
             self.reconcile(job)
         return job
 
@@ -130,27 +133,31 @@ class JobController:
         if self.metrics:
             self.metrics.deleted()
 
-    def _adopt_orphans(self, job: TorchJob):
+    def _adopt_orphans(self, job: TorchJob) -> int:
         """Manager-restart durability: rebuild handles over task
         processes a previous manager left running (their pid records
         live in the job state dir), claim their GPU slots, and restore
         the job's persisted generation/annotations/master port so the
-        gang is neither lost, duplicated, nor spuriously scaled."""
+        gang is neither lost, duplicated, nor spuriously scaled.
+        Returns the number of adopted tasks."""
         rt = self.runtime
         if not hasattr(rt, "adoptable_tasks"):
-            return
+            return 0
         recs = rt.adoptable_tasks(job.name)
         if not recs:
-            return
+            return 0
         # restore the controller's persisted job view (job.json): the
         # running tasks were launched against THAT generation and port
         import json
         import os
-        try:
-            with open(os.path.join(rt.workdir, job.name, "job.json")) as f:
-                saved = json.load(f)
-        except (OSError, ValueError):
-            saved = {}
+        saved = {}
+        workdir = getattr(rt, "workdir", None)
+        if workdir:
+            try:
+                with open(os.path.join(workdir, job.name, "job.json")) as f:
+                    saved = json.load(f)
+            except (OSError, ValueError):
+                pass
         if saved.get("generation"):
             job.generation = max(job.generation, int(saved["generation"]))
         for k, v in (saved.get("annotations") or {}).items():
@@ -168,6 +175,7 @@ class JobController:
             hs[h.key] = h
             self.event(job.name, "Normal", "TaskAdopted",
                        f"{t.value}-{idx} pid {h.pid} on GPUs {h.gpu_slots}")
+        return len(recs)
 
     def _master_port(self, job: TorchJob) -> int:
         """Per-job master port: RANDOM start within the configured range

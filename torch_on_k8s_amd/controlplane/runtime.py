@@ -103,6 +103,8 @@ class FakeRuntime(Runtime):
         self.killed: list = []
         # default behavior: tasks move PENDING->RUNNING on first poll
         self.auto_run = True
+        # adoption surface (tests seed job -> [task records])
+        self.adoptable: dict = {}
 
     def start_task(self, job, t, index, gpu_slots, extra_env):
         h = TaskHandle(job.name, t, index, gpu_slots=tuple(gpu_slots),
@@ -130,6 +132,18 @@ class FakeRuntime(Runtime):
         h.phase = phase
         h.exit_code = exit_code
         h.reason = reason
+
+    def adoptable_tasks(self, job_name: str) -> list:
+        return list(self.adoptable.get(job_name, []))
+
+    def adopt_task(self, job, t, index, rec) -> TaskHandle:
+        h = TaskHandle(job.name, t, index,
+                       gpu_slots=tuple(rec.get("gpu_slots") or ()),
+                       generation=int(rec.get("generation", job.generation)))
+        h.pid = int(rec.get("pid", 0))
+        h.phase = TaskPhase.RUNNING
+        self.tasks[h.key] = h
+        return h
 
 
 class LocalProcessRuntime(Runtime):
