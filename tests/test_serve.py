@@ -122,3 +122,22 @@ def test_serve_metrics_endpoint(trained_ckpt):
     assert r.status_code == 200
     assert "tok_serve_requests_total 1" in r.text
     assert "tok_serve_tokens_out_total 3" in r.text
+
+
+@pytest.mark.gpu
+def test_serve_on_gpu():
+    """Serving path on metal: bf16 model, flash-decode + hipGraph."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    srv = InferenceServer.from_preset("llama-tiny", "cuda", torch.bfloat16)
+    c = _client(build_app(srv))
+    r = c.post("/v1/generate", json={"prompt_ids": [[1, 2, 3, 4]] * 4,
+                                     "max_new_tokens": 16})
+    assert r.status_code == 200, r.text
+    out = r.json()
+    assert len(out["output_ids"]) == 4
+    assert all(len(o) == 20 for o in out["output_ids"])
+    # deterministic greedy decode on the graph path
+    r2 = c.post("/v1/generate", json={"prompt_ids": [[1, 2, 3, 4]] * 4,
+                                      "max_new_tokens": 16})
+    assert r2.json()["output_ids"] == out["output_ids"]
