@@ -438,8 +438,12 @@ DEVINL void repack_pa(const f32x16& pt, bf16x8v& pa0, bf16x8v& pa1) {
   pa1 = f1.f;
 }
 
-template <int D, bool CAUSAL>
-__global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
+// NWT (waves per block) A/B: 8 = one block/CU (lockstep phases);
+// 4 = TWO resident blocks/CU whose barrier phases drift, so each SIMD
+// hosts one wave of each block and staging overlaps the partner
+// block's MFMA segments (guide's compute/load wave-pair regime).
+template <int D, bool CAUSAL, int NWT = NW>
+__global__ __launch_bounds__(NWT * WAVE) void attn_bwd_dv_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ dot_t, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ dsum,
@@ -447,7 +451,8 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
     float scale) {
   constexpr int QC = D / 16;
   constexpr int DT = D / 32;
-  constexpr int BNK = NW * 32;   // 256 kv rows per block
+  constexpr int NTH = NWT * WAVE;
+  constexpr int BNK = NWT * 32;  // kv rows per block
   constexpr int KB = BN * D * 2;
   // double-buffered (Q rm | dOT): one barrier per q tile
   __shared__ __attribute__((aligned(16))) char smem[2 * (2 * KB)];
@@ -483,8 +488,8 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
 
   const int m_start = CAUSAL ? (n0 / BN) * BN : 0;
   const long dplane = (long)D * S_pad;
-  TileStage<D> q_st;
-  TileStageT<D> do_st;
+  TileStage<D, NTH> q_st;
+  TileStageT<D, NTH> do_st;
   const bf16_t* dot0 = dot_t + ((long)b * Hq + hkv * rep) * dplane;
   q_st.issue(q + ((long)b * S * q_tok) + (long)(hkv * rep) * D, m_start, S,
              q_tok);
@@ -581,8 +586,8 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dv_kernel(
   }
 }
 
-template <int D, bool CAUSAL>
-__global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
+template <int D, bool CAUSAL, int NWT = NW>
+__global__ __launch_bounds__(NWT * WAVE) void attn_bwd_dk_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const bf16_t* __restrict__ q_t, const float* __restrict__ lse,
@@ -590,7 +595,8 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
     int Hq, int Hkv, int S_pad, float scale) {
   constexpr int QC = D / 16;
   constexpr int DT = D / 32;
-  constexpr int BNK = NW * 32;
+  constexpr int NTH = NWT * WAVE;
+  constexpr int BNK = NWT * 32;
   constexpr int KB = BN * D * 2;
   // Q rm | QT | dO rm | V block tile (staged once; keeping V in
   // registers alongside K pushed the kernel to 256 VGPR + scratch spill)
@@ -626,7 +632,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
   {  // stage the block's 256-row V tile (row-major, swizzled) once
     constexpr int VPR = D / 8;
 #pragma unroll 4
-    for (int vi = threadIdx.x; vi < BNK * VPR; vi += NTHREADS) {
+    for (int vi = threadIdx.x; vi < BNK * VPR; vi += NTH) {
       const int row = vi / VPR, cv = vi % VPR;
       uint4 val = {0, 0, 0, 0};
       if (n0 + row < S)
@@ -654,8 +660,8 @@ __global__ __launch_bounds__(NTHREADS) void attn_bwd_dk_kernel(
     for (int m0 = m_start; m0 < S; m0 += BN) {
       {  // synchronous staging: the prefetch ring costs ~16 VGPRs and
          // tips this kernel into scratch spill, which is worse
-        TileStage<D> q_st, do_st;
-        TileStageT<D> qt_st;
+        TileStage<D, NTH> q_st, do_st;
+        TileStageT<D, NTH> qt_st;
         q_st.issue(qp, m0, S, q_tok);
         do_st.issue(dop, m0, S, q_tok);
         qt_st.issue(qtp, m0, S_pad);
@@ -955,17 +961,35 @@ hipError_t tok_attn_bwd(const void* q, const void* k, const void* v,
     return hipErrorInvalidValue;
 
   dim3 gkv((S + NW * 32 - 1) / (NW * 32), Hkv, B);
+  dim3 gkv4((S + 4 * 32 - 1) / (4 * 32), Hkv, B);
   dim3 gq((S + NW * 32 - 1) / (NW * 32), Hq, B);
+  // dv/dk block-shape A/B: TOK_BWD_NW4=1 runs 4-wave blocks (two
+  // resident per CU -> cross-block phase drift on each SIMD)
+  static const bool nw4 = [] {
+    const char* e = getenv("TOK_BWD_NW4");
+    return e && e[0] == '1';
+  }();
 #define LAUNCH_BWD(DD, CC)                                                    \
   do {                                                                        \
-    attn_bwd_dv_kernel<DD, CC><<<gkv, NTHREADS, 0, stream>>>(                 \
-        (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)dot_t,             \
-        (const bf16_t*)dout, lse, dsum_ws, (bf16_t*)dv, B, S, Hq, Hkv,       \
-        S_pad, scale);                                                        \
-    attn_bwd_dk_kernel<DD, CC><<<gkv, NTHREADS, 0, stream>>>(                 \
-        (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,                 \
-        (const bf16_t*)dout, (const bf16_t*)q_t, lse, dsum_ws, (bf16_t*)dk,  \
-        B, S, Hq, Hkv, S_pad, scale);                                         \
+    if (nw4) {                                                                \
+      attn_bwd_dv_kernel<DD, CC, 4><<<gkv4, 4 * WAVE, 0, stream>>>(           \
+          (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)dot_t,           \
+          (const bf16_t*)dout, lse, dsum_ws, (bf16_t*)dv, B, S, Hq, Hkv,     \
+          S_pad, scale);                                                      \
+      attn_bwd_dk_kernel<DD, CC, 4><<<gkv4, 4 * WAVE, 0, stream>>>(           \
+          (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,               \
+          (const bf16_t*)dout, (const bf16_t*)q_t, lse, dsum_ws,              \
+          (bf16_t*)dk, B, S, Hq, Hkv, S_pad, scale);                          \
+    } else {                                                                  \
+      attn_bwd_dv_kernel<DD, CC><<<gkv, NTHREADS, 0, stream>>>(               \
+          (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)dot_t,           \
+          (const bf16_t*)dout, lse, dsum_ws, (bf16_t*)dv, B, S, Hq, Hkv,     \
+          S_pad, scale);                                                      \
+      attn_bwd_dk_kernel<DD, CC><<<gkv, NTHREADS, 0, stream>>>(               \
+          (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,               \
+          (const bf16_t*)dout, (const bf16_t*)q_t, lse, dsum_ws,              \
+          (bf16_t*)dk, B, S, Hq, Hkv, S_pad, scale);                          \
+    }                                                                         \
     attn_bwd_dq_kernel<DD, CC><<<gq, NTHREADS, 0, stream>>>(                  \
         (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,                 \
         (const bf16_t*)dout, (const bf16_t*)k_t, lse, dsum_ws, (bf16_t*)dq,  \
