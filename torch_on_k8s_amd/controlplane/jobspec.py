@@ -19,7 +19,8 @@ TWO dialects are accepted (auto-detected per document):
       tasks:
         master: {replicas: 1, gpusPerTask: 1, restartPolicy: OnExitCode,
                  env: {TOK_TRAIN_STEPS: "100"}}
-        worker: {replicas: 7, gpusPerTask: 1}
+        worker: {replicas: 7, gpusPerTask: 1,
+                 cpusPerTask: 8, memMbPerTask: 65536}  # quota-tracked
 
 2. The reference CRD dialect (group train.distributed.io/v1alpha1,
    apis/train/v1alpha1/torchjob_types.go:88-206): `spec.torchTaskSpecs`
