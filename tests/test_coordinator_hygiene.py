@@ -217,3 +217,68 @@ def test_adopted_job_bypasses_admission_queue():
     assert len(ctl.handles["big"]) == 8
     assert len(node.free_slots) == 0
     assert any(e.reason == "TaskAdopted" for e in ctl.events)
+
+
+def test_adoption_churn_conserves_slots():
+    """Randomized manager-crash/adopt cycles: across repeated 'crashes'
+    (new controller + node over the surviving gang), GPU slots are
+    conserved, handles are never duplicated, and every job still runs
+    to completion."""
+    import random
+    rng = random.Random(42)
+    for trial in range(10):
+        num_gpus = 8
+        rt = FakeRuntime()
+        ctl = JobController(NodeState(num_gpus=num_gpus), rt,
+                            ControllerConfig())
+        jobs = []
+        for j in range(rng.randint(1, 3)):
+            job = ctl.create_job(mk_job(f"c{trial}-{j}",
+                                        workers=rng.randint(0, 2),
+                                        queue=f"q{j}"))
+            jobs.append(job)
+        for _ in range(3):
+            for job in jobs:
+                ctl.reconcile(job)
+
+        for _crash in range(rng.randint(1, 3)):
+            # snapshot live handles as adoption records
+            recs = {}
+            for name, hs in ctl.handles.items():
+                recs[name] = [
+                    {"pid": 10_000 + i, "task_type": h.task_type.value,
+                     "index": h.index, "generation": h.generation,
+                     "gpu_slots": list(h.gpu_slots)}
+                    for i, h in enumerate(hs.values()) if not h.finished]
+            # "crash": new controller + node + runtime
+            rt = FakeRuntime()
+            rt.adoptable = recs
+            ctl = JobController(NodeState(num_gpus=num_gpus), rt,
+                                ControllerConfig())
+            jobs = [ctl.create_job(mk_job(j.name,
+                                          workers=j.tasks.get(
+                                              TaskType.WORKER,
+                                              TaskSpec(replicas=0)).replicas,
+                                          queue=j.scheduling.queue))
+                    for j in jobs]
+            # no duplicate handles per (job, type, index)
+            for name, hs in ctl.handles.items():
+                assert len(hs) == len({h.key for h in hs.values()})
+            for _ in range(3):
+                for job in jobs:
+                    ctl.reconcile(job)
+            used = sum(len(h.gpu_slots) for hs in ctl.handles.values()
+                       for h in hs.values())
+            assert used + len(ctl.node.free_slots) == num_gpus
+
+        # drive everything to completion
+        for _ in range(6):
+            for key, h in list(rt.tasks.items()):
+                if not h.finished:
+                    rt.set_phase(key, TaskPhase.SUCCEEDED, exit_code=0)
+            for job in jobs:
+                ctl.reconcile(job)
+        for job in jobs:
+            assert job.status.phase is not None, job.name
+        assert len(ctl.node.free_slots) == num_gpus, \
+            (trial, ctl.node.alloc)
