@@ -11,6 +11,8 @@ import time
 
 import pytest
 
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
 from torch_on_k8s_amd.controlplane.api import (JobConditionType, TaskSpec,
                                                TaskType, TorchJob)
 from torch_on_k8s_amd.controlplane.controller import (ControllerConfig,
@@ -154,3 +156,60 @@ def test_arbitrary_command_task(tmp_path):
     drive(ctl, job, timeout=60)
     assert job.status.phase == JobConditionType.SUCCEEDED
     assert (tmp_path / "work" / "custom-cmd" / "ok").read_text() == "done"
+
+
+@pytest.mark.timeout(300)
+def test_aimaster_sidecar_and_periodic_ckpt(tmp_path):
+    """AIMaster supervision sidecar journals training metrics while the
+    master trains with TOK_CKPT_EVERY periodic async checkpoints; the
+    checkpoint exists mid-run and the job completes."""
+    import json as _json
+    from torch_on_k8s_amd.controlplane.api import (JobConditionType,
+                                                   TaskSpec, TaskType,
+                                                   TorchJob)
+    from torch_on_k8s_amd.controlplane.controller import (ControllerConfig,
+                                                          JobController)
+    from torch_on_k8s_amd.controlplane.node import NodeState
+    from torch_on_k8s_amd.controlplane.runtime import LocalProcessRuntime
+
+    rt = LocalProcessRuntime(str(tmp_path / "work"))
+    ctl = JobController(NodeState(num_gpus=0), rt,
+                        ControllerConfig(enable_gang_scheduling=False))
+    env = {
+        "TOK_BACKEND": "gloo",
+        "TOK_TRAIN_STEPS": "14",
+        "TOK_CKPT_EVERY": "5",
+        "TOK_STEP_DELAY": "0.15",
+        "TOK_TRAINER_CONFIG": _json.dumps(
+            {"model": "llama-tiny", "micro_batch": 1, "seq_len": 32}),
+        "PYTHONPATH": ROOT,
+    }
+    job = TorchJob(name="aim-e2e", tasks={
+        TaskType.AIMASTER: TaskSpec(replicas=1, gpus_per_task=0, env=env),
+        TaskType.MASTER: TaskSpec(replicas=1, gpus_per_task=0, env=env),
+    })
+    ctl.create_job(job)
+    state = tmp_path / "work" / "aim-e2e"
+    saw_midrun_ckpt = False
+    t0 = time.time()
+    try:
+        while time.time() - t0 < 240:
+            ctl.reconcile(job)
+            if (state / "ckpt" / "meta.json").exists() and \
+                    job.status.phase == JobConditionType.RUNNING:
+                saw_midrun_ckpt = True
+            if job.status.phase in (JobConditionType.SUCCEEDED,
+                                    JobConditionType.FAILED):
+                break
+            time.sleep(0.2)
+        assert job.status.phase == JobConditionType.SUCCEEDED, \
+            (job.status.phase, [(e.reason, e.message) for e in ctl.events])
+        assert saw_midrun_ckpt, "periodic checkpoint never published mid-run"
+        meta = _json.load(open(state / "ckpt" / "meta.json"))
+        assert meta["step"] % 5 == 0 and meta["step"] >= 5
+        # aimaster journaled structured metrics
+        journal = (state / "aimaster.log").read_text().strip().splitlines()
+        assert len(journal) >= 2
+        assert all("loss" in ln for ln in journal)
+    finally:
+        ctl.delete_job("aim-e2e")
