@@ -517,3 +517,23 @@ def test_rejoin_gate_holds_new_task_creation():
     job.annotations["rejoin-ready"] = str(job.generation)
     ctl.reconcile(job)
     assert ("rg", TaskType.WORKER, 1) in ctl.handles["rg"]
+
+
+def test_dag_on_phase_succeeded_pipeline():
+    """Workflow-style DAG: a task gated on upstream SUCCEEDED (not just
+    Running) starts only after the upstream completes — the reference's
+    generic DAGCondition onPhase semantics (dag.go:30-54, 111-116)."""
+    from torch_on_k8s_amd.controlplane.api import DAGCondition, TaskPhase
+    ctl, node, rt = mk_ctl()
+    job = mk_job("pipe", workers=1)
+    job.tasks[TaskType.WORKER].dag_conditions = [
+        DAGCondition(upstream=TaskType.MASTER,
+                     on_phase=TaskPhase.SUCCEEDED)]
+    ctl.create_job(job)
+    ctl.reconcile(job)  # master running
+    ctl.reconcile(job)
+    assert TaskType.WORKER not in {k[1] for k in rt.started}
+    rt.set_phase(("pipe", TaskType.MASTER, 0), TaskPhase.SUCCEEDED,
+                 exit_code=0)
+    ctl.reconcile(job)
+    assert TaskType.WORKER in {k[1] for k in rt.started}
