@@ -19,6 +19,9 @@ Trainer configuration:
                       <state_dir>/bench.json (read by `bench.py
                       --via-manager` for the gang-scheduled headline
                       number). TOK_TRAIN_STEPS must equal warmup+steps.
+  TOK_CKPT_EVERY      periodic async checkpoints every N steps
+  TOK_DISABLE_REJOIN / TOK_REJOIN_TIMEOUT / TOK_STEP_DELAY
+                      fast-rejoin + test knobs (docs/CONFIG.md)
 Exit codes follow the failover contract (controlplane/failover.py):
   0 success; 143 on SIGTERM after a clean checkpoint (retryable).
 
@@ -114,9 +117,12 @@ def main() -> int:
 
     signal.signal(signal.SIGTERM, on_term)
 
-    # agent.json is owned by rank 0; keep a merged view so checkpoint
-    # acks and rejoin-ready handshakes don't clobber each other
+    # agent.json is owned by rank 0; keep a merged view (seeded from any
+    # previous life's file) so checkpoint acks and rejoin-ready
+    # handshakes never clobber each other across restarts
     agent_state: dict = {}
+    if ctx.is_main and agent_file:
+        agent_state.update(_read_json(agent_file) or {})
 
     def write_agent(**kv):
         if ctx.is_main and agent_file:
