@@ -151,3 +151,16 @@ def test_client_logs(tmp_path):
     assert "line2" in cli.logs("j1")
     assert cli.logs("j1", tail=1).strip() == "line3"
     assert cli.logs("nope") == ""
+
+
+def test_feature_gate_defaults_parity():
+    """Same 5 gates, same defaults as the reference
+    (features/features.go:31-63)."""
+    from torch_on_k8s_amd.controlplane import features as feat
+    g = feat.FeatureGates()
+    assert g.enabled(feat.GANG_SCHEDULING)
+    assert g.enabled(feat.DAG_SCHEDULING)
+    assert g.enabled(feat.JOB_COORDINATOR)
+    assert g.enabled(feat.TORCH_LOCAL_MASTER_ADDR)
+    assert not g.enabled(feat.HOST_NET_WITH_HEADLESS_SVC)
+    assert len(g.as_dict()) == 5
