@@ -164,3 +164,50 @@ def test_feature_gate_defaults_parity():
     assert g.enabled(feat.TORCH_LOCAL_MASTER_ADDR)
     assert not g.enabled(feat.HOST_NET_WITH_HEADLESS_SVC)
     assert len(g.as_dict()) == 5
+
+
+def test_job_status_dict_contract(tmp_path):
+    """The status-file schema is the client API contract: keys must
+    stay stable (client/watchers parse them)."""
+    from torch_on_k8s_amd.controlplane.api import (TaskSpec, TaskType,
+                                                   TorchJob, set_defaults)
+    from torch_on_k8s_amd.controlplane.jobspec import job_status_dict
+    job = set_defaults(TorchJob(name="s", tasks={
+        TaskType.MASTER: TaskSpec(replicas=1)}))
+    job.status.set_condition(  # Created
+        __import__("torch_on_k8s_amd.controlplane.api",
+                   fromlist=["JobConditionType"]).JobConditionType.CREATED)
+    d = job_status_dict(job, events=[])
+    assert set(d) == {"events", "name", "phase", "generation",
+                      "restartCount", "conditions", "tasks",
+                      "modelVersion"}
+    assert d["phase"] == "Created"
+    assert d["generation"] == 1
+
+
+def test_client_cli_roundtrip(tmp_path):
+    """The kubectl-style CLI end-to-end over a workdir (no manager:
+    apply/list/delete against the spool, logs against state files)."""
+    import subprocess
+    import sys
+    env = dict(os.environ)
+    spec = tmp_path / "j.yaml"
+    spec.write_text("kind: TorchJob\nmetadata: {name: cj}\n"
+                    "spec:\n  tasks:\n    master: {replicas: 1}\n")
+
+    def cli(*args):
+        return subprocess.run(
+            [sys.executable, "-m", "torch_on_k8s_amd.client",
+             "--workdir", str(tmp_path)] + list(args),
+            capture_output=True, text=True, env=env, timeout=60)
+
+    r = cli("apply", str(spec))
+    assert r.returncode == 0 and "cj applied" in r.stdout
+    assert (tmp_path / "spool" / "cj.yaml").exists()
+    (tmp_path / "jobs" / "cj").mkdir(parents=True)
+    (tmp_path / "jobs" / "cj" / "cj-master-0.log").write_text("hello\n")
+    r = cli("logs", "cj")
+    assert "hello" in r.stdout
+    r = cli("delete", "cj")
+    assert "deleted" in r.stdout
+    assert not (tmp_path / "spool" / "cj.yaml").exists()
