@@ -188,6 +188,7 @@ def test_job_status_dict_contract(tmp_path):
 def test_client_cli_roundtrip(tmp_path):
     """The kubectl-style CLI end-to-end over a workdir (no manager:
     apply/list/delete against the spool, logs against state files)."""
+    import os
     import subprocess
     import sys
     env = dict(os.environ)

@@ -197,7 +197,8 @@ class ModelRegistry:
             diff_id = self._sha256_file(tmp_tar)
             tmp_gz = tmp_tar + ".gz"
             with open(tmp_tar, "rb") as fin, \
-                    gzip.GzipFile(filename="", fileobj=open(tmp_gz, "wb"),
+                    gzip.GzipFile(filename="", mode="wb",
+                                  fileobj=open(tmp_gz, "wb"),
                                   mtime=0) as fout:
                 shutil.copyfileobj(fin, fout)
             os.unlink(tmp_tar)
