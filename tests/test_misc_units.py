@@ -212,3 +212,18 @@ def test_client_cli_roundtrip(tmp_path):
     r = cli("delete", "cj")
     assert "deleted" in r.stdout
     assert not (tmp_path / "spool" / "cj.yaml").exists()
+
+
+def test_node_allocate_specific_conflict():
+    """Adoption must not silently steal slots another job holds."""
+    import pytest as _pytest
+    from torch_on_k8s_amd.controlplane.node import NodeState
+    n = NodeState(num_gpus=4)
+    n.allocate(2, ("a", "master", 0))
+    with _pytest.raises(RuntimeError):
+        n.allocate_specific((0, 1), ("b", "master", 0))
+    # re-claiming by the SAME owner is idempotent
+    got = n.allocate_specific((0,), ("a", "master", 0))
+    assert got == (0,)
+    n.allocate_specific((2, 3), ("b", "master", 0))
+    assert not n.free_slots
