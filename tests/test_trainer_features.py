@@ -113,3 +113,25 @@ def test_async_snapshot_resumable(tmp_path):
     tr2.train_step()
     for b2, br in zip(tr2.fb.buckets, ref.fb.buckets):
         assert torch.equal(b2.flat_param, br.flat_param)
+
+
+def test_lr_schedule_warmup_and_cosine_floor():
+    """current_lr: linear warmup then cosine to the floor."""
+    import torch
+    from torch_on_k8s_amd.engine.trainer import Trainer, TrainerConfig
+    from torch_on_k8s_amd.parallel.env import DistContext
+    torch.manual_seed(0)
+    tr = Trainer(TrainerConfig(model="llama-tiny", micro_batch=1,
+                               seq_len=32, lr=1e-2, lr_warmup_steps=4,
+                               lr_decay_steps=10, lr_min_ratio=0.1),
+                 DistContext())
+    lrs = []
+    for step in range(12):
+        tr.step_count = step
+        lrs.append(tr.current_lr())
+    # warmup strictly increasing to peak
+    assert lrs[0] < lrs[1] < lrs[2] < lrs[3]
+    assert abs(lrs[3] - 1e-2) < 1e-9
+    # decay monotonically down to the floor
+    assert all(a >= b - 1e-12 for a, b in zip(lrs[3:], lrs[4:]))
+    assert abs(lrs[-1] - 1e-3) < 1e-4  # floor = lr * min_ratio
