@@ -227,3 +227,18 @@ def test_node_allocate_specific_conflict():
     assert got == (0,)
     n.allocate_specific((2, 3), ("b", "master", 0))
     assert not n.free_slots
+
+
+def test_synthetic_data_deterministic_per_rank_and_step():
+    """SyntheticTokens: same (rank, step) -> same batch; different
+    ranks/steps -> different batches (DP correctness substrate)."""
+    import torch
+    from torch_on_k8s_amd.engine.data import SyntheticTokens
+    a = SyntheticTokens(512, 2, 16, torch.device("cpu"), rank=0, seed=7)
+    b = SyntheticTokens(512, 2, 16, torch.device("cpu"), rank=0, seed=7)
+    c = SyntheticTokens(512, 2, 16, torch.device("cpu"), rank=1, seed=7)
+    i0, l0 = a.batch(3)
+    i1, l1 = b.batch(3)
+    assert torch.equal(i0, i1) and torch.equal(l0, l1)
+    assert not torch.equal(a.batch(4)[0], i0)
+    assert not torch.equal(c.batch(3)[0], i0)
