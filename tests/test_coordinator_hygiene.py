@@ -282,3 +282,39 @@ def test_adoption_churn_conserves_slots():
             assert job.status.phase is not None, job.name
         assert len(ctl.node.free_slots) == num_gpus, \
             (trial, ctl.node.alloc)
+
+
+def test_priority_class_name_resolution():
+    """SchedulingPolicy.priorityClassName resolves against the manager's
+    priority-class table when `priority` is unset; an explicit priority
+    wins over the class (reference plugins/priority.go:48-85)."""
+    from torch_on_k8s_amd.controlplane.coordinator import PriorityPlugin
+
+    plug = PriorityPlugin({"high-priority": 100, "low": -5})
+    j_class = set_defaults(mk_job("by-class"))
+    j_class.scheduling.priority_class_name = "high-priority"
+    j_both = set_defaults(mk_job("both"))
+    j_both.scheduling.priority = 7
+    j_both.scheduling.priority_class_name = "high-priority"
+    j_unknown = set_defaults(mk_job("unknown-class"))
+    j_unknown.scheduling.priority_class_name = "nonexistent"
+    assert plug.score(QueueUnit(j_class)) == 100.0
+    assert plug.score(QueueUnit(j_both)) == 7.0     # explicit wins
+    assert plug.score(QueueUnit(j_unknown)) == 0.0  # unknown class -> 0
+
+
+def test_priority_class_orders_admission():
+    """Two jobs in one tenant queue: the one whose priorityClassName maps
+    to the larger value is admitted first."""
+    admitted = []
+    co = Coordinator(dequeue_fn=lambda job: admitted.append(job.name),
+                     tenant_usage_fn=lambda t: 0, default_quota=100,
+                     priority_classes={"gold": 50, "bronze": 1})
+    lo = set_defaults(mk_job("lo", queue="t1"))
+    lo.scheduling.priority_class_name = "bronze"
+    hi = set_defaults(mk_job("hi", queue="t1"))
+    hi.scheduling.priority_class_name = "gold"
+    co.enqueue_or_update(lo)
+    co.enqueue_or_update(hi)
+    co.schedule_once()
+    assert admitted[0] == "hi"

@@ -30,7 +30,8 @@ spec:
   clenPodPolicy: Running
   TTLSecondsAfterFinished: 600
   activeDurations: 3600
-  schedulingPolicy: {minAvailable: 3, queue: prod, priority: 7}
+  schedulingPolicy: {minAvailable: 3, queue: prod, priority: 7,
+                     priorityClassName: high-priority}
   torchTaskSpecs:
     Master:
       numTasks: 1
@@ -47,7 +48,8 @@ spec:
     Worker:
       numTasks: 2
       restartPolicy: OnFailure
-      spotTaskSpec: {numSpotTasks: 1, labels: {tier: spot}}
+      spotTaskSpec: {numSpotTasks: 1, labels: {tier: spot},
+                     priorityClassName: spot-low}
       template:
         spec:
           containers:
@@ -60,6 +62,8 @@ spec:
     numMinReplicas: 2
     numMaxReplicas: 8
     numWorkersPerNodePolicy: 1
+    rendezvousBackend: etcd
+    rendezvousEndpoint: etcd-host:2379
   modelVersion: {modelName: my-model}
 """
 
@@ -93,6 +97,10 @@ def test_reference_manifest_parses():
     assert job.scheduling.min_available == 3
     assert job.scheduling.queue == "prod"
     assert job.scheduling.priority == 7
+    assert job.scheduling.priority_class_name == "high-priority"
+    assert w.spot.priority_class_name == "spot-low"
+    assert job.elastic.rdzv_backend == "etcd"
+    assert job.elastic.rdzv_endpoint == "etcd-host:2379"
     assert job.min_members[TaskType.MASTER] == 1
     assert job.elastic is not None
     assert job.elastic.min_replicas == 2 and job.elastic.max_replicas == 8
@@ -130,6 +138,10 @@ def test_round_trip_crd_dialect():
     assert job2.scheduling.queue == job1.scheduling.queue
     assert (job2.elastic.min_replicas, job2.elastic.max_replicas) == \
         (job1.elastic.min_replicas, job1.elastic.max_replicas)
+    assert (job2.elastic.rdzv_backend, job2.elastic.rdzv_endpoint) == \
+        ("etcd", "etcd-host:2379")
+    assert job2.scheduling.priority_class_name == "high-priority"
+    assert job2.tasks[TaskType.WORKER].spot.priority_class_name == "spot-low"
     assert job2.model_name == job1.model_name
     # emitted dialect uses the node's GPU resource name
     res = doc["spec"]["torchTaskSpecs"]["Master"]["template"]["spec"][
@@ -204,3 +216,22 @@ def test_crd_job_runs_through_manager(tmp_path):
     assert job is not None
     assert job.status.phase == JobConditionType.SUCCEEDED, \
         (job.status.phase, mgr.controller.events_for("crd-e2e"))
+
+
+def test_rdzv_env_exported_for_custom_elastic_jobs():
+    """TorchElasticPolicy rendezvous fields surface in the task env
+    (reference emits torchrun --rdzv_* args, torchjob_controller.go:385-392;
+    here custom-command jobs read TOK_RDZV_BACKEND/ENDPOINT)."""
+    from torch_on_k8s_amd.controlplane.runtime import cluster_env
+
+    job = job_from_yaml(REFERENCE_STYLE)
+    env = cluster_env(job, TaskType.WORKER, 0)
+    assert env["TOK_RDZV_BACKEND"] == "etcd"
+    assert env["TOK_RDZV_ENDPOINT"] == "etcd-host:2379"
+    # endpoint defaults to the master address when unset
+    job.elastic.rdzv_endpoint = ""
+    env = cluster_env(job, TaskType.WORKER, 0, master_port=29400)
+    assert env["TOK_RDZV_ENDPOINT"] == "127.0.0.1:29400"
+    # non-elastic jobs carry no rendezvous keys
+    job.elastic = None
+    assert "TOK_RDZV_BACKEND" not in cluster_env(job, TaskType.WORKER, 0)

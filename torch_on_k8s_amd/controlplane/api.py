@@ -94,6 +94,9 @@ class SpotTaskSpec:
     """Lower-priority overflow replicas (torchjob_types.go:50-61)."""
     num_spot_replicas: int = 0
     priority: int = -10
+    # reference SpotTaskSpec.PriorityClassName (torchjob_types.go:56-58):
+    # resolved against the manager's priority-class table when set
+    priority_class_name: str = ""
     labels: dict = field(default_factory=dict)
 
 
@@ -118,6 +121,10 @@ class SchedulingPolicy:
     min_available: int | None = None
     queue: str = ""
     priority: int | None = None
+    # torchjob_types.go:127-130: used when `priority` is unset; the
+    # coordinator's PriorityPlugin resolves it against the manager's
+    # priority-class table (the PriorityClass-object analog)
+    priority_class_name: str = ""
 
 
 @dataclass
@@ -135,6 +142,14 @@ class ElasticPolicy:
     nproc_per_node: int = 1
     metric_window: int = 5     # observations before a scale decision
     max_num_metrics: int = 50  # stop autoscaling after this many samples
+    # reference TorchElasticPolicy rendezvous fields
+    # (torchjob_types.go:168-171, emitted as torchrun --rdzv_* args at
+    # torchjob_controller.go:385-392). The in-tree entrypoint rendezvous
+    # is the fast-rejoin TCPStore; these are exported to the task env as
+    # TOK_RDZV_BACKEND/TOK_RDZV_ENDPOINT for custom-command jobs that
+    # run torchrun themselves.
+    rdzv_backend: str = ""
+    rdzv_endpoint: str = ""
 
 
 @dataclass

@@ -199,8 +199,19 @@ class QuotaPlugin:
 
 
 class PriorityPlugin:
+    """Score = SchedulingPolicy.priority, else the value of its
+    priorityClassName in the manager's class table (the PriorityClass-
+    object analog; reference plugins/priority.go:48-85: explicit
+    priority wins, otherwise the named PriorityClass's value, else 0)."""
+
+    def __init__(self, classes: dict | None = None):
+        self.classes = dict(classes or {})  # class name -> int value
+
     def score(self, qu: QueueUnit) -> float:
-        return float(qu.job.scheduling.priority or 0)
+        pol = qu.job.scheduling
+        if pol.priority is not None:
+            return float(pol.priority)
+        return float(self.classes.get(pol.priority_class_name, 0))
 
 
 class Coordinator:
@@ -211,13 +222,14 @@ class Coordinator:
     SCHEDULE_PERIOD = 0.1  # 100ms (plugins/registry.go:27)
 
     def __init__(self, dequeue_fn, tenant_usage_fn=None, quotas=None,
-                 default_quota=8, selector: str = "wrr"):
+                 default_quota=8, selector: str = "wrr",
+                 priority_classes: dict | None = None):
         self.queues: dict[str, Queue] = {}
         self.dequeue_fn = dequeue_fn       # called with the TorchJob
         # returns in-use resources for a tenant: dict or bare gpu count
         self.tenant_usage_fn = tenant_usage_fn or (lambda tenant: 0)
         self.quota = QuotaPlugin(quotas, default_quota)
-        self.priority = PriorityPlugin()
+        self.priority = PriorityPlugin(priority_classes)
         self.selector = (WeightedRoundRobinSelector() if selector == "wrr"
                          else RoundRobinSelector())
         self._lock = threading.RLock()

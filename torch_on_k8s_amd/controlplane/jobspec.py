@@ -79,7 +79,8 @@ def _task_spec(d: dict) -> TaskSpec:
         s = d["spot"]
         spec.spot = SpotTaskSpec(
             num_spot_replicas=int(s.get("numSpotReplicas", 0)),
-            priority=int(s.get("priority", -10)))
+            priority=int(s.get("priority", -10)),
+            priority_class_name=s.get("priorityClassName", ""))
     return spec
 
 
@@ -101,6 +102,7 @@ def _task_spec_from_crd(d: dict) -> TaskSpec:
     if st:
         spec.spot = SpotTaskSpec(
             num_spot_replicas=int(st.get("numSpotTasks", 0)),
+            priority_class_name=st.get("priorityClassName", ""),
             labels=dict(st.get("labels") or {}))
     tmpl = (d.get("template") or {}).get("spec") or {}
     containers = tmpl.get("containers") or []
@@ -161,7 +163,8 @@ def _job_from_crd_dict(doc: dict) -> TorchJob:
     job.scheduling = SchedulingPolicy(
         min_available=sp.get("minAvailable"),
         queue=sp.get("queue", ""),
-        priority=sp.get("priority"))
+        priority=sp.get("priority"),
+        priority_class_name=sp.get("priorityClassName", ""))
     if spec.get("minMembers"):
         job.min_members = {TaskType(t.lower()): int(v)
                            for t, v in spec["minMembers"].items()}
@@ -173,7 +176,9 @@ def _job_from_crd_dict(doc: dict) -> TorchJob:
         job.elastic = ElasticPolicy(
             min_replicas=int(ep.get("numMinReplicas", 1)),
             max_replicas=int(ep.get("numMaxReplicas", 1)),
-            nproc_per_node=int(ep.get("numWorkersPerNodePolicy", 1)))
+            nproc_per_node=int(ep.get("numWorkersPerNodePolicy", 1)),
+            rdzv_backend=ep.get("rendezvousBackend", ""),
+            rdzv_endpoint=ep.get("rendezvousEndpoint", ""))
     return set_defaults(job)
 
 
@@ -204,6 +209,9 @@ def job_to_crd_dict(job: TorchJob) -> dict:
         if s.spot is not None:
             d["spotTaskSpec"] = {"numSpotTasks": s.spot.num_spot_replicas,
                                  "labels": dict(s.spot.labels)}
+            if s.spot.priority_class_name:
+                d["spotTaskSpec"]["priorityClassName"] = \
+                    s.spot.priority_class_name
         task_specs[t.value.capitalize() if t != TaskType.AIMASTER
                    else "AIMaster"] = d
     spec: dict = {"torchTaskSpecs": task_specs,
@@ -215,11 +223,13 @@ def job_to_crd_dict(job: TorchJob) -> dict:
     if job.run_policy.active_deadline_seconds is not None:
         spec["activeDurations"] = job.run_policy.active_deadline_seconds
     if job.scheduling.min_available or job.scheduling.queue or \
-            job.scheduling.priority:
+            job.scheduling.priority or job.scheduling.priority_class_name:
         spec["schedulingPolicy"] = {
             k: v for k, v in [("minAvailable", job.scheduling.min_available),
                               ("queue", job.scheduling.queue or None),
-                              ("priority", job.scheduling.priority)]
+                              ("priority", job.scheduling.priority),
+                              ("priorityClassName",
+                               job.scheduling.priority_class_name or None)]
             if v is not None}
     if job.min_members:
         spec["minMembers"] = {
@@ -234,6 +244,12 @@ def job_to_crd_dict(job: TorchJob) -> dict:
             "numMaxReplicas": job.elastic.max_replicas,
             "numWorkersPerNodePolicy": job.elastic.nproc_per_node,
         }
+        if job.elastic.rdzv_backend:
+            spec["torchElasticPolicy"]["rendezvousBackend"] = \
+                job.elastic.rdzv_backend
+        if job.elastic.rdzv_endpoint:
+            spec["torchElasticPolicy"]["rendezvousEndpoint"] = \
+                job.elastic.rdzv_endpoint
     return {
         "apiVersion": "train.distributed.io/v1alpha1",
         "kind": "TorchJob",
@@ -264,7 +280,8 @@ def job_from_dict(doc: dict) -> TorchJob:
     job.scheduling = SchedulingPolicy(
         min_available=sp.get("minAvailable"),
         queue=sp.get("queue", ""),
-        priority=sp.get("priority"))
+        priority=sp.get("priority"),
+        priority_class_name=sp.get("priorityClassName", ""))
     rp = spec.get("runPolicy") or {}
     job.run_policy = RunPolicy(
         clean_task_policy=CleanPodPolicy(rp.get("cleanTaskPolicy", "Running")),
@@ -278,7 +295,9 @@ def job_from_dict(doc: dict) -> TorchJob:
             max_replicas=int(ep.get("maxReplicas", 1)),
             nproc_per_node=int(ep.get("nprocPerNode", 1)),
             metric_window=int(ep.get("metricWindow", 5)),
-            max_num_metrics=int(ep.get("maxNumMetrics", 50)))
+            max_num_metrics=int(ep.get("maxNumMetrics", 50)),
+            rdzv_backend=ep.get("rdzvBackend", ""),
+            rdzv_endpoint=ep.get("rdzvEndpoint", ""))
     return set_defaults(job)
 
 

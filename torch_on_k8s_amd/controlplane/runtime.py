@@ -77,6 +77,13 @@ def cluster_env(job: TorchJob, t: TaskType, index: int,
         "TOK_TASK_INDEX": str(index),
         "TOK_GENERATION": str(job.generation),
     }
+    # torchelastic rendezvous args (torchjob_controller.go:385-392): the
+    # in-tree entrypoint rendezvouses over the fast-rejoin TCPStore, but
+    # custom-command jobs that run torchrun themselves read these.
+    if job.elastic is not None and job.elastic.rdzv_backend:
+        env["TOK_RDZV_BACKEND"] = job.elastic.rdzv_backend
+        env["TOK_RDZV_ENDPOINT"] = (job.elastic.rdzv_endpoint or
+                                    f"127.0.0.1:{master_port or DEFAULT_MASTER_PORT}")
     return env
 
 

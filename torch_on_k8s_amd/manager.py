@@ -42,7 +42,8 @@ log = get_logger("manager")
 class Manager:
     def __init__(self, workdir: str, num_gpus: int = 8, quotas=None,
                  gates: feat.FeatureGates | None = None,
-                 sync_period: float = 0.5, storage_spec: dict | None = None):
+                 sync_period: float = 0.5, storage_spec: dict | None = None,
+                 priority_classes: dict | None = None):
         self.workdir = workdir
         self.spool = os.path.join(workdir, "spool")
         self.status_dir = os.path.join(workdir, "status")
@@ -68,7 +69,8 @@ class Manager:
             self.coordinator = Coordinator(
                 dequeue_fn=self.controller.reconcile,
                 tenant_usage_fn=self.controller.tenant_resource_usage,
-                quotas=quotas, default_quota=num_gpus)
+                quotas=quotas, default_quota=num_gpus,
+                priority_classes=priority_classes)
             self.controller.coordinator = self.coordinator
         else:
             self.coordinator = None
@@ -215,6 +217,9 @@ def main():
                     help='tenant=gpus or tenant={"gpu":8,"cpu":64,'
                          '"memory_mb":512000}; repeatable')
     ap.add_argument("--sync-period", type=float, default=0.5)
+    ap.add_argument("--priority-class", action="append", default=[],
+                    help="name=value PriorityClass-object analog consulted "
+                         "when a job sets priorityClassName; repeatable")
     ap.add_argument("--storage", default="",
                     help='storage spec JSON, e.g. {"nfs": {"server": '
                          '"10.0.0.2", "path": "/exports/models"}} or '
@@ -229,11 +234,16 @@ def main():
         except ValueError:
             quotas[k] = json.loads(v)  # resource-map quota
     gates = feat.FeatureGates.from_flag(args.feature_gates)
+    prio_classes = {}
+    for pc in args.priority_class:
+        k, _, v = pc.partition("=")
+        prio_classes[k] = int(v)
     mgr = Manager(args.workdir, num_gpus=args.num_gpus,
                   quotas=quotas or None, gates=gates,
                   sync_period=args.sync_period,
                   storage_spec=json.loads(args.storage) if args.storage
-                  else None)
+                  else None,
+                  priority_classes=prio_classes or None)
     start_metrics_server(args.metrics_addr)
     log.info("workdir=%s gpus=%d gates=%s", args.workdir, args.num_gpus,
              gates.as_dict())
