@@ -318,3 +318,28 @@ def test_priority_class_orders_admission():
     co.enqueue_or_update(hi)
     co.schedule_once()
     assert admitted[0] == "hi"
+
+
+def test_spot_preemption_uses_priority_class():
+    """A gang blocked on GPUs preempts spot replicas of a job whose
+    priorityClassName resolves LOWER, even with no explicit priority set
+    on either job (pod.go:592-603 overlay + priority.go resolution)."""
+    from torch_on_k8s_amd.controlplane.api import SpotTaskSpec
+    from torch_on_k8s_amd.controlplane.controller import ControllerConfig
+
+    node = NodeState(num_gpus=2)
+    rt = FakeRuntime()
+    ctl = JobController(node, rt, ControllerConfig(
+        priority_classes={"gold": 50, "bronze": 1}))
+    lo = set_defaults(mk_job("lo-spot", workers=2, master=False))
+    lo.tasks[TaskType.WORKER].spot = SpotTaskSpec(num_spot_replicas=2)
+    lo.scheduling.priority_class_name = "bronze"
+    ctl.create_job(lo)
+    ctl.reconcile(lo)
+    assert len(node.free_slots) == 0
+    hi = set_defaults(mk_job("hi-gang", workers=2, master=False))
+    hi.scheduling.priority_class_name = "gold"
+    ctl.create_job(hi)
+    for _ in range(4):
+        ctl.reconcile(hi)
+    assert rt.killed, "gold-class gang should preempt bronze spot tasks"

@@ -48,6 +48,9 @@ class ControllerConfig:
     enable_gang_scheduling: bool = True
     enable_dag_scheduling: bool = True
     master_port_range: tuple = (20000, 30000)  # hostnetwork range analog
+    # PriorityClass-object analog (name -> value); consulted when a job
+    # sets priorityClassName without an explicit priority
+    priority_classes: dict = field(default_factory=dict)
 
 
 class JobController:
@@ -376,6 +379,14 @@ class JobController:
         if nh:
             nh.restart_count = rc
 
+    def _effective_priority(self, job: TorchJob) -> int:
+        """Explicit priority, else the job's priorityClassName resolved
+        against cfg.priority_classes (plugins/priority.go:48-85)."""
+        if job.scheduling.priority is not None:
+            return job.scheduling.priority
+        return int(self.cfg.priority_classes.get(
+            job.scheduling.priority_class_name, 0))
+
     def _preempt_spot_for(self, job: TorchJob):
         """Free GPU slots by gracefully killing spot replicas of
         lower-priority jobs (spot-task priority overlay, pod.go:592-603)."""
@@ -385,12 +396,12 @@ class JobController:
         need = pg.min_gpus - len(self.node.free_slots)
         if need <= 0:
             return
-        my_prio = job.scheduling.priority or 0
+        my_prio = self._effective_priority(job)
         victims = []
         for other, ohs in self.handles.items():
             if other == job.name:
                 continue
-            oprio = (self.jobs[other].scheduling.priority or 0) \
+            oprio = self._effective_priority(self.jobs[other]) \
                 if other in self.jobs else 0
             for h in ohs.values():
                 if h.spot and not h.finished and oprio < my_prio:

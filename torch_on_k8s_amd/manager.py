@@ -61,7 +61,8 @@ class Manager:
         self.metrics = JobMetrics()
         cfg = ControllerConfig(
             enable_gang_scheduling=self.gates.enabled(feat.GANG_SCHEDULING),
-            enable_dag_scheduling=self.gates.enabled(feat.DAG_SCHEDULING))
+            enable_dag_scheduling=self.gates.enabled(feat.DAG_SCHEDULING),
+            priority_classes=dict(priority_classes or {}))
         self.controller = JobController(
             node, runtime, cfg, metrics=self.metrics,
             model_registry=self.registry, elastic=ElasticScaler())
