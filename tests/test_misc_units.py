@@ -242,3 +242,40 @@ def test_synthetic_data_deterministic_per_rank_and_step():
     assert torch.equal(i0, i1) and torch.equal(l0, l1)
     assert not torch.equal(a.batch(4)[0], i0)
     assert not torch.equal(c.batch(3)[0], i0)
+
+
+def test_model_crd_schemas_parse_and_cover_registry_phases():
+    """The L6 model CRD schemas (configs/crd/, counterpart of the
+    reference's model.distributed.io_{models,modelversions}.yaml) are
+    valid YAML with the reference group/kinds, and the ModelVersion
+    phase enum covers every phase the in-tree registry emits."""
+    import os
+    import yaml
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    with open(os.path.join(repo, "configs", "crd",
+                           "modelversions.model.distributed.io.yaml")) as f:
+        mv = yaml.safe_load(f)
+    assert mv["spec"]["group"] == "model.distributed.io"
+    assert mv["spec"]["names"]["kind"] == "ModelVersion"
+    assert mv["spec"]["names"]["shortNames"] == ["mv"]
+    schema = mv["spec"]["versions"][0]["schema"]["openAPIV3Schema"]
+    spec_props = schema["properties"]["spec"]["properties"]
+    # reference ModelVersionSpec fields (modelversion_types.go:58-79)
+    for k in ("modelName", "createdBy", "storage", "imageRepo", "imageTag"):
+        assert k in spec_props, k
+    for st in ("nfs", "localStorage"):
+        assert st in spec_props["storage"]["properties"], st
+    phases = set(schema["properties"]["status"]["properties"]
+                 ["imageBuildPhase"]["enum"])
+    # every phase modelregistry.py writes must validate
+    assert {"Created", "Building", "Succeeded", "Failed"} <= phases
+
+    with open(os.path.join(repo, "configs", "crd",
+                           "models.model.distributed.io.yaml")) as f:
+        m = yaml.safe_load(f)
+    assert m["spec"]["group"] == "model.distributed.io"
+    assert m["spec"]["names"]["kind"] == "Model"
+    assert "latestVersion" in (m["spec"]["versions"][0]["schema"]
+                               ["openAPIV3Schema"]["properties"]["status"]
+                               ["properties"])
