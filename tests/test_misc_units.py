@@ -180,9 +180,16 @@ def test_job_status_dict_contract(tmp_path):
     d = job_status_dict(job, events=[])
     assert set(d) == {"events", "name", "phase", "generation",
                       "restartCount", "conditions", "tasks",
-                      "modelVersion"}
+                      "modelVersion", "elastic"}
     assert d["phase"] == "Created"
     assert d["generation"] == 1
+    assert d["elastic"] is None  # no TorchElasticStatus on non-elastic jobs
+    # elastic jobs publish the TorchElasticStatus analog
+    from torch_on_k8s_amd.controlplane.api import ElasticStatus
+    job.status.elastic = ElasticStatus(replicas=4, last_replicas=2)
+    e = job_status_dict(job, events=[])["elastic"]
+    assert e == {"currentReplicas": 4, "lastReplicas": 2,
+                 "continue": True, "elasticCondition": "Start"}
 
 
 def test_client_cli_roundtrip(tmp_path):

@@ -323,4 +323,11 @@ def job_status_dict(job: TorchJob, events=None) -> dict:
                             "failed": s.failed}
                   for t, s in job.status.tasks.items()},
         "modelVersion": job.status.model_version,
+        # TorchElasticStatus analog (torchjob_types.go:259-289)
+        "elastic": None if job.status.elastic is None else {
+            "currentReplicas": job.status.elastic.replicas,
+            "lastReplicas": job.status.elastic.last_replicas,
+            "continue": job.status.elastic.continue_training,
+            "elasticCondition": job.status.elastic.condition.value,
+        },
     }
