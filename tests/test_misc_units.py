@@ -286,3 +286,46 @@ def test_model_crd_schemas_parse_and_cover_registry_phases():
     assert "latestVersion" in (m["spec"]["versions"][0]["schema"]
                                ["openAPIV3Schema"]["properties"]["status"]
                                ["properties"])
+
+
+def test_client_describe(tmp_path):
+    """`describe` renders every status-schema section (kubectl describe
+    analog) and degrades gracefully for unknown jobs."""
+    import json
+    import os
+    import time
+    from torch_on_k8s_amd.client import TorchJobClient
+
+    cli = TorchJobClient(str(tmp_path))
+    assert "not found" in cli.describe("ghost")
+    os.makedirs(os.path.join(str(tmp_path), "status"), exist_ok=True)
+    st = {
+        "name": "d1", "phase": "Running", "generation": 2,
+        "restartCount": 1, "modelVersion": "mv-d1-ab123",
+        "elastic": {"currentReplicas": 4, "lastReplicas": 2,
+                    "continue": True, "elasticCondition": "ContinueTraining"},
+        "tasks": {"master": {"active": 1, "succeeded": 0, "failed": 0},
+                  "worker": {"active": 3, "succeeded": 0, "failed": 1}},
+        "conditions": [{"type": "Created", "reason": "JobCreated",
+                        "ts": time.time() - 90},
+                       {"type": "Running", "reason": "JobRunning",
+                        "ts": time.time() - 30}],
+        "events": [{"type": "Normal", "reason": "TaskStarted",
+                    "message": "d1-master-0", "ts": time.time() - 30}],
+    }
+    with open(os.path.join(str(tmp_path), "status", "d1.json"), "w") as f:
+        json.dump(st, f)
+    out = cli.describe("d1")
+    for needle in ("Phase:       Running", "Generation:  2",
+                   "mv-d1-ab123", "replicas=4", "ContinueTraining",
+                   "master", "worker", "failed=1", "Created", "Running",
+                   "TaskStarted", "d1-master-0"):
+        assert needle in out, (needle, out)
+    # CLI path
+    import subprocess
+    import sys
+    r = subprocess.run(
+        [sys.executable, "-m", "torch_on_k8s_amd.client",
+         "--workdir", str(tmp_path), "describe", "d1"],
+        capture_output=True, text=True)
+    assert r.returncode == 0 and "Phase:       Running" in r.stdout

@@ -81,6 +81,57 @@ class TorchJobClient:
             text = "\n".join(text.splitlines()[-tail:])
         return text
 
+    def describe(self, name: str) -> str:
+        """Human-readable status rendering (kubectl describe analog):
+        phase, tasks, conditions with ages, recent events, elastic
+        state — everything the status schema carries."""
+        st = self.get(name)
+        if st is None:
+            return f"torchjob {name!r} not found"
+        now = time.time()
+
+        def age(ts):
+            if not ts:
+                return "-"
+            d = max(0.0, now - ts)
+            if d < 120:
+                return f"{d:.0f}s"
+            if d < 7200:
+                return f"{d / 60:.0f}m"
+            return f"{d / 3600:.1f}h"
+
+        lines = [f"Name:        {st.get('name', name)}",
+                 f"Phase:       {st.get('phase')}",
+                 f"Generation:  {st.get('generation')}",
+                 f"Restarts:    {st.get('restartCount')}"]
+        if st.get("modelVersion"):
+            lines.append(f"ModelVersion: {st['modelVersion']}")
+        e = st.get("elastic")
+        if e:
+            lines.append(f"Elastic:     replicas={e.get('currentReplicas')} "
+                         f"last={e.get('lastReplicas')} "
+                         f"condition={e.get('elasticCondition')} "
+                         f"continue={e.get('continue')}")
+        lines.append("Tasks:")
+        for t, c in (st.get("tasks") or {}).items():
+            lines.append(f"  {t:10s} active={c.get('active', 0)} "
+                         f"succeeded={c.get('succeeded', 0)} "
+                         f"failed={c.get('failed', 0)}")
+        lines.append("Conditions:")
+        for c in st.get("conditions") or []:
+            lines.append(f"  {c.get('type', ''):12s} "
+                         f"{age(c.get('ts')):>6s}  {c.get('reason', '')}")
+        evs = st.get("events") or []
+        lines.append("Events:")
+        for ev in evs[-10:]:
+            lines.append(f"  {ev.get('type', ''):8s} "
+                         f"{age(ev.get('ts')):>6s}  "
+                         f"{ev.get('reason', ''):24s} "
+                         f"{ev.get('message', '')}")
+        if not evs:
+            lines.append("  <none>")
+        return "\n".join(lines)
+
     # -- model registry view (Model/ModelVersion read API analog) ------
     def _registry(self):
         from torch_on_k8s_amd.controlplane.modelregistry import (
@@ -122,7 +173,7 @@ def main():
     sub = ap.add_subparsers(dest="cmd", required=True)
     p_apply = sub.add_parser("apply")
     p_apply.add_argument("file")
-    for c in ("get", "delete", "wait"):
+    for c in ("get", "delete", "wait", "describe"):
         pc = sub.add_parser(c)
         pc.add_argument("name")
     sub.add_parser("list")
@@ -145,6 +196,8 @@ def main():
     elif args.cmd == "get":
         st = cli.get(args.name)
         print(json.dumps(st, indent=2) if st else f"not found: {args.name}")
+    elif args.cmd == "describe":
+        print(cli.describe(args.name))
     elif args.cmd == "delete":
         ok = cli.delete(args.name)
         print(f"torchjob/{args.name} {'deleted' if ok else 'not found'}")
