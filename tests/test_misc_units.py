@@ -329,3 +329,48 @@ def test_client_describe(tmp_path):
          "--workdir", str(tmp_path), "describe", "d1"],
         capture_output=True, text=True)
     assert r.returncode == 0 and "Phase:       Running" in r.stdout
+
+
+def test_negative_resource_requests_rejected():
+    """gpusPerTask: -3 must be a spec error, not a silent 5-GPU grant:
+    NodeState.allocate(-3) would slice free[:-3] and allocate
+    len(free)-3 slots to a request for negative GPUs."""
+    import pytest as _pytest
+    from torch_on_k8s_amd.controlplane.node import NodeState
+
+    n = NodeState(num_gpus=8)
+    with _pytest.raises(ValueError):
+        n.allocate(-3, "j")
+    assert len(n.free_slots) == 8  # nothing leaked
+
+    # native dialect
+    with _pytest.raises(ValueError):
+        job_from_dict({"metadata": {"name": "x"}, "spec": {"tasks": {
+            "master": {"replicas": 1, "gpusPerTask": -3}}}})
+    with _pytest.raises(ValueError):
+        job_from_dict({"metadata": {"name": "x"}, "spec": {"tasks": {
+            "master": {"replicas": 1, "memMbPerTask": -1}}}})
+    # reference CRD dialect
+    with _pytest.raises(ValueError):
+        job_from_dict({"metadata": {"name": "x"}, "spec": {
+            "torchTaskSpecs": {"Master": {"numTasks": 1, "template": {
+                "spec": {"containers": [{"name": "torch", "resources": {
+                    "limits": {"amd.com/gpu": -2}}}]}}}}}})
+
+
+def test_bad_crd_spec_surfaces_as_event(tmp_path):
+    """A CRD-dialect spec with an unknown task type or negative
+    resources becomes a BadJobSpec event; the manager keeps running."""
+    from torch_on_k8s_amd.manager import Manager
+    import os as _os
+    import yaml as _yaml
+
+    mgr = Manager(str(tmp_path), num_gpus=2, sync_period=0.05)
+    bad = {"metadata": {"name": "bad-crd"}, "spec": {"torchTaskSpecs": {
+        "Evaluator": {"numTasks": 1}}}}
+    with open(_os.path.join(mgr.spool, "bad-crd.yaml"), "w") as f:
+        _yaml.safe_dump(bad, f)
+    for _ in range(5):
+        mgr.step()
+    assert any(e.reason == "BadJobSpec" for e in mgr.controller.events)
+    assert "bad-crd" not in mgr.controller.jobs

@@ -60,6 +60,14 @@ def parse_quantity(q) -> float:
     return float(s)
 
 
+def _check_non_negative(**named) -> None:
+    """Resource amounts must be >= 0; a negative value is a spec error
+    (surfaced as a BadJobSpec event by the manager), not a default."""
+    for name, v in named.items():
+        if v is not None and v < 0:
+            raise ValueError(f"negative {name}: {v}")
+
+
 def _task_spec(d: dict) -> TaskSpec:
     spec = TaskSpec(
         replicas=int(d.get("replicas", 1)),
@@ -69,6 +77,9 @@ def _task_spec(d: dict) -> TaskSpec:
         command=d.get("command"),
         env={str(k): str(v) for k, v in (d.get("env") or {}).items()},
     )
+    _check_non_negative(gpusPerTask=spec.gpus_per_task,
+                        cpusPerTask=spec.cpus_per_task,
+                        memMbPerTask=spec.mem_mb_per_task)
     if d.get("restartPolicy"):
         spec.restart_policy = RestartPolicy(d["restartPolicy"])
     for c in d.get("dagConditions") or []:
@@ -133,6 +144,8 @@ def _task_spec_from_crd(d: dict) -> TaskSpec:
         if "memory" in req:
             spec.mem_mb_per_task = int(parse_quantity(req["memory"]) /
                                        (1 << 20))
+    _check_non_negative(gpu=spec.gpus_per_task, cpu=spec.cpus_per_task,
+                        memory=spec.mem_mb_per_task)
     return spec
 
 

@@ -23,6 +23,9 @@ class NodeState:
         return [s for s, owner in self.alloc.items() if owner is None]
 
     def allocate(self, n: int, owner) -> tuple:
+        if n < 0:
+            # a negative slice would silently grant len(free)+n slots
+            raise ValueError(f"negative GPU request: {n}")
         free = self.free_slots
         if len(free) < n:
             raise RuntimeError(
