@@ -374,3 +374,29 @@ def test_bad_crd_spec_surfaces_as_event(tmp_path):
         mgr.step()
     assert any(e.reason == "BadJobSpec" for e in mgr.controller.events)
     assert "bad-crd" not in mgr.controller.jobs
+
+
+def test_nonsense_policy_values_rejected():
+    """minAvailable < 0 would trivially satisfy the gang and fire
+    Running-at-MinMember with zero tasks; backoffLimit < 0 and inverted
+    elastic bounds are spec errors too (all -> BadJobSpec events)."""
+    import pytest as _pytest
+
+    def mk(spec_extra):
+        spec = {"tasks": {"master": {"replicas": 1}}}
+        spec.update(spec_extra)
+        return {"metadata": {"name": "x"}, "spec": spec}
+
+    with _pytest.raises(ValueError):
+        job_from_dict(mk({"schedulingPolicy": {"minAvailable": -4}}))
+    with _pytest.raises(ValueError):
+        job_from_dict(mk({"runPolicy": {"backoffLimit": -1}}))
+    with _pytest.raises(ValueError):
+        job_from_dict(mk({"elasticPolicy": {"minReplicas": 8,
+                                            "maxReplicas": 2}}))
+    # valid edges still accepted
+    ok = job_from_dict(mk({"schedulingPolicy": {"minAvailable": 0},
+                           "runPolicy": {"backoffLimit": 0},
+                           "elasticPolicy": {"minReplicas": 2,
+                                             "maxReplicas": 2}}))
+    assert ok.run_policy.backoff_limit == 0

@@ -301,4 +301,20 @@ def set_defaults(job: TorchJob) -> TorchJob:
     # torchjob_defaults.go:192-197 / SURVEY.md §2.3)
     if not job.min_members:
         job.min_members = {t: s.replicas for t, s in job.tasks.items()}
+    # spec sanity (surfaced as BadJobSpec by the manager): a negative
+    # MinAvailable would make the gang trivially satisfied and the
+    # Running-at-MinMember rule fire with zero tasks running; inverted
+    # elastic bounds would make the autoscaler oscillate
+    if job.scheduling.min_available is not None and \
+            job.scheduling.min_available < 0:
+        raise ValueError(
+            f"negative minAvailable: {job.scheduling.min_available}")
+    if job.run_policy.backoff_limit < 0:
+        raise ValueError(
+            f"negative backoffLimit: {job.run_policy.backoff_limit}")
+    if job.elastic is not None and \
+            job.elastic.min_replicas > job.elastic.max_replicas:
+        raise ValueError(
+            f"elastic minReplicas {job.elastic.min_replicas} > "
+            f"maxReplicas {job.elastic.max_replicas}")
     return job
