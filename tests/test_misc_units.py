@@ -400,3 +400,61 @@ def test_nonsense_policy_values_rejected():
                            "elasticPolicy": {"minReplicas": 2,
                                              "maxReplicas": 2}}))
     assert ok.run_policy.backoff_limit == 0
+
+
+def test_oci_extract_rejects_path_traversal(tmp_path):
+    """A crafted layer blob with '../' members or an escaping symlink
+    must not extract outside dest (py3.10 tarfile has no filter arg;
+    modelregistry._safe_extract validates members)."""
+    import io
+    import tarfile as _tar
+    import pytest as _pytest
+    from torch_on_k8s_amd.controlplane.modelregistry import _safe_extract
+
+    def mk_tar(add):
+        buf = io.BytesIO()
+        with _tar.open(fileobj=buf, mode="w:gz") as tf:
+            add(tf)
+        buf.seek(0)
+        return _tar.open(fileobj=buf, mode="r:gz")
+
+    def add_file(tf, name, data=b"x"):
+        info = _tar.TarInfo(name)
+        info.size = len(data)
+        tf.addfile(info, io.BytesIO(data))
+
+    dest = tmp_path / "out"
+    dest.mkdir()
+    # benign tar extracts fine
+    with mk_tar(lambda tf: add_file(tf, "model/weights.pt")) as tf:
+        _safe_extract(tf, str(dest))
+    assert (dest / "model" / "weights.pt").exists()
+    # ../ traversal refused
+    with mk_tar(lambda tf: add_file(tf, "../evil.txt")) as tf:
+        with _pytest.raises(RuntimeError, match="unsafe tar member"):
+            _safe_extract(tf, str(dest))
+    assert not (tmp_path / "evil.txt").exists()
+    # absolute path refused
+    with mk_tar(lambda tf: add_file(tf, "/etc/evil")) as tf:
+        with _pytest.raises(RuntimeError):
+            _safe_extract(tf, str(dest))
+    # escaping symlink refused
+    def add_sym(tf):
+        info = _tar.TarInfo("link")
+        info.type = _tar.SYMTYPE
+        info.linkname = "../../outside"
+        tf.addfile(info)
+    with mk_tar(add_sym) as tf:
+        with _pytest.raises(RuntimeError, match="unsafe tar link"):
+            _safe_extract(tf, str(dest))
+    # registry round-trip still works end-to-end
+    from torch_on_k8s_amd.controlplane.modelregistry import (ModelRegistry,
+                                                             StorageProvider)
+    reg = ModelRegistry(StorageProvider(str(tmp_path / "models")))
+    src = tmp_path / "src"
+    src.mkdir()
+    (src / "w").write_text("d")
+    reg.build_version("m", "v1", str(src))
+    out = reg.extract("m", "v1", str(tmp_path / "x"))
+    import os as _os
+    assert _os.path.exists(_os.path.join(out, "w"))

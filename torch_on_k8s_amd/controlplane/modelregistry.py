@@ -32,6 +32,28 @@ MODEL_PATH_ENV = "TORCH_ON_K8S_MODEL_PATH"
 MODEL_IMAGE_PATH = "/torch-on-k8s-model"  # path baked into the artifact
 
 
+def _safe_extract(tf: tarfile.TarFile, dest: str) -> None:
+    """extractall with member validation (py3.10 tarfile has no filter
+    arg): artifacts under the models dir are data, not trusted input —
+    reject absolute paths, '..' traversal, and symlinks/hardlinks that
+    escape dest."""
+    root = os.path.realpath(dest)
+    for m in tf.getmembers():
+        target = os.path.realpath(os.path.join(root, m.name))
+        if not (target == root or target.startswith(root + os.sep)):
+            raise RuntimeError(f"unsafe tar member path: {m.name!r}")
+        if m.issym() or m.islnk():
+            linkt = m.linkname
+            base = os.path.dirname(target) if m.issym() else root
+            link_target = os.path.realpath(os.path.join(base, linkt))
+            if not link_target.startswith(root + os.sep):
+                raise RuntimeError(
+                    f"unsafe tar link: {m.name!r} -> {linkt!r}")
+        if m.isdev():
+            raise RuntimeError(f"device node in artifact: {m.name!r}")
+    tf.extractall(dest)
+
+
 @dataclass
 class ModelVersion:
     model: str
@@ -322,5 +344,5 @@ class ModelRegistry:
         for layer in manifest["layers"]:
             ldigest = layer["digest"].split(":", 1)[1]
             with tarfile.open(os.path.join(blobs, ldigest), "r:gz") as tf:
-                tf.extractall(dest)
+                _safe_extract(tf, dest)
         return os.path.join(dest, MODEL_IMAGE_PATH.lstrip("/"))
