@@ -458,3 +458,35 @@ def test_oci_extract_rejects_path_traversal(tmp_path):
     out = reg.extract("m", "v1", str(tmp_path / "x"))
     import os as _os
     assert _os.path.exists(_os.path.join(out, "w"))
+
+
+def test_job_name_dns1123_validation():
+    """job.name becomes filesystem paths (spool/status/jobs); names that
+    k8s would reject at the API server must be rejected here too —
+    '../x' would otherwise escape the manager workdir."""
+    import pytest as _pytest
+
+    def mk(name):
+        return job_from_dict({"metadata": {"name": name},
+                              "spec": {"tasks": {"master": {"replicas": 1}}}})
+
+    for bad in ("../evil", "a/b", "a\\b", ".", "-x", "x-", "", "UP PER",
+                "x" * 300):
+        with _pytest.raises(ValueError):
+            mk(bad)
+    # canonicalization still applies before validation
+    assert mk("My_Job").name == "my-job"
+    assert mk("a.b-c9").name == "a.b-c9"
+
+
+def test_client_apply_rejects_traversal_names(tmp_path):
+    from torch_on_k8s_amd.client import TorchJobClient
+    import pytest as _pytest
+    cli = TorchJobClient(str(tmp_path))
+    with _pytest.raises(ValueError):
+        cli.apply({"metadata": {"name": "../../etc/x"},
+                   "spec": {"tasks": {"master": {"replicas": 1}}}})
+    # normal apply unaffected (underscores allowed pre-canonicalization)
+    assert cli.apply({"metadata": {"name": "ok_name"},
+                      "spec": {"tasks": {"master": {"replicas": 1}}}}) \
+        == "ok_name"

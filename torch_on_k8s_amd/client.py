@@ -28,6 +28,12 @@ class TorchJobClient:
                 import yaml
                 name = yaml.safe_load(text).get("metadata", {}).get(
                     "name", "torchjob")
+        # name becomes a spool filename: refuse anything k8s-invalid
+        # (controlplane.api.set_defaults applies the same rule at parse)
+        import re
+        if not re.fullmatch(r"[a-zA-Z0-9_]([-a-zA-Z0-9._]{0,251}"
+                            r"[a-zA-Z0-9_])?", name or ""):
+            raise ValueError(f"invalid job name {name!r}")
         tmp = os.path.join(self.spool, f".{name}.tmp")
         with open(tmp, "w") as f:
             f.write(text)

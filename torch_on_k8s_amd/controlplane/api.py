@@ -26,6 +26,7 @@ from __future__ import annotations
 
 import enum
 import itertools
+import re
 import time
 from dataclasses import dataclass, field
 
@@ -270,6 +271,12 @@ DEFAULT_MASTER_PORT = 23456  # reference constants.go:96-103
 def set_defaults(job: TorchJob) -> TorchJob:
     """SetDefaults_TorchJob parity (torchjob_defaults.go:29-74)."""
     job.name = job.name.lower().replace("_", "-")
+    # k8s enforces DNS-1123 object names before any controller runs;
+    # here job.name becomes filesystem paths (spool/status/jobs dirs),
+    # so an unvalidated name like '../x' would escape the workdir
+    if not re.fullmatch(r"[a-z0-9]([-a-z0-9.]{0,251}[a-z0-9])?", job.name):
+        raise ValueError(f"invalid job name {job.name!r}: must match "
+                         f"DNS-1123 (lowercase alphanumerics, '-', '.')")
     if not job.tasks:
         job.tasks = {TaskType.MASTER: TaskSpec()}
     for t, spec in job.tasks.items():
