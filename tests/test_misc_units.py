@@ -490,3 +490,21 @@ def test_client_apply_rejects_traversal_names(tmp_path):
     assert cli.apply({"metadata": {"name": "ok_name"},
                       "spec": {"tasks": {"master": {"replicas": 1}}}}) \
         == "ok_name"
+
+
+def test_world_size_annotation_garbage_falls_back():
+    """A non-numeric/non-positive world-size annotation must not crash
+    every task at env parse — fall back to the computed world."""
+    from torch_on_k8s_amd.controlplane.api import (TaskSpec, TaskType,
+                                                   TorchJob, set_defaults)
+    from torch_on_k8s_amd.controlplane.runtime import cluster_env
+    job = set_defaults(TorchJob(name="w", tasks={
+        TaskType.MASTER: TaskSpec(replicas=1),
+        TaskType.WORKER: TaskSpec(replicas=3)}))
+    assert cluster_env(job, TaskType.MASTER, 0)["WORLD_SIZE"] == "4"
+    job.annotations["world-size"] = "6"   # elastic override honored
+    assert cluster_env(job, TaskType.MASTER, 0)["WORLD_SIZE"] == "6"
+    for garbage in ("abc", "", "-2", "0"):
+        job.annotations["world-size"] = garbage
+        assert cluster_env(job, TaskType.MASTER, 0)["WORLD_SIZE"] == "4", \
+            garbage

@@ -51,6 +51,21 @@ def task_name(job: str, t: TaskType, index: int) -> str:
     return f"{job}-{t.value}-{index}"
 
 
+def _world_override(job: TorchJob, default: int) -> int:
+    """The elastic WORLD_SIZE annotation (torchjob_controller.go:419-439
+    downward-API analog) — but annotations are user-writable strings, so
+    a non-numeric or non-positive value falls back to the computed world
+    instead of crashing every task at env parse."""
+    v = job.annotations.get("world-size")
+    if v is None:
+        return default
+    try:
+        n = int(v)
+    except ValueError:
+        return default
+    return n if n >= 1 else default
+
+
 def cluster_env(job: TorchJob, t: TaskType, index: int,
                 master_port: int | None = None) -> dict:
     """The torch env contract (SetClusterSpec parity,
@@ -69,7 +84,7 @@ def cluster_env(job: TorchJob, t: TaskType, index: int,
     env = {
         "MASTER_ADDR": "127.0.0.1",
         "MASTER_PORT": str(master_port or DEFAULT_MASTER_PORT),
-        "WORLD_SIZE": str(job.annotations.get("world-size", world)),
+        "WORLD_SIZE": str(_world_override(job, world)),
         "RANK": str(rank),
         "PYTHONUNBUFFERED": "1",  # reference torchjob_controller.go:394-445
         "TOK_JOB_NAME": job.name,
