@@ -69,6 +69,14 @@ def test_serve_request_validation(trained_ckpt):
         "prompt_ids": [[10**6]]}).status_code == 400      # out of vocab
     assert c.post("/v1/generate", json={
         "prompt_ids": [[1]], "max_new_tokens": 0}).status_code == 422
+    # past the trained context: RoPE extrapolation degrades silently -> 400
+    max_seq = srv.meta["model_config"]["max_seq_len"]
+    r = c.post("/v1/generate", json={"prompt_ids": [[1] * (max_seq - 4)],
+                                     "max_new_tokens": 16})
+    assert r.status_code == 400 and "context" in r.json()["detail"]
+    # exactly at the limit is allowed
+    assert c.post("/v1/generate", json={"prompt_ids": [[1] * 4],
+                                        "max_new_tokens": 2}).status_code == 200
 
 
 def test_serve_from_oci_version(tmp_path, trained_ckpt):

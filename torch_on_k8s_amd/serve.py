@@ -171,6 +171,12 @@ def build_app(srv: InferenceServer):
         if vocab and any(t < 0 or t >= vocab
                          for p in req.prompt_ids for t in p):
             raise HTTPException(400, f"token id out of range [0,{vocab})")
+        max_seq = srv.meta.get("model_config", {}).get("max_seq_len")
+        if max_seq and len(req.prompt_ids[0]) + req.max_new_tokens > max_seq:
+            raise HTTPException(
+                400, f"prompt+max_new_tokens exceeds the model's trained "
+                     f"context ({max_seq}); RoPE extrapolation beyond it "
+                     f"degrades silently")
         return srv.generate(req.prompt_ids, req.max_new_tokens,
                             req.temperature)
 
