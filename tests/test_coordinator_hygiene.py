@@ -343,3 +343,51 @@ def test_spot_preemption_uses_priority_class():
     for _ in range(4):
         ctl.reconcile(hi)
     assert rt.killed, "gold-class gang should preempt bronze spot tasks"
+
+
+def test_quota_reject_emits_events():
+    """Quota rejection must be visible (the reference's flow-controlled
+    quota-plugin events, quota.go:59): a request larger than the whole
+    tenant quota -> ExceedsTenantQuota; a merely-busy quota ->
+    QuotaPending. Neither job dequeues."""
+    events = []
+    co = Coordinator(dequeue_fn=lambda job: events.append(("deq", job.name)),
+                     tenant_usage_fn=lambda t: 6,
+                     quotas={"t1": 8},
+                     event_fn=lambda j, r, m: events.append((r, j.name)))
+    giant = set_defaults(mk_job("giant", workers=15, queue="t1"))  # 16 GPUs
+    co.enqueue_or_update(giant)
+    co.schedule_once()
+    assert ("ExceedsTenantQuota", "giant") in events
+    assert ("deq", "giant") not in events
+    events.clear()
+    busy = set_defaults(mk_job("busy", workers=3, queue="t1"))  # 4 > 8-6
+    co.enqueue_or_update(busy)
+    co.schedule_once()
+    assert ("QuotaPending", "busy") in events
+    assert ("deq", "busy") not in events
+
+
+def test_quota_reject_event_reaches_manager_status(tmp_path):
+    """Through the Manager wiring: the Warning event lands in the job's
+    event ring (and hence its published status)."""
+    from torch_on_k8s_amd.manager import Manager
+    import os as _os
+    import yaml as _yaml
+    mgr = Manager(str(tmp_path), num_gpus=2, sync_period=0.05,
+                  quotas={"default": 2})
+    doc = {"metadata": {"name": "too-big"},
+           "spec": {"tasks": {"master": {"replicas": 1},
+                              "worker": {"replicas": 7}}}}  # 8 > 2
+    with open(_os.path.join(mgr.spool, "too-big.yaml"), "w") as f:
+        _yaml.safe_dump(doc, f)
+    import time as _t
+    deadline = _t.time() + 10
+    while _t.time() < deadline:
+        mgr.step()
+        if any(e.reason == "ExceedsTenantQuota"
+               for e in mgr.controller.events_for("too-big")):
+            break
+        _t.sleep(0.02)
+    assert any(e.reason == "ExceedsTenantQuota"
+               for e in mgr.controller.events_for("too-big"))

@@ -223,13 +223,17 @@ class Coordinator:
 
     def __init__(self, dequeue_fn, tenant_usage_fn=None, quotas=None,
                  default_quota=8, selector: str = "wrr",
-                 priority_classes: dict | None = None):
+                 priority_classes: dict | None = None, event_fn=None):
         self.queues: dict[str, Queue] = {}
         self.dequeue_fn = dequeue_fn       # called with the TorchJob
         # returns in-use resources for a tenant: dict or bare gpu count
         self.tenant_usage_fn = tenant_usage_fn or (lambda tenant: 0)
         self.quota = QuotaPlugin(quotas, default_quota)
         self.priority = PriorityPlugin(priority_classes)
+        # quota-reject visibility (the reference's flow-controlled event
+        # recorder in the quota plugin, quota.go:59): called with
+        # (job, reason, message); dedup/flow-control is the sink's job
+        self.event_fn = event_fn or (lambda job, reason, msg: None)
         self.selector = (WeightedRoundRobinSelector() if selector == "wrr"
                          else RoundRobinSelector())
         self._lock = threading.RLock()
@@ -285,6 +289,23 @@ class Coordinator:
                     continue
                 in_use = self.tenant_usage_fn(qu.tenant)
                 if not self.quota.filter(qu, in_use):
+                    # a request larger than the tenant's whole quota can
+                    # NEVER admit — tell the user, don't queue silently
+                    quota = self.quota.tenant_quota(qu.tenant)
+                    over = {r: (qu.request.get(r, 0), lim)
+                            for r, lim in quota.items()
+                            if qu.request.get(r, 0) > lim}
+                    if over:
+                        self.event_fn(
+                            qu.job, "ExceedsTenantQuota",
+                            f"request exceeds tenant {qu.tenant!r} quota "
+                            f"outright (resource: (requested, limit)) "
+                            f"{over}; the job cannot admit")
+                    else:
+                        self.event_fn(
+                            qu.job, "QuotaPending",
+                            f"tenant {qu.tenant!r} quota busy "
+                            f"(in use + assumed); job queued")
                     continue
                 candidates.append((self.priority.score(qu), qu))
             if not candidates:

@@ -71,7 +71,9 @@ class Manager:
                 dequeue_fn=self.controller.reconcile,
                 tenant_usage_fn=self.controller.tenant_resource_usage,
                 quotas=quotas, default_quota=num_gpus,
-                priority_classes=priority_classes)
+                priority_classes=priority_classes,
+                event_fn=lambda job, reason, msg: self.controller.event(
+                    job.name, "Warning", reason, msg))
             self.controller.coordinator = self.coordinator
         else:
             self.coordinator = None
