@@ -391,3 +391,26 @@ def test_quota_reject_event_reaches_manager_status(tmp_path):
         _t.sleep(0.02)
     assert any(e.reason == "ExceedsTenantQuota"
                for e in mgr.controller.events_for("too-big"))
+
+
+def test_gang_unsatisfiable_event():
+    """A gang larger than the whole node warns GangUnsatisfiable instead
+    of the generic waiting event (waiting can never help)."""
+    node = NodeState(num_gpus=2)
+    ctl = JobController(node, FakeRuntime())
+    big = set_defaults(mk_job("big", workers=7))  # 8 GPUs on a 2-GPU node
+    ctl.create_job(big)
+    ctl.reconcile(big)
+    assert any(e.reason == "GangUnsatisfiable"
+               for e in ctl.events_for("big"))
+    # a merely-busy gang still gets the Normal waiting event
+    ok1 = set_defaults(mk_job("ok1", workers=1))  # 2 GPUs
+    ctl.create_job(ok1)
+    ctl.reconcile(ok1)  # admits, node now full
+    ok2 = set_defaults(mk_job("ok2", workers=1))
+    ctl.create_job(ok2)
+    ctl.reconcile(ok2)
+    assert any(e.reason == "GangNotAdmitted"
+               for e in ctl.events_for("ok2"))
+    assert not any(e.reason == "GangUnsatisfiable"
+                   for e in ctl.events_for("ok2"))

@@ -260,8 +260,15 @@ class JobController:
             if not hs and not self.gang.can_admit(job):
                 self._preempt_spot_for(job)
             if not hs and not self.gang.can_admit(job):
-                self.event(job.name, "Normal", "GangNotAdmitted",
-                           "waiting for gang quota")
+                pg = self.gang.groups.get(job.name)
+                if pg is not None and pg.min_gpus > self.node.num_gpus:
+                    # larger than the node itself: waiting can never help
+                    self.event(job.name, "Warning", "GangUnsatisfiable",
+                               f"gang needs {pg.min_gpus} GPUs but the "
+                               f"node has {self.node.num_gpus}")
+                else:
+                    self.event(job.name, "Normal", "GangNotAdmitted",
+                               "waiting for gang quota")
                 return
 
         # per-task-type reconcile in AIMaster -> Master -> Worker order
