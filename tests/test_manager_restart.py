@@ -158,3 +158,24 @@ def test_manager_graceful_shutdown_keeps_tasks(tmp_path):
             os.kill(json.load(open(rec))["pid"], signal.SIGKILL)
         except (OSError, ValueError, FileNotFoundError):
             pass
+
+
+def test_leader_election_lock(tmp_path):
+    """Leader-election analog (reference main.go:77-83): a second
+    manager on the same workdir cannot become leader until the first
+    releases; a different workdir is independent."""
+    import pytest as _pytest
+    from torch_on_k8s_amd.manager import Manager
+
+    m1 = Manager(str(tmp_path), num_gpus=1, sync_period=0.05)
+    m1.acquire_leadership(block=False)
+    m2 = Manager(str(tmp_path), num_gpus=1, sync_period=0.05)
+    with _pytest.raises(RuntimeError, match="another manager"):
+        m2.acquire_leadership(block=False)
+    other = Manager(str(tmp_path / "other"), num_gpus=1, sync_period=0.05)
+    other.acquire_leadership(block=False)  # independent workdir: fine
+    other.release_leadership()
+    # leader exits -> takeover succeeds
+    m1.release_leadership()
+    m2.acquire_leadership(block=False)
+    m2.release_leadership()

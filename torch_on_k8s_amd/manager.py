@@ -188,11 +188,41 @@ class Manager:
         self.autoscale_pass()
         self.publish_status()
 
+    def acquire_leadership(self, block: bool = True):
+        """Leader election analog (reference main.go:77-83, election id
+        torch-on-k8s-election): an exclusive flock on
+        <workdir>/manager.lock. Two daemons on one workdir would both
+        start tasks and oversubscribe the node's real GPUs; the lock
+        makes the second one wait (block=True, takeover on leader exit,
+        the k8s semantics) or fail fast (block=False). Held for the
+        process lifetime; the OS releases it on ANY exit, so a crashed
+        leader never wedges the next one."""
+        import fcntl
+        self._lock_file = open(os.path.join(self.workdir, "manager.lock"),
+                               "w")
+        try:
+            fcntl.flock(self._lock_file,
+                        fcntl.LOCK_EX | (0 if block else fcntl.LOCK_NB))
+        except OSError:
+            self._lock_file.close()
+            self._lock_file = None
+            raise RuntimeError(
+                f"another manager owns {self.workdir} (manager.lock held)")
+        self._lock_file.truncate(0)
+        self._lock_file.write(f"{os.getpid()}\n")
+        self._lock_file.flush()
+
+    def release_leadership(self):
+        if getattr(self, "_lock_file", None) is not None:
+            self._lock_file.close()  # closing the fd drops the flock
+            self._lock_file = None
+
     def run_forever(self):
         """Main loop with graceful shutdown: on SIGTERM/SIGINT, publish
         final statuses and exit 0 WITHOUT killing task processes — a
         restarted manager adopts them (controller._adopt_orphans), so a
         manager upgrade never interrupts running gangs."""
+        self.acquire_leadership(block=True)
         import signal as _signal
         stop = {"flag": False}
 
