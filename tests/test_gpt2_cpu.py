@@ -52,3 +52,31 @@ def test_gpt2_trainer_loss_decreases():
         losses.append(tr.train_step())
         tr.step_count = 1
     assert losses[-1] < losses[0], losses
+
+
+def test_gpt2_generate_and_serve():
+    """GPT-2 serves through the same HTTP endpoint as Llama: greedy
+    generate is deterministic, respects the learned-position cap, and
+    the FastAPI route round-trips it."""
+    import pytest
+    import torch
+    from torch_on_k8s_amd.serve import InferenceServer, build_app
+    from starlette.testclient import TestClient
+
+    torch.manual_seed(0)
+    srv = InferenceServer.from_preset("gpt2-tiny", "cpu")
+    out = srv.generate([[1, 2, 3]], max_new_tokens=4)
+    assert len(out["output_ids"][0]) == 7
+    out2 = srv.generate([[1, 2, 3]], max_new_tokens=4)
+    assert out["output_ids"] == out2["output_ids"]  # greedy = deterministic
+    with pytest.raises(ValueError, match="learned positions"):
+        srv.model.generate(torch.zeros(1, 250, dtype=torch.long),
+                           max_new_tokens=16)
+    c = TestClient(build_app(srv))
+    r = c.post("/v1/generate", json={"prompt_ids": [[5, 6]],
+                                     "max_new_tokens": 3})
+    assert r.status_code == 200 and len(r.json()["new_ids"][0]) == 3
+    # past max_seq_len -> 400 from the endpoint guard
+    r = c.post("/v1/generate", json={"prompt_ids": [[1] * 250],
+                                     "max_new_tokens": 16})
+    assert r.status_code == 400

@@ -144,3 +144,28 @@ class GPT2Model(nn.Module):
         if labels is None:
             return logits
         return ops.cross_entropy(logits, labels)
+
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
+                 temperature: float = 0.0, **_) -> torch.Tensor:
+        """Greedy/sampled decoding so GPT-2 checkpoints serve through
+        the same endpoint as Llama (serve.py). Full-context recompute
+        per token — honest but unoptimized: the KV-cache/flash-decode/
+        hipGraph serving fast path is Llama's (models/llama.py
+        generate()); GPT-2 is the op-layer-generality family. Learned
+        positions cap the total length at max_seq_len."""
+        total = input_ids.shape[1] + max_new_tokens
+        if total > self.cfg.max_seq_len:
+            raise ValueError(
+                f"prompt+max_new_tokens {total} exceeds learned positions "
+                f"({self.cfg.max_seq_len})")
+        out = input_ids
+        for _ in range(max_new_tokens):
+            logits = self.forward(out)[:, -1]
+            if temperature and temperature > 0:
+                probs = torch.softmax(logits.float() / temperature, dim=-1)
+                nxt = torch.multinomial(probs, 1)
+            else:
+                nxt = logits.argmax(-1, keepdim=True)
+            out = torch.cat([out, nxt], dim=1)
+        return out
