@@ -130,6 +130,13 @@ class SchedulingPolicy:
 
 @dataclass
 class RunPolicy:
+    # DELIBERATE divergence from the reference default of None
+    # (torchjob_defaults.go:30-33): on k8s, pods left behind only hold
+    # kubelet bookkeeping, but here a straggler process holds REAL GPU
+    # slots on the one node until TTL/delete. Running (kill stragglers
+    # on completion) is the safe single-node default; set
+    # cleanTaskPolicy/clenPodPolicy: None explicitly for reference
+    # semantics (both honored, tests cover both).
     clean_task_policy: CleanPodPolicy = CleanPodPolicy.RUNNING
     ttl_seconds_after_finished: float | None = None
     active_deadline_seconds: float | None = None
