@@ -179,6 +179,10 @@ def main():
     sub = ap.add_subparsers(dest="cmd", required=True)
     p_apply = sub.add_parser("apply")
     p_apply.add_argument("file")
+    p_val = sub.add_parser(
+        "validate", help="parse a manifest (either dialect) without "
+                         "submitting; prints the canonical CRD form")
+    p_val.add_argument("file")
     for c in ("get", "delete", "wait", "describe"):
         pc = sub.add_parser(c)
         pc.add_argument("name")
@@ -199,6 +203,21 @@ def main():
         with open(args.file) as f:
             name = cli.apply(f.read())
         print(f"torchjob/{name} applied")
+    elif args.cmd == "validate":
+        from torch_on_k8s_amd.controlplane.jobspec import (job_from_yaml,
+                                                           job_to_crd_dict)
+        import sys as _sys
+        import yaml as _yaml
+        with open(args.file) as f:
+            text = f.read()
+        try:
+            job = job_from_yaml(text)
+        except (ValueError, KeyError, TypeError) as e:
+            print(f"INVALID: {e}", file=_sys.stderr)
+            raise SystemExit(1)
+        print(f"torchjob/{job.name} valid "
+              f"({job.total_replicas()} tasks, {job.total_gpus()} GPUs)")
+        print(_yaml.safe_dump(job_to_crd_dict(job), sort_keys=False), end="")
     elif args.cmd == "get":
         st = cli.get(args.name)
         print(json.dumps(st, indent=2) if st else f"not found: {args.name}")
