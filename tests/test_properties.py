@@ -77,3 +77,49 @@ def test_crd_round_trip_replicas(workers, gpus, cpus):
     assert job2.total_replicas() == job.total_replicas()
     assert abs(job2.total_resources()["cpu"] -
                job.total_resources()["cpu"]) < 1e-9
+
+
+@given(st.text(min_size=0, max_size=40))
+@settings(max_examples=300)
+def test_job_name_validation_is_filesystem_safe(name):
+    """Property: any name that survives set_defaults stays inside a
+    directory when joined (no separators, no '..', non-empty) — the
+    invariant the spool/status/jobs layout depends on."""
+    import os.path
+    from torch_on_k8s_amd.controlplane.api import (TaskSpec, TaskType,
+                                                   TorchJob, set_defaults)
+    try:
+        job = set_defaults(TorchJob(
+            name=name, tasks={TaskType.MASTER: TaskSpec(replicas=1)}))
+    except ValueError:
+        return  # rejected: nothing to check
+    n = job.name
+    assert n and "/" not in n and "\\" not in n and n not in (".", "..")
+    joined = os.path.normpath(os.path.join("/base", n))
+    assert joined.startswith("/base/") or joined == "/base/" + n
+
+
+@given(st.integers(min_value=-4, max_value=12),
+       st.integers(min_value=0, max_value=8))
+@settings(max_examples=60)
+def test_allocate_never_overgrants(n, preallocated):
+    """Property: allocate(n) either returns exactly n free slots or
+    raises; the free count never drops by more than n."""
+    import pytest
+    from torch_on_k8s_amd.controlplane.node import NodeState
+    node = NodeState(num_gpus=8)
+    if preallocated:
+        node.allocate(min(preallocated, 8), "other")
+    free_before = len(node.free_slots)
+    if n < 0:
+        with pytest.raises(ValueError):
+            node.allocate(n, "j")
+        assert len(node.free_slots) == free_before
+    elif n > free_before:
+        with pytest.raises(RuntimeError):
+            node.allocate(n, "j")
+        assert len(node.free_slots) == free_before
+    else:
+        got = node.allocate(n, "j")
+        assert len(got) == n
+        assert len(node.free_slots) == free_before - n
