@@ -66,6 +66,9 @@ def test_manager_autoscales_on_metrics(tmp_path):
             "autoscaler never doubled the workers"
         est = job.status.elastic
         assert est is not None and est.last_replicas == 1
+        # the decision surfaced as a job event (describe/status visible)
+        assert any(e.reason.startswith("Elastic")
+                   for e in mgr.controller.events_for("auto-e2e"))
         assert job.status.phase == JobConditionType.SUCCEEDED, (
             job.status.phase, mgr.controller.events_for("auto-e2e"))
         # the gang really ran at world 3 after the scale (master + 2)

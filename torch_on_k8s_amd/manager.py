@@ -178,7 +178,13 @@ class Manager:
             if job.status.phase != JobConditionType.RUNNING:
                 continue
             self.autoscaler.observe(job)
-            self.autoscaler.decide(job)
+            decision = self.autoscaler.decide(job)
+            if decision is not None:
+                st = job.status.elastic
+                self.controller.event(
+                    job.name, "Normal", f"Elastic{decision.value}",
+                    f"autoscaler: replicas {st.last_replicas} -> "
+                    f"{st.replicas}" if st else "")
 
     def step(self):
         self.sync_spool()
