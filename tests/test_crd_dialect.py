@@ -235,3 +235,62 @@ def test_rdzv_env_exported_for_custom_elastic_jobs():
     # non-elastic jobs carry no rendezvous keys
     job.elastic = None
     assert "TOK_RDZV_BACKEND" not in cluster_env(job, TaskType.WORKER, 0)
+
+
+def test_emitted_crd_fields_declared_in_schema():
+    """Schema/parser drift check: every key job_to_crd_dict emits for a
+    maximal job must appear somewhere in the CRD schema document
+    (configs/crd/torchjobs...yaml), so a manifest round-tripped through
+    the exporter always validates against the published schema."""
+    from torch_on_k8s_amd.controlplane.api import (ElasticPolicy,
+                                                   SchedulingPolicy,
+                                                   SpotTaskSpec, TaskSpec,
+                                                   TorchJob, set_defaults)
+
+    job = set_defaults(TorchJob(
+        name="maximal",
+        tasks={TaskType.MASTER: TaskSpec(replicas=1, gpus_per_task=1,
+                                         cpus_per_task=2.0,
+                                         mem_mb_per_task=1024,
+                                         command=["python", "x.py"]),
+               TaskType.WORKER: TaskSpec(
+                   replicas=2, gpus_per_task=1,
+                   spot=SpotTaskSpec(num_spot_replicas=1,
+                                     priority_class_name="spot"))},
+        scheduling=SchedulingPolicy(min_available=3, queue="q", priority=5,
+                                    priority_class_name="gold"),
+        elastic=ElasticPolicy(min_replicas=1, max_replicas=2,
+                              rdzv_backend="etcd",
+                              rdzv_endpoint="e:2379"),
+        model_name="m"))
+    job.run_policy.ttl_seconds_after_finished = 60
+    job.run_policy.active_deadline_seconds = 600
+    doc = job_to_crd_dict(job)
+
+    def all_keys(d):
+        out = set()
+        if isinstance(d, dict):
+            for k, v in d.items():
+                out.add(k)
+                out |= all_keys(v)
+        elif isinstance(d, list):
+            for v in d:
+                out |= all_keys(v)
+        return out
+
+    with open(os.path.join(REPO, "configs", "crd",
+                           "torchjobs.train.distributed.io.yaml")) as f:
+        schema_text = f.read()
+    emitted = all_keys(doc["spec"]) | all_keys(doc["metadata"])
+    # not schema keys: task-type map keys, resource names, standard
+    # ObjectMeta, and the pod-template subtree (declared open via
+    # x-kubernetes-preserve-unknown-fields — the hand-maintained
+    # equivalent of the reference's 7k-line expanded PodTemplateSpec)
+    assert "x-kubernetes-preserve-unknown-fields: true" in schema_text
+    skip = {"Master", "Worker", "AIMaster", "amd.com/gpu", "cpu", "memory",
+            "name", "value", "annotations", "labels", "namespace",
+            "containers", "limits", "spec"}
+    missing = sorted(k for k in emitted - skip
+                     if f"{k}:" not in schema_text and
+                     f"{k} " not in schema_text)
+    assert not missing, f"emitted keys absent from CRD schema: {missing}"
