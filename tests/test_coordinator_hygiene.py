@@ -414,3 +414,51 @@ def test_gang_unsatisfiable_event():
                for e in ctl.events_for("ok2"))
     assert not any(e.reason == "GangUnsatisfiable"
                    for e in ctl.events_for("ok2"))
+
+
+def test_sigterm_ignoring_task_keeps_slots_until_reaped():
+    """A deleted job's task that survives the graceful SIGTERM (it is
+    checkpointing, or stuck) still occupies its physical GPUs: the
+    slots must stay HELD until the process really exits, with SIGKILL
+    escalation after kill_grace_seconds — releasing at delete time
+    would hand a busy GPU to the next gang."""
+    from torch_on_k8s_amd.controlplane.api import TaskPhase
+    from torch_on_k8s_amd.controlplane.controller import ControllerConfig
+
+    class StubbornRuntime(FakeRuntime):
+        def kill(self, h, grace=True):
+            self.killed.append((h.key, grace))
+            if not grace:  # only SIGKILL fells it, and only on poll
+                h._dying = True
+
+        def poll(self, h):
+            if getattr(h, "_dying", False):
+                h.phase = TaskPhase.FAILED
+                h.exit_code = 137
+                return h
+            return super().poll(h)
+
+    node = NodeState(num_gpus=2)
+    rt = StubbornRuntime()
+    ctl = JobController(node, rt, ControllerConfig(kill_grace_seconds=0.2))
+    job = set_defaults(mk_job("stuck", workers=1, master=False))
+    ctl.create_job(job)
+    ctl.reconcile(job)
+    assert len(node.free_slots) == 1
+    ctl.delete_job(job.name)
+    # SIGTERM ignored: the slot is NOT back yet
+    assert len(node.free_slots) == 1
+    assert ctl._reaping
+    ctl.reap_pass()  # inside the grace window: still held
+    assert len(node.free_slots) == 1
+    time.sleep(0.25)
+    ctl.reap_pass()  # past deadline: escalates to SIGKILL
+    assert ((job.name, TaskType.WORKER, 0), False) in rt.killed
+    ctl.reap_pass()  # process now gone: slot released
+    assert len(node.free_slots) == 2
+    assert not ctl._reaping
+    # a new gang can use the GPU only now
+    nxt = set_defaults(mk_job("next", workers=2, master=False))
+    ctl.create_job(nxt)
+    ctl.reconcile(nxt)
+    assert len(node.free_slots) == 0

@@ -48,6 +48,10 @@ class ControllerConfig:
     enable_gang_scheduling: bool = True
     enable_dag_scheduling: bool = True
     master_port_range: tuple = (20000, 30000)  # hostnetwork range analog
+    # graceful-kill window before SIGKILL escalation during cleanup;
+    # sized for the trainer's SIGTERM checkpoint (~8s for the 75 GiB
+    # flagship state, profiles/async_ckpt.log)
+    kill_grace_seconds: float = 60.0
     # PriorityClass-object analog (name -> value); consulted when a job
     # sets priorityClassName without an explicit priority
     priority_classes: dict = field(default_factory=dict)
@@ -76,6 +80,9 @@ class JobController:
         self._events_by_job: dict[str, object] = {}
         self._event_last: dict = {}
         self._ports: dict[str, int] = {}
+        # handles whose processes outlived their job (graceful-kill
+        # window / CleanPodPolicy None): (handle, escalate-deadline|None)
+        self._reaping: list = []
         self._last_jobjson: dict[str, str] = {}
         import random as _random
         lo, hi = self.cfg.master_port_range
@@ -546,12 +553,46 @@ class JobController:
         hs = self.handles.get(job.name, {})
         for h in list(hs.values()):
             if kill_all and not h.finished:
-                self.runtime.kill(h)
-            if h.gpu_slots:
-                self.node.release(h.gpu_slots)
-                h.gpu_slots = ()
+                self.runtime.kill(h)       # graceful: trainer checkpoints
+                self.runtime.poll(h)
+            if h.finished:
+                if h.gpu_slots:
+                    self.node.release(h.gpu_slots)
+                    h.gpu_slots = ()
+            else:
+                # still alive (checkpointing on SIGTERM, stuck, or
+                # deliberately left by CleanPodPolicy None): the process
+                # OCCUPIES its physical GPUs, so the slots must stay
+                # held until it actually exits — releasing now would
+                # hand a busy GPU to the next gang. reap_pass() polls,
+                # escalates to SIGKILL after kill_grace_seconds (only
+                # for killed tasks), and releases on real exit.
+                deadline = (time.time() + self.cfg.kill_grace_seconds
+                            if kill_all else None)
+                self._reaping.append((h, deadline))
         if self.gang is not None and job.deleted:
             self.gang.delete_pod_group(job.name)
+
+    def reap_pass(self):
+        """Poll handles whose processes outlived their job: release GPU
+        slots only when the process is really gone; SIGKILL tasks that
+        ignored the graceful kill past their deadline."""
+        still = []
+        for h, deadline in self._reaping:
+            self.runtime.poll(h)
+            if h.finished:
+                if h.gpu_slots:
+                    self.node.release(h.gpu_slots)
+                    h.gpu_slots = ()
+                continue
+            if deadline is not None and time.time() >= deadline:
+                self.event(h.job_name, "Warning", "KillEscalated",
+                           f"{h.task_type.value}-{h.index} ignored SIGTERM "
+                           f"for {self.cfg.kill_grace_seconds:.0f}s; SIGKILL")
+                self.runtime.kill(h, grace=False)
+                deadline = None  # escalated once; keep polling for exit
+            still.append((h, deadline))
+        self._reaping = still
 
     def _maybe_ttl_cleanup(self, job: TorchJob):
         ttl = job.run_policy.ttl_seconds_after_finished
@@ -610,6 +651,7 @@ class JobController:
     def reconcile_all(self):
         for job in list(self.jobs.values()):
             self.reconcile(job)
+        self.reap_pass()
 
     def tenant_gpu_usage(self, tenant: str) -> int:
         return self.tenant_resource_usage(tenant)["gpu"]
