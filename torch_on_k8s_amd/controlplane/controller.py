@@ -426,10 +426,17 @@ class JobController:
             if freed >= need:
                 break
             self.runtime.kill(h, grace=True)  # trainer checkpoints on TERM
+            self.runtime.poll(h)
             freed += len(h.gpu_slots)
-            if h.gpu_slots:
-                self.node.release(h.gpu_slots)
-                h.gpu_slots = ()
+            if h.finished:
+                if h.gpu_slots:
+                    self.node.release(h.gpu_slots)
+                    h.gpu_slots = ()
+            else:
+                # victim still checkpointing: its GPUs free only when it
+                # exits (reap_pass); the preempting gang admits then
+                self._reaping.append(
+                    (h, time.time() + self.cfg.kill_grace_seconds))
             # shrink the victim's spot replica count so its controller
             # does not immediately recreate the preempted replica
             ospec = self.jobs[other].tasks.get(h.task_type)
