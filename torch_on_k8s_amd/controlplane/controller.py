@@ -311,12 +311,20 @@ class JobController:
                 continue
             if h.finished:
                 self._handle_finished(job, t, h, hs)
-        # remove excess indices (scale-in)
+        # remove excess indices (scale-in); slots come back only when
+        # the victim process really exits (reap_pass)
         for key, h in list(hs.items()):
             if h.task_type == t and h.index >= desired:
-                self.runtime.kill(h)
-                if h.gpu_slots:
-                    self.node.release(h.gpu_slots)
+                if not h.finished:
+                    self.runtime.kill(h)
+                    self.runtime.poll(h)
+                if h.finished:
+                    if h.gpu_slots:
+                        self.node.release(h.gpu_slots)
+                        h.gpu_slots = ()
+                else:
+                    self._reaping.append(
+                        (h, time.time() + self.cfg.kill_grace_seconds))
                 hs.pop(key, None)
 
     def _start_task(self, job: TorchJob, t: TaskType, idx: int, hs: dict):
