@@ -103,3 +103,35 @@ def test_bad_spec_surfaces_event_and_skips(tmp_path):
                    "gpusPerTask: 0, command: ['true']}\n")
     mgr.step()
     assert "j" in mgr.controller.jobs
+
+
+def test_duplicate_spool_file_cannot_hijack_or_delete(tmp_path):
+    """Two spool files declaring the same job name: the second is
+    ignored with a DuplicateJobName warning, and REMOVING it must not
+    delete the first file's running job (k8s name-uniqueness analog)."""
+    import os
+    import yaml
+    from torch_on_k8s_amd.manager import Manager
+
+    mgr = Manager(str(tmp_path), num_gpus=1, sync_period=0.05)
+    doc = {"kind": "TorchJob", "metadata": {"name": "uniq"},
+           "spec": {"tasks": {"master": {
+               "replicas": 1, "gpusPerTask": 0, "command": ["sleep", "30"]}}}}
+    with open(os.path.join(mgr.spool, "a.yaml"), "w") as f:
+        yaml.safe_dump(doc, f)
+    mgr.step()
+    assert "uniq" in mgr.controller.jobs
+    with open(os.path.join(mgr.spool, "b.yaml"), "w") as f:
+        yaml.safe_dump(doc, f)
+    mgr.step()
+    assert any(e.reason == "DuplicateJobName"
+               for e in mgr.controller.events_for("uniq"))
+    os.unlink(os.path.join(mgr.spool, "b.yaml"))
+    mgr.step()
+    assert "uniq" in mgr.controller.jobs, \
+        "removing the duplicate file deleted the original job"
+    # removing the OWNING file still deletes
+    os.unlink(os.path.join(mgr.spool, "a.yaml"))
+    for _ in range(3):
+        mgr.step()
+    assert "uniq" not in mgr.controller.jobs

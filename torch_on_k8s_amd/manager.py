@@ -118,6 +118,17 @@ class Manager:
                 # through the elastic generation bump so running gangs
                 # checkpoint + restart at the new world size
                 self._apply_update(existing, job)
+            else:
+                # a DIFFERENT spool file already owns this job name
+                # (k8s would reject the create: names are unique). Track
+                # the file without ownership so removing it can't delete
+                # the other file's job.
+                self.controller.event(
+                    job.name, "Warning", "DuplicateJobName",
+                    f"{f} ignored: job {job.name!r} already exists from "
+                    f"another spool file")
+                self._spooled[f] = ("", mtime)
+                continue
             self._spooled[f] = (job.name, mtime)
         for f in set(self._spooled) - files:   # spool file removed
             name, _ = self._spooled.pop(f)
