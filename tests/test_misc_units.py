@@ -540,3 +540,17 @@ def test_client_validate_cli(tmp_path):
          "--workdir", str(tmp_path), "validate", str(bad)],
         capture_output=True, text=True)
     assert r.returncode == 1 and "INVALID" in r.stderr
+
+
+def test_client_delete_finds_mismatched_filename(tmp_path):
+    """delete(name) must work when the spool file's name differs from
+    metadata.name (hand-dropped manifests)."""
+    import os
+    from torch_on_k8s_amd.client import TorchJobClient
+    cli = TorchJobClient(str(tmp_path))
+    with open(os.path.join(cli.spool, "whatever.yaml"), "w") as f:
+        f.write("kind: TorchJob\nmetadata: {name: oddname}\n"
+                "spec: {tasks: {master: {replicas: 1}}}\n")
+    assert cli.delete("oddname") is True
+    assert not os.listdir(cli.spool)
+    assert cli.delete("oddname") is False  # already gone

@@ -45,7 +45,30 @@ class TorchJobClient:
             os.unlink(os.path.join(self.spool, f"{name}.yaml"))
             return True
         except FileNotFoundError:
+            pass
+        # a hand-dropped spool file may use any filename: find the one
+        # whose metadata.name owns this job
+        import yaml
+        try:
+            files = os.listdir(self.spool)
+        except OSError:
             return False
+        for f in files:
+            if not f.endswith((".yaml", ".yml", ".json")):
+                continue
+            path = os.path.join(self.spool, f)
+            try:
+                with open(path) as fh:
+                    doc = yaml.safe_load(fh) or {}
+            except (OSError, yaml.YAMLError):
+                continue
+            if (doc.get("metadata") or {}).get("name") == name:
+                try:
+                    os.unlink(path)
+                    return True
+                except OSError:
+                    return False
+        return False
 
     def get(self, name: str) -> dict | None:
         try:
