@@ -135,3 +135,59 @@ def test_duplicate_spool_file_cannot_hijack_or_delete(tmp_path):
     for _ in range(3):
         mgr.step()
     assert "uniq" not in mgr.controller.jobs
+
+
+def test_resubmitted_job_name_starts_fresh(tmp_path):
+    """Deleting a job and resubmitting the SAME name must train from
+    scratch, not silently resume the old incarnation's checkpoint
+    (TOK_TRAIN_STEPS already 'reached' meant instant Succeeded with
+    zero training)."""
+    import json
+    import os
+    import time
+    import yaml
+    from torch_on_k8s_amd.manager import Manager
+    from torch_on_k8s_amd.controlplane.api import JobConditionType
+
+    mgr = Manager(str(tmp_path), num_gpus=0, sync_period=0.05)
+    env = {"TOK_BACKEND": "gloo", "TOK_TRAIN_STEPS": "2",
+           "TOK_TRAINER_CONFIG": json.dumps(
+               {"model": "llama-tiny", "micro_batch": 1, "seq_len": 32})}
+    doc = {"kind": "TorchJob", "metadata": {"name": "fresh"},
+           "spec": {"tasks": {"master": {"replicas": 1, "gpusPerTask": 0,
+                                         "env": env}}}}
+
+    def run_once():
+        with open(os.path.join(mgr.spool, "fresh.yaml"), "w") as f:
+            yaml.safe_dump(doc, f)
+        deadline = time.time() + 180
+        while time.time() < deadline:
+            mgr.step()
+            job = mgr.controller.jobs.get("fresh")
+            if job is not None and job.status.phase in (
+                    JobConditionType.SUCCEEDED, JobConditionType.FAILED):
+                return job
+            time.sleep(0.05)
+        raise AssertionError("job never finished")
+
+    job1 = run_once()
+    assert job1.status.phase == JobConditionType.SUCCEEDED
+    # plant a fake old checkpoint claiming step 2 (== TOK_TRAIN_STEPS)
+    ck = tmp_path / "jobs" / "fresh" / "ckpt"
+    ck.mkdir(parents=True, exist_ok=True)
+    (ck / "meta.json").write_text('{"step": 2}')
+    os.unlink(os.path.join(mgr.spool, "fresh.yaml"))
+    for _ in range(5):
+        mgr.step()
+        time.sleep(0.05)
+    assert "fresh" not in mgr.controller.jobs
+    job2 = run_once()
+    assert job2.status.phase == JobConditionType.SUCCEEDED
+    # the stale ckpt was cleared at create: the new life really trained
+    # (its metrics.json exists and reports step 2 of THIS run)
+    m = json.load(open(tmp_path / "jobs" / "fresh" / "metrics.json"))
+    assert m["step"] == 2
+    # and no trace of the planted stale checkpoint survived the create
+    log = (tmp_path / "jobs" / "fresh" /
+           "fresh-master-0.log").read_text()
+    assert "resumed at step" not in log

@@ -19,6 +19,7 @@ reservation/rewrite semantics live in _master_port.
 """
 from __future__ import annotations
 
+import os
 import time
 from dataclasses import dataclass, field
 
@@ -83,6 +84,8 @@ class JobController:
         # handles whose processes outlived their job (graceful-kill
         # window / CleanPodPolicy None): (handle, escalate-deadline|None)
         self._reaping: list = []
+        # deleted jobs whose state dir awaits clearing once reaped
+        self._pending_state_clear: set = set()
         self._last_jobjson: dict[str, str] = {}
         import random as _random
         lo, hi = self.cfg.master_port_range
@@ -133,6 +136,25 @@ class JobController:
             # filter counts their own live GPUs against the tenant
             self.reconcile(job)
         return job
+
+    def _clear_stale_state(self, job_name: str):
+        """Remove a previous incarnation's dynamic state (checkpoints,
+        agent/bench/metrics files, task records) from the job state dir.
+        Logs and output/ are kept — they are the previous life's
+        artifacts, not state the new gang reads."""
+        workdir = getattr(self.runtime, "workdir", None)
+        if not workdir:
+            return
+        import shutil
+        jobdir = os.path.join(workdir, job_name)
+        for sub in ("ckpt", "tasks"):
+            shutil.rmtree(os.path.join(jobdir, sub), ignore_errors=True)
+        for f in ("agent.json", "job.json", "metrics.json", "bench.json"):
+            try:
+                os.unlink(os.path.join(jobdir, f))
+            except OSError:
+                pass
+        self._last_jobjson.pop(job_name, None)
 
     def delete_job(self, name: str):
         job = self.jobs.get(name)
@@ -205,6 +227,19 @@ class JobController:
 
         if job.deleted:
             self._cleanup(job, kill_all=True)
+            # deletion is the user declaring the job DONE: its dynamic
+            # state (checkpoints, records) must not leak into a later
+            # resubmission of the same name — a stale ckpt/ would
+            # silently resume and TOK_TRAIN_STEPS already "reached"
+            # means instant Succeeded with zero training. A manager
+            # CRASH is different: no delete happens, so recreated jobs
+            # still resume their checkpoints. Tasks still exiting
+            # (reaping) may write a final SIGTERM checkpoint — defer
+            # the clear until they are gone.
+            if any(h.job_name == job.name for h, _ in self._reaping):
+                self._pending_state_clear.add(job.name)
+            else:
+                self._clear_stale_state(job.name)
             self.jobs.pop(job.name, None)
             self.handles.pop(job.name, None)
             # daemon hygiene: every per-job cache is released with the
@@ -608,6 +643,12 @@ class JobController:
                 deadline = None  # escalated once; keep polling for exit
             still.append((h, deadline))
         self._reaping = still
+        if self._pending_state_clear:
+            live = {h.job_name for h, _ in still}
+            for name in list(self._pending_state_clear):
+                if name not in live:
+                    self._clear_stale_state(name)
+                    self._pending_state_clear.discard(name)
 
     def _maybe_ttl_cleanup(self, job: TorchJob):
         ttl = job.run_policy.ttl_seconds_after_finished
