@@ -462,3 +462,37 @@ def test_sigterm_ignoring_task_keeps_slots_until_reaped():
     ctl.create_job(nxt)
     ctl.reconcile(nxt)
     assert len(node.free_slots) == 0
+
+
+def test_policy_none_straggler_reaped_on_natural_exit():
+    """CleanPodPolicy None: a straggler left running at job completion
+    (the GPU-less AIMaster sidecar — Succeeded requires master+workers
+    done) is tracked by the reap list WITHOUT a kill deadline, never
+    killed, and drained when it exits on its own."""
+    from torch_on_k8s_amd.controlplane.api import (CleanPodPolicy,
+                                                   JobConditionType,
+                                                   RunPolicy, TaskPhase,
+                                                   TaskSpec)
+
+    node = NodeState(num_gpus=2)
+    rt = FakeRuntime()
+    ctl = JobController(node, rt)
+    job = mk_job("straggle", workers=0)
+    job.tasks[TaskType.AIMASTER] = TaskSpec(replicas=1)
+    job.run_policy = RunPolicy(clean_task_policy=CleanPodPolicy.NONE)
+    set_defaults(job)
+    ctl.create_job(job)
+    for _ in range(3):
+        ctl.reconcile(job)
+    rt.set_phase(("straggle", TaskType.MASTER, 0), TaskPhase.SUCCEEDED, 0)
+    for _ in range(3):
+        ctl.reconcile(job)
+    assert job.status.phase == JobConditionType.SUCCEEDED
+    assert not rt.killed                      # None policy: no kill
+    assert any(h.task_type == TaskType.AIMASTER and d is None
+               for h, d in ctl._reaping)      # polled, no SIGKILL clock
+    ctl.reap_pass()
+    assert ctl._reaping                       # still alive: still tracked
+    rt.set_phase(("straggle", TaskType.AIMASTER, 0), TaskPhase.SUCCEEDED, 0)
+    ctl.reap_pass()
+    assert not ctl._reaping                   # drained on natural exit
