@@ -496,3 +496,39 @@ def test_policy_none_straggler_reaped_on_natural_exit():
     rt.set_phase(("straggle", TaskType.AIMASTER, 0), TaskPhase.SUCCEEDED, 0)
     ctl.reap_pass()
     assert not ctl._reaping                   # drained on natural exit
+
+
+def test_per_task_termination_grace_overrides_default():
+    """terminationGracePeriodSeconds (pod-spec analog) overrides the
+    manager's kill_grace_seconds for that task's SIGKILL escalation."""
+    from torch_on_k8s_amd.controlplane.api import TaskPhase
+    from torch_on_k8s_amd.controlplane.controller import ControllerConfig
+    from torch_on_k8s_amd.controlplane.jobspec import job_from_dict
+
+    class StubbornRuntime(FakeRuntime):
+        def kill(self, h, grace=True):
+            self.killed.append((h.key, grace))
+            if not grace:
+                h.phase = TaskPhase.FAILED
+                h.exit_code = 137
+
+    # CRD dialect carries the grace in the pod template
+    job = job_from_dict({"metadata": {"name": "graceful"}, "spec": {
+        "torchTaskSpecs": {"Master": {"numTasks": 1, "template": {"spec": {
+            "terminationGracePeriodSeconds": 0.05,
+            "containers": [{"name": "torch", "resources": {
+                "limits": {"amd.com/gpu": 1}}}]}}}}}})
+    assert job.tasks[TaskType.MASTER].termination_grace_seconds == 0.05
+
+    node = NodeState(num_gpus=1)
+    rt = StubbornRuntime()
+    ctl = JobController(node, rt, ControllerConfig(kill_grace_seconds=9999))
+    ctl.create_job(job)
+    ctl.reconcile(job)
+    ctl.delete_job(job.name)
+    assert len(node.free_slots) == 0  # held while SIGTERM pending
+    time.sleep(0.1)
+    ctl.reap_pass()  # per-task 0.05s grace elapsed despite 9999 default
+    assert any(not g for _, g in rt.killed), "never escalated to SIGKILL"
+    ctl.reap_pass()
+    assert len(node.free_slots) == 1

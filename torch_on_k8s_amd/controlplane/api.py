@@ -115,6 +115,9 @@ class TaskSpec:
     # entrypoint override: argv list; None -> framework training entrypoint
     command: list[str] | None = None
     env: dict = field(default_factory=dict)
+    # pod terminationGracePeriodSeconds analog: per-task override of the
+    # manager's kill_grace_seconds (SIGTERM -> SIGKILL window)
+    termination_grace_seconds: float | None = None
 
 
 @dataclass

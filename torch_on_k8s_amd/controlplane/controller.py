@@ -137,6 +137,14 @@ class JobController:
             self.reconcile(job)
         return job
 
+    def _grace_for(self, job: TorchJob | None, h: TaskHandle) -> float:
+        """SIGTERM->SIGKILL window: the task's pod-spec
+        terminationGracePeriodSeconds analog, else the manager default."""
+        spec = job.tasks.get(h.task_type) if job is not None else None
+        if spec is not None and spec.termination_grace_seconds is not None:
+            return spec.termination_grace_seconds
+        return self.cfg.kill_grace_seconds
+
     def _clear_stale_state(self, job_name: str):
         """Remove a previous incarnation's dynamic state (checkpoints,
         agent/bench/metrics files, task records) from the job state dir.
@@ -359,7 +367,7 @@ class JobController:
                         h.gpu_slots = ()
                 else:
                     self._reaping.append(
-                        (h, time.time() + self.cfg.kill_grace_seconds))
+                        (h, time.time() + self._grace_for(job, h)))
                 hs.pop(key, None)
 
     def _start_task(self, job: TorchJob, t: TaskType, idx: int, hs: dict):
@@ -479,7 +487,8 @@ class JobController:
                 # victim still checkpointing: its GPUs free only when it
                 # exits (reap_pass); the preempting gang admits then
                 self._reaping.append(
-                    (h, time.time() + self.cfg.kill_grace_seconds))
+                    (h, time.time() +
+                     self._grace_for(self.jobs.get(other), h)))
             # shrink the victim's spot replica count so its controller
             # does not immediately recreate the preempted replica
             ospec = self.jobs[other].tasks.get(h.task_type)
@@ -617,7 +626,7 @@ class JobController:
                 # hand a busy GPU to the next gang. reap_pass() polls,
                 # escalates to SIGKILL after kill_grace_seconds (only
                 # for killed tasks), and releases on real exit.
-                deadline = (time.time() + self.cfg.kill_grace_seconds
+                deadline = (time.time() + self._grace_for(job, h)
                             if kill_all else None)
                 self._reaping.append((h, deadline))
         if self.gang is not None and job.deleted:
@@ -638,7 +647,7 @@ class JobController:
             if deadline is not None and time.time() >= deadline:
                 self.event(h.job_name, "Warning", "KillEscalated",
                            f"{h.task_type.value}-{h.index} ignored SIGTERM "
-                           f"for {self.cfg.kill_grace_seconds:.0f}s; SIGKILL")
+                           f"past its grace period; SIGKILL")
                 self.runtime.kill(h, grace=False)
                 deadline = None  # escalated once; keep polling for exit
             still.append((h, deadline))

@@ -77,6 +77,9 @@ def _task_spec(d: dict) -> TaskSpec:
         command=d.get("command"),
         env={str(k): str(v) for k, v in (d.get("env") or {}).items()},
     )
+    if d.get("terminationGracePeriodSeconds") is not None:
+        spec.termination_grace_seconds = float(
+            d["terminationGracePeriodSeconds"])
     _check_non_negative(gpusPerTask=spec.gpus_per_task,
                         cpusPerTask=spec.cpus_per_task,
                         memMbPerTask=spec.mem_mb_per_task)
@@ -116,6 +119,9 @@ def _task_spec_from_crd(d: dict) -> TaskSpec:
             priority_class_name=st.get("priorityClassName", ""),
             labels=dict(st.get("labels") or {}))
     tmpl = (d.get("template") or {}).get("spec") or {}
+    if tmpl.get("terminationGracePeriodSeconds") is not None:
+        spec.termination_grace_seconds = float(
+            tmpl["terminationGracePeriodSeconds"])
     containers = tmpl.get("containers") or []
     cont = None
     for c in containers:
@@ -215,6 +221,9 @@ def job_to_crd_dict(job: TorchJob) -> dict:
             cont["resources"] = {"limits": res}
         d = {"numTasks": s.replicas,
              "template": {"spec": {"containers": [cont]}}}
+        if s.termination_grace_seconds is not None:
+            d["template"]["spec"]["terminationGracePeriodSeconds"] = \
+                s.termination_grace_seconds
         if s.restart_policy is not None:
             inv = {v: k for k, v in _CRD_RESTART.items()}
             d["restartPolicy"] = inv.get(s.restart_policy,

@@ -43,7 +43,8 @@ class Manager:
     def __init__(self, workdir: str, num_gpus: int = 8, quotas=None,
                  gates: feat.FeatureGates | None = None,
                  sync_period: float = 0.5, storage_spec: dict | None = None,
-                 priority_classes: dict | None = None):
+                 priority_classes: dict | None = None,
+                 kill_grace_seconds: float = 60.0):
         self.workdir = workdir
         self.spool = os.path.join(workdir, "spool")
         self.status_dir = os.path.join(workdir, "status")
@@ -62,7 +63,8 @@ class Manager:
         cfg = ControllerConfig(
             enable_gang_scheduling=self.gates.enabled(feat.GANG_SCHEDULING),
             enable_dag_scheduling=self.gates.enabled(feat.DAG_SCHEDULING),
-            priority_classes=dict(priority_classes or {}))
+            priority_classes=dict(priority_classes or {}),
+            kill_grace_seconds=kill_grace_seconds)
         self.controller = JobController(
             node, runtime, cfg, metrics=self.metrics,
             model_registry=self.registry, elastic=ElasticScaler())
@@ -270,6 +272,9 @@ def main():
     ap.add_argument("--priority-class", action="append", default=[],
                     help="name=value PriorityClass-object analog consulted "
                          "when a job sets priorityClassName; repeatable")
+    ap.add_argument("--kill-grace", type=float, default=60.0,
+                    help="SIGTERM->SIGKILL window in seconds for task "
+                         "cleanup (terminationGracePeriodSeconds default)")
     ap.add_argument("--storage", default="",
                     help='storage spec JSON, e.g. {"nfs": {"server": '
                          '"10.0.0.2", "path": "/exports/models"}} or '
@@ -293,7 +298,8 @@ def main():
                   sync_period=args.sync_period,
                   storage_spec=json.loads(args.storage) if args.storage
                   else None,
-                  priority_classes=prio_classes or None)
+                  priority_classes=prio_classes or None,
+                  kill_grace_seconds=args.kill_grace)
     start_metrics_server(args.metrics_addr)
     log.info("workdir=%s gpus=%d gates=%s", args.workdir, args.num_gpus,
              gates.as_dict())
