@@ -294,3 +294,20 @@ def test_emitted_crd_fields_declared_in_schema():
                      if f"{k}:" not in schema_text and
                      f"{k} " not in schema_text)
     assert not missing, f"emitted keys absent from CRD schema: {missing}"
+
+
+def test_production_knobs_sample_parses():
+    """The kitchen-sink sample exercises every late-round policy field
+    in one manifest and must round-trip cleanly."""
+    with open(os.path.join(REPO, "configs", "samples",
+                           "production-knobs.yaml")) as f:
+        job = job_from_yaml(f.read())
+    assert job.scheduling.priority_class_name == "gold"
+    assert job.elastic.rdzv_backend == "tcpstore"
+    assert job.tasks[TaskType.MASTER].termination_grace_seconds == 90
+    w = job.tasks[TaskType.WORKER]
+    assert w.spot.priority_class_name == "spot-low"
+    assert w.mem_mb_per_task == 65536
+    job2 = job_from_dict(job_to_crd_dict(job))
+    assert job2.tasks[TaskType.WORKER].termination_grace_seconds == 90
+    assert job2.scheduling.priority_class_name == "gold"
