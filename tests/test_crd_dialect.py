@@ -311,3 +311,18 @@ def test_production_knobs_sample_parses():
     job2 = job_from_dict(job_to_crd_dict(job))
     assert job2.tasks[TaskType.WORKER].termination_grace_seconds == 90
     assert job2.scheduling.priority_class_name == "gold"
+
+
+def test_all_sample_manifests_parse():
+    """Every file under configs/samples/ must parse in whichever dialect
+    it is written (docs stay executable)."""
+    d = os.path.join(REPO, "configs", "samples")
+    parsed = 0
+    for f in sorted(os.listdir(d)):
+        if not f.endswith((".yaml", ".yml")):
+            continue
+        with open(os.path.join(d, f)) as fh:
+            job = job_from_yaml(fh.read())
+        assert job.tasks, f
+        parsed += 1
+    assert parsed >= 6  # grows with new samples
