@@ -113,3 +113,54 @@ def test_chaos_with_elastic_scaling():
                if h.task_type == TaskType.WORKER]
     assert len(workers) == job.tasks[TaskType.WORKER].replicas
     assert all(h.generation == job.generation for h in workers)
+
+
+@pytest.mark.parametrize("seed", [11, 47, 333])
+def test_chaos_delete_resubmit_storm(seed):
+    """Random deletes + resubmits of the same names under failure
+    mayhem: slots are conserved (counting the reap list), the reap list
+    and pending-state-clear set drain, and the final world is clean."""
+    rng = random.Random(seed)
+    node = NodeState(num_gpus=8)
+    rt = FakeRuntime()
+    ctl = JobController(node, rt, ControllerConfig())
+
+    def submit(i):
+        return ctl.create_job(TorchJob(
+            name=f"dr-{i}",
+            tasks={TaskType.MASTER: TaskSpec(replicas=1, gpus_per_task=0),
+                   TaskType.WORKER: TaskSpec(
+                       replicas=rng.randint(1, 3), gpus_per_task=1)},
+            run_policy=RunPolicy(backoff_limit=1)))
+
+    for i in range(3):
+        submit(i)
+    for it in range(400):
+        live = [h for h in rt.tasks.values() if not h.finished]
+        r = rng.random()
+        if live and r < 0.4:
+            h = rng.choice(live)
+            code = rng.choice(EXIT_CODES)
+            rt.set_phase(h.key, TaskPhase.SUCCEEDED if code == 0
+                         else TaskPhase.FAILED, exit_code=code)
+        elif r < 0.5 and ctl.jobs:
+            ctl.delete_job(rng.choice(list(ctl.jobs)))
+        elif r < 0.6:
+            name = f"dr-{rng.randint(0, 2)}"
+            if name not in ctl.jobs:
+                submit(int(name[3:]))
+        ctl.reconcile_all()
+        held = sum(len(h.gpu_slots) for hs in ctl.handles.values()
+                   for h in hs.values())
+        held += sum(len(h.gpu_slots) for h, _ in ctl._reaping)
+        used = 8 - len(node.free_slots)
+        assert held == used, f"slot leak at iter {it}"
+
+    for name in list(ctl.jobs):
+        ctl.delete_job(name)
+    for _ in range(5):
+        ctl.reconcile_all()
+    assert not ctl._reaping
+    assert not ctl._pending_state_clear
+    assert len(node.free_slots) == 8
+    assert not ctl.jobs and not ctl.handles
