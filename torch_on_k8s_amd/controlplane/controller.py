@@ -715,7 +715,16 @@ class JobController:
 
     def reconcile_all(self):
         for job in list(self.jobs.values()):
-            self.reconcile(job)
+            try:
+                self.reconcile(job)
+            except Exception as e:  # reconcile isolation: one poisoned
+                # job must not take the manager down for every tenant
+                # (controller-runtime requeues errored reconciles; here
+                # the next 0.5s pass is the requeue)
+                import traceback
+                traceback.print_exc()
+                self.event(job.name, "Warning", "ReconcileError",
+                           f"{type(e).__name__}: {e}")
         self.reap_pass()
 
     def tenant_gpu_usage(self, tenant: str) -> int:
