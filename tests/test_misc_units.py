@@ -554,3 +554,25 @@ def test_client_delete_finds_mismatched_filename(tmp_path):
     assert cli.delete("oddname") is True
     assert not os.listdir(cli.spool)
     assert cli.delete("oddname") is False  # already gone
+
+
+def test_metrics_reader_rejects_non_numeric_latency(tmp_path):
+    """A custom entrypoint writing null/string step_time_s must yield no
+    observation (not a TypeError inside the autoscaler math)."""
+    import json as _json
+    from torch_on_k8s_amd.controlplane.api import (TaskSpec, TaskType,
+                                                   TorchJob, set_defaults)
+    from torch_on_k8s_amd.controlplane.elastic import read_trainer_metrics
+
+    p = tmp_path / "metrics.json"
+    read = read_trainer_metrics(lambda job: str(p))
+    job = set_defaults(TorchJob(name="m", tasks={
+        TaskType.MASTER: TaskSpec(replicas=1)}))
+    for bad in ({"step": 1, "step_time_s": None},
+                {"step": 1, "step_time_s": "fast"},
+                {"step": "x", "step_time_s": 0.5}):
+        p.write_text(_json.dumps(bad))
+        assert read(job) is None, bad
+    p.write_text(_json.dumps({"step": 3, "step_time_s": 0.25}))
+    obs = read(job)
+    assert obs is not None and obs.latency == 0.25 and obs.step == 3

@@ -287,9 +287,13 @@ def read_trainer_metrics(metrics_path_fn):
         try:
             with open(path) as f:
                 rec = json.load(f)
-            return Observation(step=rec["step"], latency=rec["step_time_s"],
+            lat = rec["step_time_s"]
+            if not isinstance(lat, (int, float)):
+                return None  # custom entrypoints may write null/strings;
+                # a non-numeric latency must not poison the decide() math
+            return Observation(step=int(rec["step"]), latency=float(lat),
                                tokens_per_s=rec.get("tokens_per_s") or 0.0,
                                loss=rec.get("loss") or 0.0)
-        except (OSError, ValueError, KeyError):
+        except (OSError, ValueError, TypeError, KeyError):
             return None
     return _read

@@ -190,8 +190,14 @@ class Manager:
                 continue
             if job.status.phase != JobConditionType.RUNNING:
                 continue
-            self.autoscaler.observe(job)
-            decision = self.autoscaler.decide(job)
+            try:
+                self.autoscaler.observe(job)
+                decision = self.autoscaler.decide(job)
+            except Exception as e:  # same isolation as reconcile_all:
+                # one job's bad metrics must not stop every autoscaler
+                self.controller.event(job.name, "Warning", "AutoscaleError",
+                                      f"{type(e).__name__}: {e}")
+                continue
             if decision is not None:
                 st = job.status.elastic
                 self.controller.event(
