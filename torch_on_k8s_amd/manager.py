@@ -159,11 +159,15 @@ class Manager:
             evs = self.controller.events_for(name)
             doc = json.dumps(job_status_dict(job, evs), indent=2)
             if self._last_status.get(name) != doc:
-                tmp = path + ".tmp"
-                with open(tmp, "w") as f:
-                    f.write(doc)
-                os.replace(tmp, path)
-                self._last_status[name] = doc
+                try:
+                    tmp = path + ".tmp"
+                    with open(tmp, "w") as f:
+                        f.write(doc)
+                    os.replace(tmp, path)
+                    self._last_status[name] = doc
+                except OSError as e:  # disk full etc.: stale status
+                    # beats taking every tenant's controller down
+                    log.warning("status write failed for %s: %s", name, e)
             # data-plane gauges from the trainer's structured metrics
             try:
                 with open(os.path.join(self.workdir, "jobs", name,
