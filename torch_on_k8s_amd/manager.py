@@ -172,8 +172,10 @@ class Manager:
             try:
                 with open(os.path.join(self.workdir, "jobs", name,
                                        "metrics.json")) as f:
-                    self.metrics.set_training_metrics(name, json.load(f))
-            except (OSError, ValueError):
+                    rec = json.load(f)
+                if isinstance(rec, dict):  # metrics.json is task-written
+                    self.metrics.set_training_metrics(name, rec)
+            except (OSError, ValueError, TypeError):
                 pass
         # drop status files of deleted jobs (daemon hygiene)
         try:
