@@ -179,3 +179,71 @@ def test_leader_election_lock(tmp_path):
     m1.release_leadership()
     m2.acquire_leadership(block=False)
     m2.release_leadership()
+
+
+def test_zero_downtime_upgrade_with_overlapping_managers(tmp_path):
+    """The upgrade story end to end across PROCESSES: manager B starts
+    while A is still leader, BLOCKS on the manager.lock flock, and on
+    A's SIGTERM takes over and adopts A's still-running task (same
+    pid)."""
+    import signal
+    import subprocess
+    import sys
+
+    def start_mgr():
+        return subprocess.Popen(
+            [sys.executable, "-m", "torch_on_k8s_amd.manager",
+             "--workdir", str(tmp_path), "--num-gpus", "0",
+             "--metrics-addr", "0", "--sync-period", "0.05"],
+            env=dict(os.environ, PYTHONPATH=ROOT),
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+
+    a = start_mgr()
+    b = None
+    rec = tmp_path / "jobs" / "zd" / "tasks" / "zd-master-0.json"
+    try:
+        spec = ("kind: TorchJob\nmetadata: {name: zd}\nspec:\n  tasks:\n"
+                "    master: {replicas: 1, gpusPerTask: 0, env: {"
+                "TOK_BACKEND: gloo, TOK_TRAIN_STEPS: '400', "
+                "TOK_STEP_DELAY: '0.2', TOK_TRAINER_CONFIG: '"
+                '{"model": "llama-tiny", "micro_batch": 1, "seq_len": 32}'
+                "'}}\n")
+        os.makedirs(tmp_path / "spool", exist_ok=True)
+        (tmp_path / "spool" / "zd.yaml").write_text(spec)
+        t0 = time.time()
+        while time.time() - t0 < 60 and not rec.exists():
+            time.sleep(0.2)
+        assert rec.exists(), a.stdout
+        task_pid = json.load(open(rec))["pid"]
+
+        b = start_mgr()                      # overlapping daemon
+        time.sleep(3.0)
+        assert b.poll() is None              # B is alive, blocked on lock
+        # B must NOT have acted yet: A's pid is still in the lockfile
+        assert (tmp_path / "manager.lock").read_text().strip() == str(a.pid)
+
+        os.kill(a.pid, signal.SIGTERM)       # upgrade: stop A
+        assert a.wait(timeout=30) == 0
+        # B becomes leader and adopts the SAME process
+        t0 = time.time()
+        while time.time() - t0 < 60:
+            txt = (tmp_path / "manager.lock").read_text().strip()
+            if txt == str(b.pid):
+                break
+            time.sleep(0.2)
+        assert (tmp_path / "manager.lock").read_text().strip() == str(b.pid)
+        os.kill(task_pid, 0)                 # task survived both managers
+        # and B is reconciling it: status file keeps updating
+        st = tmp_path / "status" / "zd.json"
+        t0 = time.time()
+        while time.time() - t0 < 60 and not st.exists():
+            time.sleep(0.2)
+        assert st.exists(), b.stdout
+    finally:
+        for p in (a, b):
+            if p is not None and p.poll() is None:
+                p.kill()
+        try:
+            os.kill(json.load(open(rec))["pid"], signal.SIGKILL)
+        except (OSError, ValueError, FileNotFoundError):
+            pass

@@ -229,8 +229,11 @@ class Manager:
         process lifetime; the OS releases it on ANY exit, so a crashed
         leader never wedges the next one."""
         import fcntl
+        # "a" not "w": opening must NOT truncate — a second daemon
+        # blocked on the flock would otherwise wipe the live leader's
+        # pid record the moment it starts waiting
         self._lock_file = open(os.path.join(self.workdir, "manager.lock"),
-                               "w")
+                               "a")
         try:
             fcntl.flock(self._lock_file,
                         fcntl.LOCK_EX | (0 if block else fcntl.LOCK_NB))
@@ -239,6 +242,7 @@ class Manager:
             self._lock_file = None
             raise RuntimeError(
                 f"another manager owns {self.workdir} (manager.lock held)")
+        self._lock_file.seek(0)
         self._lock_file.truncate(0)
         self._lock_file.write(f"{os.getpid()}\n")
         self._lock_file.flush()
