@@ -25,7 +25,8 @@ def test_manager_autoscales_on_metrics(tmp_path):
     env = {
         "TOK_BACKEND": "gloo",
         "TOK_TRAIN_STEPS": str(steps),
-        "TOK_STEP_DELAY": "0.15",
+        "TOK_STEP_DELAY": "0.2",  # slow enough that metricWindow samples
+        # land before the run ends even on a contended box (xdist -n 4)
         "TOK_TRAINER_CONFIG": json.dumps(
             {"model": "llama-tiny", "micro_batch": 1, "seq_len": 32}),
         "PYTHONPATH": ROOT,
@@ -47,7 +48,7 @@ def test_manager_autoscales_on_metrics(tmp_path):
 
     job = None
     scaled = False
-    deadline = time.time() + 300
+    deadline = time.time() + 360
     try:
         while time.time() < deadline:
             mgr.step()
